@@ -1,0 +1,150 @@
+// kubebrain_amd/csrc/slab_dev.h — host-visible interface of the HBM slab
+// engine (implemented in slab.hip, gfx950 kernels). PRODUCT code: no CPU
+// fallback; creation fails without a HIP device.
+//
+// Data model: one sorted columnar run in HBM, ordered by
+// (userKey, rev) == the reference's internal-key order
+// (coder/normal.go:42-50), plus an append-only value heap.
+#pragma once
+
+#include <cstdint>
+#include <string>
+#include <utility>
+#include <vector>
+
+namespace kbslab {
+
+constexpr int KEYW = 96;  // fixed-width zero-padded key column
+
+// meta word bits (per row)
+constexpr uint64_t M_SAME_NEXT = 1ull << 0;  // next row has the same user key
+constexpr uint64_t M_TOMB = 1ull << 1;       // value == "tombstone" (util.go:28)
+constexpr uint64_t M_FLAG9 = 1ull << 2;      // 9-byte revision value (rev.go:26)
+constexpr uint64_t M_EVENTS = 1ull << 3;     // key contains "/events/" (TTL)
+
+#ifdef __HIPCC__
+#define KB_HD __host__ __device__
+#else
+#define KB_HD
+#endif
+KB_HD inline uint64_t meta_make(bool tomb, bool flag9, bool events,
+                                uint32_t klen, uint32_t vlen) {
+  return (tomb ? M_TOMB : 0) | (flag9 ? M_FLAG9 : 0) | (events ? M_EVENTS : 0) |
+         ((uint64_t)(klen & 0xffff) << 16) | ((uint64_t)vlen << 32);
+}
+KB_HD inline uint32_t meta_klen(uint64_t m) { return (uint32_t)((m >> 16) & 0xffff); }
+KB_HD inline uint32_t meta_vlen(uint64_t m) { return (uint32_t)(m >> 32); }
+
+#pragma pack(push, 1)
+struct DevRangeQ {
+  uint8_t start[KEYW];
+  uint8_t end[KEYW];
+  uint64_t read_rev;
+  int64_t cap;        // winner-write cap per query (limit+1); <=0 => unbounded
+  int32_t count_only;
+  int32_t _pad;
+};
+struct DevGetQ {
+  uint8_t key[KEYW];
+  uint64_t read_rev;  // UINT64_MAX for "latest"
+};
+#pragma pack(pop)
+
+struct RangeResult {
+  int64_t written = 0;   // winners materialized (<= cap)
+  int64_t total = 0;     // winners seen before stop (exact when unbounded)
+  int64_t bytes = 0;     // gathered record bytes for this query
+  bool overflow = false; // per-query arena too small; retry with bigger qcap
+  // parsed records (filled only in d2h mode): {rev, key, val}
+  struct Rec { uint64_t rev; std::string key, val; };
+  std::vector<Rec> recs;
+};
+
+struct GetResult {
+  bool found = false;
+  bool tomb = false;
+  uint64_t rev = 0;
+  std::string val;
+};
+
+// sorted delta (memtable flush) — vo entries already ABSOLUTE heap offsets
+// (host adds the pre-merge heap_used base) or objRev for rev rows.
+struct DeltaRows {
+  std::vector<uint8_t> keys;  // m * KEYW
+  std::vector<uint64_t> meta, rev, vo;
+  std::vector<uint8_t> heap;  // new value bytes (4B-aligned records)
+  int64_t m = 0;
+};
+
+struct DumpRow {
+  std::string key;  // user key
+  uint64_t rev;
+  uint64_t meta;
+  std::string val;  // object rows: heap bytes; rev rows: empty (vo=objRev)
+  uint64_t vo;
+};
+
+struct Perf {
+  double scan_ms = 0, gather_ms = 0, get_ms = 0, compact_ms = 0, merge_ms = 0,
+         filter_ms = 0, pack_d2h_ms = 0;
+  int64_t scan_launches = 0, gather_launches = 0, get_launches = 0,
+          merges = 0, compacts = 0, filter_launches = 0;
+  int64_t rows_scanned = 0, bytes_gathered = 0, winners = 0,
+          filter_events = 0, filter_watchers = 0;
+};
+
+struct WatchFilterBatch {
+  // E events: key column (E*KEYW) + revision per event
+  std::vector<uint8_t> ekeys;
+  std::vector<uint64_t> erevs;
+  int64_t e = 0;
+};
+
+class Slab {
+ public:
+  // device < 0: use current device. Returns nullptr + *err on failure
+  // (including "no HIP device": the product path fails loudly).
+  static Slab* Create(int64_t max_rows, int64_t heap_cap, int device,
+                      std::string* err);
+  ~Slab();
+
+  int64_t rows() const;
+  int64_t heap_used() const;
+
+  // merge sorted delta rows into the slab (GPU merge-path by ranks;
+  // delta rev-rows REPLACE base rev-rows of the same key).
+  bool Merge(const DeltaRows& d, std::string* err);
+
+  // batched Range (the north-star kernel; scanner worker.run semantics,
+  // scanner.go:389-516). d2h=false leaves records in the device arena
+  // (bench "value" mode); d2h=true packs + copies + parses them.
+  bool RangeBatch(const std::vector<DevRangeQ>& qs, bool d2h,
+                  std::vector<RangeResult>* outs, std::string* err);
+
+  // batched MVCC point read (range.go:91-121 reverse-iter semantics)
+  bool GetBatch(const std::vector<DevGetQ>& qs, std::vector<GetResult>* outs,
+                std::string* err);
+
+  // compaction mark+sweep over encoded borders (compact.go:55-68 +
+  // scanner.go:444-491, 566-591). Bounds are (key96, rev) pairs.
+  struct Bound { uint8_t key[KEYW]; uint64_t rev; };
+  bool Compact(const std::vector<std::pair<Bound, Bound>>& borders,
+               uint64_t compact_rev, uint64_t timeout_rev, std::string* err);
+
+  // full slab dump for parity diffs (debug; D2H of all columns + used heap)
+  bool Dump(std::vector<DumpRow>* rows_out, std::string* err);
+
+  // watch fan-out filter: bitmap[w][ceil(E/64)] over registered watchers
+  bool WatchFilter(const WatchFilterBatch& b, std::vector<uint64_t>* bitmap,
+                   int64_t* n_watch_slots, std::string* err);
+  bool WatcherSet(int64_t slot, const uint8_t* prefix, uint32_t plen,
+                  uint64_t from_rev, std::string* err);  // slot grows table
+  void WatcherClear(int64_t slot);
+
+  Perf perf;
+
+  struct Impl;
+  Impl* p;
+};
+
+}  // namespace kbslab

@@ -1,0 +1,933 @@
+// kubebrain_amd/csrc/store.cc — host MVCC logic over the HBM slab.
+// PRODUCT code (no oracle/ dependency, no CPU scan fallback: all range /
+// point-read / compact / watch-filter compute runs on the GPU slab; the host
+// only owns the write path and the bounded memtable).
+// All semantics cites are into /root/reference.
+
+#include "store.h"
+
+#include <algorithm>
+#include <chrono>
+#include <cstring>
+
+namespace kbstore {
+
+using kbslab::DevGetQ;
+using kbslab::DevRangeQ;
+using kbslab::KEYW;
+using kbslab::M_EVENTS;
+using kbslab::M_FLAG9;
+using kbslab::M_SAME_NEXT;
+using kbslab::M_TOMB;
+
+static const Bytes kTombstone = "tombstone";  // backend/util.go:28
+static const Bytes kEvents = "/events/";      // backend/util.go:30
+
+Bytes U64ToBytes(uint64_t v) {
+  Bytes b(8, '\0');
+  for (int i = 7; i >= 0; --i) { b[i] = (char)(v & 0xff); v >>= 8; }
+  return b;
+}
+
+Bytes EncodeObjectKey(const Bytes& userKey, uint64_t rev) {
+  // coder/normal.go:42-50
+  Bytes key;
+  key.reserve(4 + userKey.size() + 9);
+  key.append("\x57\xfb\x80\x8b", 4);
+  key += userKey;
+  key += '$';
+  key += U64ToBytes(rev);
+  return key;
+}
+
+Bytes PrefixEnd(const Bytes& prefix) {  // util.go PrefixEnd
+  Bytes end = prefix;
+  for (int i = (int)end.size() - 1; i >= 0; --i) {
+    if ((uint8_t)end[i] < 0xff) {
+      end[i] = (char)((uint8_t)end[i] + 1);
+      end.resize(i + 1);
+      return end;
+    }
+  }
+  return Bytes("\x00", 1);
+}
+
+static void pad96(const Bytes& k, uint8_t out[KEYW]) {
+  memset(out, 0, KEYW);
+  memcpy(out, k.data(), std::min(k.size(), (size_t)KEYW));
+}
+
+static int64_t env_i64(const char* name, int64_t dflt) {
+  const char* v = getenv(name);
+  return v && *v ? atoll(v) : dflt;
+}
+
+Store* Store::Open(const Config& cfg_in, std::string* err) {
+  Config cfg = cfg_in;
+  cfg.max_rows = env_i64("KB_MAX_ROWS", cfg.max_rows);
+  cfg.heap_bytes = env_i64("KB_HEAP_BYTES", cfg.heap_bytes);
+  cfg.flush_rows = env_i64("KB_FLUSH_ROWS", cfg.flush_rows);
+  cfg.device = (int)env_i64("KB_DEVICE", cfg.device);
+  kbslab::Slab* slab = kbslab::Slab::Create(cfg.max_rows, cfg.heap_bytes,
+                                            cfg.device, err);
+  if (!slab) return nullptr;
+  Store* s = new Store();
+  s->cfg_ = cfg;
+  s->slab_ = slab;
+  s->ring_.init(cfg.watch_cache_size > 0 ? cfg.watch_cache_size : 200000);
+  return s;
+}
+
+Store::~Store() { delete slab_; }
+
+Status Store::validateKey(const Bytes& key) const {
+  if (key.size() > (size_t)KEYW) return KEYTOOLONG;  // DESIGN.md §4 round-1 limit
+  for (char c : key)
+    if ((uint8_t)c <= 0x24) return BADKEY;  // coder/normal.go:29-31 constraint
+  return OK;
+}
+
+// Range bounds additionally allow TRAILING 0x00 bytes: enc(k+"\x00"..., 0)
+// and the 96B zero-padded (k,0) bound exclude exactly the same rows, so the
+// etcd single-key idiom [k, k+"\0") keeps its meaning. Any other byte <= 0x24
+// inside a bound would order differently against '$'-terminated internal keys
+// (coder/normal.go:29-31) and is rejected loudly.
+static Status validateBound(const Bytes& b) {
+  if (b.size() > (size_t)KEYW) return KEYTOOLONG;
+  size_t end = b.size();
+  while (end > 0 && b[end - 1] == '\x00') end--;
+  for (size_t i = 0; i < end; ++i)
+    if ((uint8_t)b[i] <= 0x24) return BADKEY;
+  return OK;
+}
+
+uint64_t Store::GetCurrentRevision() { return committed_; }
+
+void Store::SetCurrentRevision(uint64_t rev) {
+  std::lock_guard<std::recursive_mutex> lk(mu_);
+  committed_ = rev;  // tso.Commit (tso.go:60-72)
+  if (dealt_ < rev) dealt_ = rev;
+}
+
+void Store::ClockAdvance(int64_t secs) { now_ += secs; }
+
+uint64_t Store::deal(uint64_t prevRevision, Status* st) {
+  // backend.go:190-206 over naiveTSO (tso.go:52-54)
+  uint64_t rev = ++dealt_;
+  if (prevRevision > 0 && rev < prevRevision) { *st = REV_DRIFT; return rev; }
+  *st = OK;
+  return rev;
+}
+
+uint64_t Store::mustDeal(uint64_t prevRevision) {  // txn.go:139-142
+  Status st;
+  return deal(prevRevision, &st);
+}
+
+void Store::putRow(const Bytes& key, uint64_t rev, const Bytes& val) {
+  MemRow r;
+  r.val = val;
+  r.tomb = (val == kTombstone);
+  memtable_[{key, rev}] = std::move(r);
+}
+
+void Store::putRevRow(const Bytes& key, uint64_t objrev, bool flag9) {
+  MemRow r;
+  r.flag9 = flag9;
+  r.objrev = objrev;
+  memtable_[{key, 0}] = std::move(r);
+  revIndex_[key] = RevEntry{objrev, flag9};
+  if (flag9) tombstoned_.insert(key); else tombstoned_.erase(key);
+  if (key.find(kEvents) != Bytes::npos) events_keys_.insert(key);
+}
+
+void Store::maybeFlush() {
+  if ((int64_t)memtable_.size() >= cfg_.flush_rows) {
+    std::string err;
+    (void)flushLocked(&err);  // errors surface on the next explicit call
+  }
+}
+
+bool Store::flushLocked(std::string* err) {
+  if (memtable_.empty()) return true;
+  auto t0 = std::chrono::steady_clock::now();
+  kbslab::DeltaRows d;
+  d.m = (int64_t)memtable_.size();
+  d.keys.resize((size_t)d.m * KEYW);
+  d.meta.reserve(d.m);
+  d.rev.reserve(d.m);
+  d.vo.reserve(d.m);
+  int64_t heap_base = slab_->heap_used();
+  int64_t i = 0;
+  for (auto& [ik, row] : memtable_) {
+    const Bytes& key = ik.first;
+    uint64_t rev = ik.second;
+    pad96(key, d.keys.data() + (size_t)i * KEYW);
+    bool ev = key.find(kEvents) != Bytes::npos;
+    if (rev == 0) {
+      uint32_t vlen = row.flag9 ? 9 : 8;
+      d.meta.push_back(kbslab::meta_make(false, row.flag9, ev, (uint32_t)key.size(), vlen));
+      d.rev.push_back(0);
+      d.vo.push_back(row.objrev);
+    } else {
+      d.meta.push_back(kbslab::meta_make(row.tomb, false, ev, (uint32_t)key.size(),
+                                         (uint32_t)row.val.size()));
+      d.rev.push_back(rev);
+      d.vo.push_back((uint64_t)(heap_base + (int64_t)d.heap.size()));
+      d.heap.insert(d.heap.end(), row.val.begin(), row.val.end());
+      d.heap.resize((d.heap.size() + 3) & ~3ull);  // 4B-aligned heap records
+    }
+    ++i;
+  }
+  if (!slab_->Merge(d, err)) return false;
+  memtable_.clear();
+  host_merge_s_ += std::chrono::duration<double>(std::chrono::steady_clock::now() - t0).count();
+  return true;
+}
+
+bool Store::Flush(std::string* err) {
+  std::lock_guard<std::recursive_mutex> lk(mu_);
+  return flushLocked(err);
+}
+
+// ---- events / watch -----------------------------------------------------
+
+void Store::notify(const Bytes& key, const Bytes& val, uint64_t revision,
+                   uint64_t prevRevision, bool valid, Event::Type type) {
+  // txn.go:267-293 + serial collector (backend.go:208-270): writes are
+  // serialized, so events commit in ascending revision order in place.
+  if (revision == 0) return;
+  committed_ = revision;
+  if (dealt_ < revision) dealt_ = revision;
+  if (!valid) return;
+  Event e;
+  e.type = type;
+  e.revision = revision;
+  e.kv_key = key;
+  if (type == Event::DELETE) { e.kv_value = val; e.kv_revision = prevRevision; }
+  else { e.kv_value = val; e.kv_revision = revision; }
+  ring_.Add(e);  // backend.go:263
+  event_log_.push_back(e);
+  pending_.push_back(std::move(e));
+  if (pending_.size() >= 300) pumpEvents();  // eventBatchSize (backend.go:41)
+}
+
+void Store::pumpEvents() {
+  if (pending_.empty()) return;
+  bool any_live = false;
+  for (auto& [id, w] : watchers_) if (!w.dropped) { any_live = true; break; }
+  if (!any_live) { pending_.clear(); return; }  // hub with no subs drops batches
+  // fan-out via the GPU ballot filter (DESIGN.md §3.3 k_watch_filter)
+  const int64_t kBatch = 512;
+  for (size_t b0 = 0; b0 < pending_.size(); b0 += kBatch) {
+    int64_t e = std::min((size_t)kBatch, pending_.size() - b0);
+    kbslab::WatchFilterBatch batch;
+    batch.e = e;
+    batch.ekeys.resize((size_t)e * KEYW);
+    batch.erevs.resize(e);
+    for (int64_t j = 0; j < e; ++j) {
+      const Event& ev = pending_[b0 + j];
+      pad96(ev.kv_key, batch.ekeys.data() + (size_t)j * KEYW);
+      batch.erevs[j] = ev.revision;
+    }
+    std::vector<uint64_t> bitmap;
+    int64_t W = 0;
+    std::string err;
+    if (!slab_->WatchFilter(batch, &bitmap, &W, &err)) { pending_.clear(); return; }
+    int64_t words = (e + 63) / 64;
+    for (auto& [id, w] : watchers_) {
+      if (w.dropped || w.slot >= W) continue;
+      for (int64_t c = 0; c < words; ++c) {
+        uint64_t bits = bitmap[(size_t)(w.slot * words + c)];
+        while (bits) {
+          int j = __builtin_ctzll(bits);
+          bits &= bits - 1;
+          w.queue.push_back(pending_[b0 + c * 64 + j]);
+        }
+      }
+      if (w.queue.size() > kWatchQueueCap) {
+        // drop slow consumer (watcherhub.go:84-94)
+        w.dropped = true;
+        w.queue.clear();
+        slab_->WatcherClear(w.slot);
+      }
+    }
+  }
+  pending_.clear();
+}
+
+int64_t Store::Watch(const Bytes& prefix, uint64_t revision, Status* st) {
+  // watch.go:37-99
+  std::lock_guard<std::recursive_mutex> lk(mu_);
+  pumpEvents();  // older events go only to older watchers
+  Watcher w;
+  w.prefix = prefix;
+  if (revision == 0) {
+    w.from_rev = 0;
+  } else {
+    // Ring.FindEvents (ring.go:84-118)
+    if (ring_.e == 0) {
+      if (revision > committed_) w.from_rev = revision;
+      else { *st = WATCH_EMPTY; return -1; }
+    } else {
+      const Event& newest = ring_.arr[(ring_.e - 1) % ring_.l];
+      const Event& oldest = ring_.arr[ring_.s % ring_.l];
+      if (revision > newest.revision) {
+        w.from_rev = revision;  // high
+      } else if (revision < oldest.revision) {
+        *st = WATCH_LOW;  // watch.go:79-84
+        return -1;
+      } else {
+        int64_t n = ring_.e - ring_.s, lo = 0, hi = n;
+        while (lo < hi) {
+          int64_t mid = lo + (hi - lo) / 2;
+          if (ring_.arr[(ring_.s + mid) % ring_.l].revision >= revision) hi = mid;
+          else lo = mid + 1;
+        }
+        std::vector<Event> catchup;
+        for (int64_t i = lo; i < n; ++i) {
+          const Event& ev = ring_.arr[(ring_.s + i) % ring_.l];
+          if (ev.kv_key.compare(0, prefix.size(), prefix) == 0) catchup.push_back(ev);
+        }
+        uint64_t lastRevision = revision;
+        if (!catchup.empty()) {
+          lastRevision = newest.revision + 1;  // watch.go:91-95
+          for (auto& ev : catchup) w.queue.push_back(ev);
+        }
+        w.from_rev = lastRevision;
+      }
+    }
+  }
+  int64_t slot;
+  if (!free_slots_.empty()) { slot = free_slots_.back(); free_slots_.pop_back(); }
+  else slot = next_slot_++;
+  w.slot = slot;
+  uint8_t p96[KEYW];
+  pad96(prefix, p96);
+  std::string err;
+  if (!slab_->WatcherSet(slot, p96, (uint32_t)prefix.size(), w.from_rev, &err)) {
+    *st = INTERNAL;
+    return -1;
+  }
+  int64_t wid = next_wid_++;
+  watchers_[wid] = std::move(w);
+  *st = OK;
+  return wid;
+}
+
+std::vector<Event> Store::WatchPoll(int64_t wid, Status* st) {
+  std::lock_guard<std::recursive_mutex> lk(mu_);
+  pumpEvents();
+  auto it = watchers_.find(wid);
+  if (it == watchers_.end()) { *st = WATCH_DROPPED; return {}; }
+  Watcher& w = it->second;
+  if (w.dropped) { *st = WATCH_DROPPED; watchers_.erase(it); return {}; }
+  std::vector<Event> out(w.queue.begin(), w.queue.end());
+  w.queue.clear();
+  *st = OK;
+  return out;
+}
+
+void Store::WatchCancel(int64_t wid) {
+  std::lock_guard<std::recursive_mutex> lk(mu_);
+  auto it = watchers_.find(wid);
+  if (it == watchers_.end()) return;
+  slab_->WatcherClear(it->second.slot);
+  free_slots_.push_back(it->second.slot);
+  watchers_.erase(it);
+}
+
+// ---- txn protocol (txn.go, creator/naive.go) ----------------------------
+
+Status Store::createInternal(const Bytes& key, const Bytes& value, uint64_t revision) {
+  // creator/naive.go:48-105 against revIndex (== live revision-row contents)
+  auto it = revIndex_.find(key);
+  if (it == revIndex_.end()) {
+    putRevRow(key, revision, false);
+    putRow(key, revision, value);
+    return OK;
+  }
+  uint64_t prevRevision = it->second.rev;
+  bool isTombstone = it->second.tomb;
+  if (isTombstone && prevRevision < revision) {  // naive.go:85-87
+    putRevRow(key, revision, false);
+    putRow(key, revision, value);
+    return OK;
+  }
+  return CAS_FAILED;
+}
+
+WriteResponse Store::Create(const Bytes& key, const Bytes& value, Status* st) {
+  // txn.go:33-77
+  std::lock_guard<std::recursive_mutex> lk(mu_);
+  WriteResponse resp;
+  Status v = validateKey(key);
+  if (v != OK) { *st = v; return resp; }
+  ops_create_++;
+  Status dst;
+  uint64_t revision = deal(0, &dst);
+  Status err = dst == OK ? createInternal(key, value, revision) : dst;
+  notify(key, value, revision, 0, err == OK, Event::CREATE);
+  maybeFlush();
+  if (err == CAS_FAILED) {
+    resp.header_revision = revision;
+    *st = OK;
+    return resp;
+  } else if (err != OK) { *st = err; return resp; }
+  resp.header_revision = revision;
+  resp.succeeded = true;
+  *st = OK;
+  return resp;
+}
+
+Status Store::get(const Bytes& key, uint64_t revision, Bytes* val, uint64_t* modRev) {
+  // range.go:82-121: largest object row (key,rev<=R); tombstone -> NOTFOUND
+  *modRev = 0;
+  uint64_t R = revision == 0 ? UINT64_MAX : revision;
+  // memtable candidates first: rows of key with 1 <= rev <= R (strictly newer
+  // than any slab row of the same key)
+  bool mem_hit = false;
+  Bytes mval;
+  uint64_t mrev = 0;
+  bool mtomb = false;
+  {
+    auto ub = memtable_.upper_bound({key, R});
+    if (ub != memtable_.begin()) {
+      auto c = std::prev(ub);
+      if (c->first.first == key && c->first.second >= 1) {
+        mem_hit = true;
+        mrev = c->first.second;
+        mval = c->second.val;
+        mtomb = c->second.tomb;
+      }
+    }
+  }
+  if (mem_hit) {
+    *modRev = mrev;
+    if (mtomb) return NOTFOUND;
+    *val = mval;
+    return OK;
+  }
+  // GPU point read (range.go:91-121 reverse-iter semantics)
+  DevGetQ q;
+  pad96(key, q.key);
+  q.read_rev = R;
+  std::vector<kbslab::GetResult> outs;
+  std::string err;
+  if (!slab_->GetBatch({q}, &outs, &err)) return INTERNAL;
+  if (!outs[0].found) return NOTFOUND;
+  *modRev = outs[0].rev;
+  if (outs[0].tomb) return NOTFOUND;
+  *val = outs[0].val;
+  return OK;
+}
+
+GetResponse Store::Get(const Bytes& key, uint64_t revision, Status* st) {
+  // range.go:34-74
+  std::lock_guard<std::recursive_mutex> lk(mu_);
+  GetResponse resp;
+  uint64_t curRev = committed_;
+  Bytes val;
+  uint64_t modRev = 0;
+  Status err = get(key, revision, &val, &modRev);
+  if (err == NOTFOUND) { resp.header_revision = curRev; *st = OK; return resp; }
+  if (err != OK) { *st = err; return resp; }
+  if (modRev > curRev) curRev = modRev;
+  resp.header_revision = curRev;
+  resp.has_kv = true;
+  resp.kv = KeyValue{key, val, modRev};
+  *st = OK;
+  return resp;
+}
+
+WriteResponse Store::Update(const Bytes& key, const Bytes& value,
+                            uint64_t prevRev, Status* st) {
+  // txn.go:193-265
+  std::lock_guard<std::recursive_mutex> lk(mu_);
+  WriteResponse resp;
+  Status v = validateKey(key);
+  if (v != OK) { *st = v; return resp; }
+  ops_update_++;
+  uint64_t curRev = 0;
+  Status err;
+  if (prevRev == 0) {
+    Status dst;
+    curRev = deal(0, &dst);
+    err = dst == OK ? createInternal(key, value, curRev) : dst;
+    notify(key, value, curRev, prevRev, err == OK, Event::CREATE);
+  } else {
+    Status dst;
+    uint64_t newRevision = deal(prevRev, &dst);
+    if (dst != OK) { curRev = 0; err = dst; }
+    else {
+      // CAS(revKey, new8, old8) (txn.go:255-264; memkv/batch.go:72-92): a
+      // 9B tombstone-flagged current value never equals the 8B expectation.
+      auto it = revIndex_.find(key);
+      if (it == revIndex_.end() || it->second.tomb || it->second.rev != prevRev) {
+        err = CAS_FAILED;
+      } else {
+        putRevRow(key, newRevision, false);
+        putRow(key, newRevision, value);
+        err = OK;
+      }
+      curRev = newRevision;
+    }
+    notify(key, value, curRev, prevRev, err == OK, Event::PUT);
+  }
+  maybeFlush();
+  resp.header_revision = curRev;
+  resp.succeeded = (err == OK);
+  if (err == CAS_FAILED) {
+    Bytes val;
+    uint64_t modRev = 0;
+    Status getErr = get(key, 0, &val, &modRev);
+    if (getErr != OK) {
+      if (getErr == NOTFOUND) { *st = OK; return resp; }
+      *st = getErr;
+      return resp;
+    }
+    resp.header_revision = std::max(resp.header_revision, modRev);
+    resp.has_kv = true;
+    resp.kv = KeyValue{key, val, modRev};
+    *st = OK;
+    return resp;
+  } else if (err != OK) { *st = err; return resp; }
+  *st = OK;
+  return resp;
+}
+
+WriteResponse Store::Delete(const Bytes& key, uint64_t prevRev, Status* st) {
+  // txn.go:79-190
+  std::lock_guard<std::recursive_mutex> lk(mu_);
+  WriteResponse resp;
+  Status v = validateKey(key);
+  if (v != OK) { *st = v; return resp; }
+  ops_delete_++;
+  uint64_t expectedRevision = prevRev;
+  Bytes oldVal;
+  uint64_t modRevision = 0;
+  Status err = get(key, 0, &oldVal, &modRevision);
+  if (err != OK) {
+    uint64_t rev = mustDeal(prevRev);  // txn.go:148-151
+    notify(key, Bytes(), rev, 0, false, Event::DELETE);
+    resp.header_revision = rev;
+    if (err == NOTFOUND) { *st = OK; return resp; }
+    *st = err;
+    return resp;
+  }
+  Status dst;
+  uint64_t newRevision = deal(prevRev, &dst);
+  if (dst != OK) { *st = dst; return resp; }
+  KeyValue old{key, oldVal, modRevision};
+  if (expectedRevision > 0 && expectedRevision != modRevision) {
+    err = CAS_FAILED;  // txn.go:162-166
+  } else {
+    if (expectedRevision == 0) expectedRevision = modRevision;
+    if (newRevision <= modRevision) err = INTERNAL;  // txn.go:171-175
+    else {
+      // CAS(revKey, rev8+0x00, expected8) + Put(objKey, tombstone) (txn.go:177-186)
+      auto it = revIndex_.find(key);
+      if (it == revIndex_.end() || it->second.tomb || it->second.rev != expectedRevision) {
+        err = CAS_FAILED;
+      } else {
+        putRevRow(key, newRevision, true);
+        putRow(key, newRevision, kTombstone);
+        err = OK;
+      }
+    }
+  }
+  notify(key, old.value, newRevision, old.revision, err == OK, Event::DELETE);
+  maybeFlush();
+  resp.header_revision = newRevision;
+  resp.succeeded = (err == OK);
+  if (err == CAS_FAILED) {
+    Bytes val;
+    uint64_t modRev = 0;
+    Status getErr = get(key, 0, &val, &modRev);
+    if (getErr != OK) {
+      resp.has_kv = true;
+      resp.kv = old;
+      *st = OK;
+      return resp;
+    }
+    resp.header_revision = std::max(resp.header_revision, modRev);
+    resp.has_kv = true;
+    resp.kv = KeyValue{key, val, modRev};
+    *st = OK;
+    return resp;
+  } else if (err != OK) { *st = err; return resp; }
+  resp.has_kv = true;
+  resp.kv = old;
+  *st = OK;
+  return resp;
+}
+
+// ---- range --------------------------------------------------------------
+
+Status Store::checkCompactRace(uint64_t revision) {
+  // scanner.go:594-626 (read path)
+  if (!compact_cell_set_) return OK;
+  if (compact_cell_ > revision) return COMPACTED;
+  return OK;
+}
+
+RangeResponse Store::List(const Bytes& start, const Bytes& end,
+                          uint64_t revision, int64_t limit, Status* st) {
+  // range.go:124-174 + scanner rangeWithLimit/worker.run on the GPU slab
+  std::lock_guard<std::recursive_mutex> lk(mu_);
+  RangeResponse resp;
+  if (end.empty()) { *st = INVALID_ARG; return resp; }
+  Status bv = validateBound(start);
+  if (bv == OK) bv = validateBound(end);
+  if (bv != OK) { *st = bv; return resp; }
+  uint64_t reqRevision = revision;
+  uint64_t curRevision = committed_;
+  if (reqRevision == 0) reqRevision = curRevision;
+  if (start >= end) { *st = INVALID_ARG; return resp; }
+  int64_t lim = limit > 0 ? limit + 1 : 0;  // range.go:154-158
+  Status cst = checkCompactRace(reqRevision);
+  if (cst != OK) { *st = cst; return resp; }
+  ops_range_++;
+
+  // memtable winners over [start, end): per key, max rev<=R (tombstones are
+  // deletion markers that override slab winners)
+  struct MemWinner { Bytes key, val; uint64_t rev; bool tomb; };
+  std::vector<MemWinner> memw;
+  int64_t mem_tombs = 0;
+  {
+    auto it = memtable_.lower_bound({start, 0});
+    auto endit = memtable_.lower_bound({end, 0});
+    const Bytes* curKey = nullptr;
+    for (; it != endit; ++it) {
+      uint64_t r = it->first.second;
+      if (r == 0 || r > reqRevision) continue;
+      if (!curKey || *curKey != it->first.first) {
+        memw.push_back(MemWinner{it->first.first, it->second.val, r, it->second.tomb});
+        curKey = &memw.back().key;
+      } else {  // ascending rev within key: later row wins
+        memw.back().val = it->second.val;
+        memw.back().rev = r;
+        memw.back().tomb = it->second.tomb;
+      }
+    }
+    for (auto& m : memw) if (m.tomb) mem_tombs++;
+  }
+
+  // GPU winners, fetched in chunks; tombstone overrides may consume up to
+  // mem_tombs extra, so over-ask. Continuation bound trick: winner_key+0x01
+  // excludes only the winner itself (key bytes <= 0x24 are rejected), so the
+  // next chunk resumes exactly after it.
+  std::string err;
+  std::vector<kbslab::RangeResult> outs;
+  std::vector<kbslab::RangeResult::Rec> dev_recs;
+  size_t di = 0, mi = 0;
+  bool dev_more = true;  // more device winners may exist past dev_recs
+  Bytes dev_frontier = start;
+  auto fetch_dev = [&](int64_t want) -> bool {
+    DevRangeQ q{};
+    memset(q.start, 0, KEYW);
+    memcpy(q.start, dev_frontier.data(), std::min(dev_frontier.size(), (size_t)KEYW));
+    pad96(end, q.end);
+    q.read_rev = reqRevision;
+    q.cap = want;
+    q.count_only = 0;
+    if (!slab_->RangeBatch({q}, true, &outs, &err)) return false;
+    kbslab::RangeResult& r = outs[0];
+    if (r.overflow) return false;
+    for (auto& rec : r.recs) dev_recs.push_back(std::move(rec));
+    // more winners beyond this chunk iff the scan stopped at its cap
+    dev_more = (want > 0 && r.total >= want) ||
+               (want <= 0 && r.total > r.written);
+    if (!dev_recs.empty())
+      dev_frontier = dev_recs.back().key + Bytes("\x01", 1);
+    return true;
+  };
+  if (!fetch_dev(lim > 0 ? lim + mem_tombs : 0)) {
+    *st = outs.empty() || !outs[0].overflow ? INTERNAL : NOBUF;
+    return resp;
+  }
+
+  // ordered merge: device winners × memtable winners (override semantics)
+  std::vector<KeyValue> kvs;
+  auto need = [&]() { return lim <= 0 || (int64_t)kvs.size() < lim; };
+  while (need()) {
+    if (di >= dev_recs.size() && dev_more) {
+      int64_t want = lim > 0 ? (lim - (int64_t)kvs.size()) + mem_tombs : 0;
+      size_t before = dev_recs.size();
+      if (!fetch_dev(want)) { *st = INTERNAL; return resp; }
+      if (dev_recs.size() == before && !dev_more) {}  // exhausted
+      if (dev_recs.size() == before && dev_more) break;  // defensive
+    }
+    bool have_d = di < dev_recs.size();
+    bool have_m = mi < memw.size();
+    if (!have_d && !have_m) break;
+    if (!have_d && dev_more) continue;  // fetch more before consuming memw tail
+    if (have_d && (!have_m || dev_recs[di].key < memw[mi].key)) {
+      kvs.push_back(KeyValue{dev_recs[di].key, dev_recs[di].val, dev_recs[di].rev});
+      di++;
+    } else if (have_m && (!have_d || memw[mi].key < dev_recs[di].key)) {
+      if (!memw[mi].tomb)
+        kvs.push_back(KeyValue{memw[mi].key, memw[mi].val, memw[mi].rev});
+      mi++;
+    } else {  // same key: memtable rows are newer -> override
+      if (!memw[mi].tomb)
+        kvs.push_back(KeyValue{memw[mi].key, memw[mi].val, memw[mi].rev});
+      di++;
+      mi++;
+    }
+  }
+
+  resp.header_revision = curRevision;
+  if (lim > 0 && (int64_t)kvs.size() > limit) {  // range.go:168-171
+    resp.more = true;
+    kvs.resize(limit);
+  }
+  resp.kvs = std::move(kvs);
+  *st = OK;
+  return resp;
+}
+
+CountResponse Store::Count(const Bytes& start, const Bytes& end, Status* st) {
+  // range.go:177-205
+  std::lock_guard<std::recursive_mutex> lk(mu_);
+  CountResponse resp;
+  uint64_t rev = committed_;
+  resp.header_revision = rev;
+  if (!cfg_.enable_etcd_compatibility) { *st = OK; return resp; }
+  Status bv = validateBound(start);
+  if (bv == OK) bv = validateBound(end);
+  if (bv != OK) { *st = bv; return resp; }
+  Status cst = checkCompactRace(rev);
+  if (cst != OK) { *st = cst; return resp; }
+  std::string err;
+  if (!flushLocked(&err)) { *st = INTERNAL; return resp; }
+  DevRangeQ q{};
+  pad96(start, q.start);
+  pad96(end, q.end);
+  q.read_rev = rev;
+  q.cap = 0;
+  q.count_only = 1;
+  std::vector<kbslab::RangeResult> outs;
+  if (!slab_->RangeBatch({q}, false, &outs, &err)) { *st = INTERNAL; return resp; }
+  resp.count = (uint64_t)outs[0].total;
+  *st = OK;
+  return resp;
+}
+
+// ---- compaction ---------------------------------------------------------
+
+uint64_t Store::getTimeoutRevision() {
+  // scanner.go:147-177
+  CompactRecord prev{0, 0};
+  while (!compact_histories_.empty()) {
+    CompactRecord head = compact_histories_.front();
+    if (now_ - head.time < cfg_.events_ttl_seconds) break;
+    compact_histories_.pop_front();
+    prev = head;
+  }
+  return prev.revision;
+}
+
+uint64_t Store::Compact(uint64_t revision, Status* st) {
+  // compact.go:31-127 (retry-queue MinRevision stub == 0)
+  std::lock_guard<std::recursive_mutex> lk(mu_);
+  uint64_t curRevision = committed_;
+  if (revision == 0 || revision > curRevision) revision = curRevision;
+  // setCompactRecord (compact.go:70-105): a larger stored record skips the
+  // cell update but NOT the border scans
+  if (!(compact_cell_set_ && compact_cell_ > revision)) {
+    compact_cell_set_ = true;
+    compact_cell_ = revision;
+  }
+  std::string err;
+  if (!flushLocked(&err)) { *st = INTERNAL; return revision; }
+  pumpEvents();
+  // borders (compact.go:108-127)
+  std::vector<Bytes> prefixes;
+  prefixes.push_back(cfg_.prefix);
+  for (auto& p : cfg_.skipped_prefixes) prefixes.push_back(p);
+  std::vector<Bytes> borders;
+  for (auto key : prefixes) {
+    if (key.empty() || key.back() != '/') key += '/';
+    borders.push_back(key);
+    borders.push_back(PrefixEnd(key));
+  }
+  std::sort(borders.begin(), borders.end());
+  std::vector<std::pair<kbslab::Slab::Bound, kbslab::Slab::Bound>> bpairs;
+  uint64_t timeout_rev = 0;
+  for (size_t i = 0; i + 1 < borders.size(); i += 2) {
+    // scanner.Compact: logCompactHistory + checkCompactRace(compact) Put +
+    // per-scan timeout revision (scanner.go:147-198, 594-603)
+    compact_histories_.push_back(CompactRecord{revision, now_});
+    compact_cell_set_ = true;
+    compact_cell_ = revision;
+    timeout_rev = getTimeoutRevision();  // identical across pairs in one call
+    kbslab::Slab::Bound lo{}, hi{};
+    pad96(borders[i], lo.key);
+    lo.rev = 0;
+    pad96(borders[i + 1], hi.key);
+    hi.rev = 0;
+    bpairs.push_back({lo, hi});
+  }
+  if (!slab_->Compact(bpairs, revision, timeout_rev, &err)) {
+    *st = INTERNAL;
+    return revision;
+  }
+  // host mirror of deleted revision rows (flagged <= compactRev; TTL'd events)
+  for (auto it = tombstoned_.begin(); it != tombstoned_.end();) {
+    auto ri = revIndex_.find(*it);
+    if (ri != revIndex_.end() && ri->second.tomb && ri->second.rev <= revision) {
+      revIndex_.erase(ri);
+      it = tombstoned_.erase(it);
+    } else ++it;
+  }
+  if (timeout_rev != 0) {
+    for (auto it = events_keys_.begin(); it != events_keys_.end();) {
+      auto ri = revIndex_.find(*it);
+      if (ri != revIndex_.end() && ri->second.rev <= timeout_rev) {
+        tombstoned_.erase(*it);
+        revIndex_.erase(ri);
+        it = events_keys_.erase(it);
+      } else ++it;
+    }
+  }
+  *st = OK;
+  return revision;
+}
+
+// ---- dump / parity ------------------------------------------------------
+
+bool Store::DumpStore(std::vector<std::pair<Bytes, Bytes>>* out, std::string* err) {
+  std::lock_guard<std::recursive_mutex> lk(mu_);
+  if (!flushLocked(err)) return false;
+  std::vector<kbslab::DumpRow> rows;
+  if (!slab_->Dump(&rows, err)) return false;
+  out->clear();
+  out->reserve(rows.size() + 1);
+  for (auto& r : rows) {
+    Bytes ik = EncodeObjectKey(r.key, r.rev);
+    Bytes val;
+    if (r.rev == 0) {
+      val = U64ToBytes(r.vo);
+      if (r.meta & M_FLAG9) val += '\x00';
+    } else {
+      val = r.val;
+    }
+    out->push_back({std::move(ik), std::move(val)});
+  }
+  if (compact_cell_set_) {
+    Bytes ck = cfg_.prefix + "/compact_key";
+    Bytes cv = U64ToBytes(compact_cell_);
+    auto pos = std::lower_bound(out->begin(), out->end(), ck,
+                                [](const std::pair<Bytes, Bytes>& a, const Bytes& b) {
+                                  return a.first < b;
+                                });
+    out->insert(pos, {ck, cv});
+  }
+  return true;
+}
+
+// ---- bench --------------------------------------------------------------
+
+bool Store::BulkCreate(const uint8_t* keys, const uint32_t* klens,
+                       const uint8_t* vals, const uint32_t* vlens, size_t n,
+                       std::string* err) {
+  std::lock_guard<std::recursive_mutex> lk(mu_);
+  const uint8_t* kp = keys;
+  const uint8_t* vp = vals;
+  for (size_t i = 0; i < n; ++i) {
+    Bytes key((const char*)kp, klens[i]);
+    Bytes val((const char*)vp, vlens[i]);
+    kp += klens[i];
+    vp += vlens[i];
+    Status st;
+    auto r = Create(key, val, &st);
+    if (st != OK || !r.succeeded) {
+      if (err) *err = "bulk create failed at " + std::to_string(i) + " key " + key;
+      return false;
+    }
+  }
+  return true;
+}
+
+bool Store::BenchRange(const uint8_t* qbuf, size_t nq, bool d2h,
+                       unsigned long long* total, double* secs, std::string* err) {
+  // the measured hot path: batched List semantics with inputs resident in HBM
+  std::lock_guard<std::recursive_mutex> lk(mu_);
+  if (!flushLocked(err)) return false;
+  struct Q { Bytes s, e; uint64_t rev; int64_t limit; };
+  std::vector<Q> qs;
+  qs.reserve(nq);
+  const uint8_t* p = qbuf;
+  for (size_t i = 0; i < nq; ++i) {
+    uint32_t slen, elen;
+    uint64_t rev, limit;
+    memcpy(&slen, p, 4); p += 4;
+    memcpy(&elen, p, 4); p += 4;
+    memcpy(&rev, p, 8); p += 8;
+    memcpy(&limit, p, 8); p += 8;
+    Q q;
+    q.s.assign((const char*)p, slen); p += slen;
+    q.e.assign((const char*)p, elen); p += elen;
+    q.rev = rev;
+    q.limit = (int64_t)limit;
+    qs.push_back(std::move(q));
+  }
+  const int64_t kMax = 1024;
+  unsigned long long tot = 0;
+  auto t0 = std::chrono::steady_clock::now();
+  for (size_t b0 = 0; b0 < qs.size(); b0 += kMax) {
+    size_t bn = std::min((size_t)kMax, qs.size() - b0);
+    std::vector<DevRangeQ> dq(bn);
+    for (size_t j = 0; j < bn; ++j) {
+      DevRangeQ& q = dq[j];
+      memset(&q, 0, sizeof(q));
+      pad96(qs[b0 + j].s, q.start);
+      pad96(qs[b0 + j].e, q.end);
+      q.read_rev = qs[b0 + j].rev == 0 ? committed_ : qs[b0 + j].rev;
+      q.cap = qs[b0 + j].limit > 0 ? qs[b0 + j].limit + 1 : 0;
+      q.count_only = 0;
+    }
+    std::vector<kbslab::RangeResult> outs;
+    if (!slab_->RangeBatch(dq, d2h, &outs, err)) return false;
+    for (size_t j = 0; j < bn; ++j) {
+      int64_t lim = qs[b0 + j].limit;
+      int64_t w = outs[j].written;
+      tot += (unsigned long long)(lim > 0 && w > lim ? lim : w);
+    }
+  }
+  *secs = std::chrono::duration<double>(std::chrono::steady_clock::now() - t0).count();
+  *total = tot;
+  ops_range_ += (int64_t)nq;
+  return true;
+}
+
+std::string Store::PerfJson() {
+  const kbslab::Perf& p = slab_->perf;
+  char buf[1536];
+  snprintf(buf, sizeof(buf),
+           "{\"scan_ms\":%.3f,\"gather_ms\":%.3f,\"get_ms\":%.3f,"
+           "\"compact_ms\":%.3f,\"merge_ms\":%.3f,\"filter_ms\":%.3f,"
+           "\"pack_d2h_ms\":%.3f,\"scan_launches\":%lld,\"rows_scanned\":%lld,"
+           "\"bytes_gathered\":%lld,\"winners\":%lld,\"merges\":%lld,"
+           "\"compacts\":%lld,\"filter_launches\":%lld,\"filter_events\":%lld,"
+           "\"filter_watchers\":%lld,\"slab_rows\":%lld,\"heap_used\":%lld,"
+           "\"ops\":{\"create\":%lld,\"update\":%lld,\"delete\":%lld,"
+           "\"range\":%lld}}",
+           p.scan_ms, p.gather_ms, p.get_ms, p.compact_ms, p.merge_ms,
+           p.filter_ms, p.pack_d2h_ms, (long long)p.scan_launches,
+           (long long)p.rows_scanned, (long long)p.bytes_gathered,
+           (long long)p.winners, (long long)p.merges, (long long)p.compacts,
+           (long long)p.filter_launches, (long long)p.filter_events,
+           (long long)p.filter_watchers, (long long)slab_->rows(),
+           (long long)slab_->heap_used(), (long long)ops_create_,
+           (long long)ops_update_, (long long)ops_delete_, (long long)ops_range_);
+  return buf;
+}
+
+void Store::PerfReset() {
+  slab_->perf = kbslab::Perf();
+  ops_create_ = ops_update_ = ops_delete_ = ops_range_ = 0;
+}
+
+}  // namespace kbstore

@@ -1,0 +1,167 @@
+// kubebrain_amd/csrc/store.h — host side of the MI355X-native MVCC store.
+// Mirrors the reference's backend.Backend surface (pkg/backend/backend.go:44-84)
+// over the HBM slab (slab_dev.h). PRODUCT code — never touches oracle/.
+//
+// Write path: single-writer; revIndex (revision-row contents,
+// the CAS target) + memtable (rows newer than the slab) on the host; reads
+// merge GPU winners with the bounded memtable.
+#pragma once
+
+#include <cstdint>
+#include <deque>
+#include <map>
+#include <memory>
+#include <mutex>
+#include <set>
+#include <string>
+#include <unordered_map>
+#include <unordered_set>
+#include <vector>
+
+#include "slab_dev.h"
+
+namespace kbstore {
+
+using Bytes = std::string;
+
+enum Status : int32_t {  // == include/kb_slab.h kb_status
+  OK = 0, NOTFOUND = 1, CAS_FAILED = 2, UNCERTAIN = 3, COMPACTED = 4,
+  INVALID_ARG = 5, UNSUPPORTED = 6, REV_DRIFT = 7, WATCH_LOW = 8,
+  WATCH_EMPTY = 9, WATCH_DROPPED = 10, KEYTOOLONG = 11, BADKEY = 12,
+  INTERNAL = 13, NOGPU = 14, NOBUF = 100,
+};
+
+struct Event {
+  enum Type : int32_t { CREATE = 0, PUT = 1, DELETE = 2 };
+  Type type;
+  uint64_t revision;
+  Bytes kv_key, kv_value;
+  uint64_t kv_revision;
+};
+
+struct KeyValue { Bytes key, value; uint64_t revision = 0; };
+struct GetResponse { uint64_t header_revision = 0; bool has_kv = false; KeyValue kv; };
+struct RangeResponse { uint64_t header_revision = 0; std::vector<KeyValue> kvs; bool more = false; };
+struct WriteResponse { uint64_t header_revision = 0; bool succeeded = false; bool has_kv = false; KeyValue kv; };
+struct CountResponse { uint64_t header_revision = 0; uint64_t count = 0; };
+
+class Store {
+ public:
+  struct Config {
+    Bytes prefix = "/registry";
+    std::vector<Bytes> skipped_prefixes;
+    int watch_cache_size = 200000;      // historyCapacity (backend.go:39)
+    bool enable_etcd_compatibility = true;
+    int64_t events_ttl_seconds = 3600;  // eventsTTL (util.go:37)
+    int64_t max_rows = 8 << 20;
+    int64_t heap_bytes = 2ll << 30;
+    int64_t flush_rows = 65536;
+    int device = -1;
+  };
+
+  static Store* Open(const Config& cfg, std::string* err);
+  ~Store();
+
+  WriteResponse Create(const Bytes& key, const Bytes& value, Status* st);
+  WriteResponse Update(const Bytes& key, const Bytes& value, uint64_t prevRev, Status* st);
+  WriteResponse Delete(const Bytes& key, uint64_t prevRev, Status* st);
+  GetResponse Get(const Bytes& key, uint64_t revision, Status* st);
+  RangeResponse List(const Bytes& start, const Bytes& end, uint64_t revision,
+                     int64_t limit, Status* st);
+  CountResponse Count(const Bytes& start, const Bytes& end, Status* st);
+  uint64_t Compact(uint64_t revision, Status* st);
+  uint64_t GetCurrentRevision();
+  void SetCurrentRevision(uint64_t rev);
+
+  int64_t Watch(const Bytes& prefix, uint64_t revision, Status* st);
+  std::vector<Event> WatchPoll(int64_t wid, Status* st);
+  void WatchCancel(int64_t wid);
+
+  void ClockAdvance(int64_t secs);
+  bool Flush(std::string* err);
+  // sorted (internal key, value) pairs, byte-diffable vs the oracle dump
+  bool DumpStore(std::vector<std::pair<Bytes, Bytes>>* out, std::string* err);
+  const std::vector<Event>& EventLog() const { return event_log_; }
+
+  // bench support
+  bool BulkCreate(const uint8_t* keys, const uint32_t* klens, const uint8_t* vals,
+                  const uint32_t* vlens, size_t n, std::string* err);
+  bool BenchRange(const uint8_t* qbuf, size_t nq, bool d2h,
+                  unsigned long long* total, double* secs, std::string* err);
+  std::string PerfJson();
+  void PerfReset();
+
+ private:
+  Store() = default;
+  // helpers (semantics cites in store.cc)
+  Status validateKey(const Bytes& key) const;
+  uint64_t deal(uint64_t prevRevision, Status* st);
+  uint64_t mustDeal(uint64_t prevRevision);
+  Status createInternal(const Bytes& key, const Bytes& value, uint64_t revision);
+  void notify(const Bytes& key, const Bytes& val, uint64_t revision,
+              uint64_t prevRevision, bool valid, Event::Type type);
+  Status get(const Bytes& key, uint64_t revision, Bytes* val, uint64_t* modRev);
+  Status checkCompactRace(uint64_t revision);
+  void putRow(const Bytes& key, uint64_t rev, const Bytes& val);
+  void putRevRow(const Bytes& key, uint64_t objrev, bool flag9);
+  void maybeFlush();
+  void pumpEvents();  // fan-out pending events via the GPU filter
+  bool flushLocked(std::string* err);
+  uint64_t getTimeoutRevision();
+
+  Config cfg_;
+  std::recursive_mutex mu_;
+  kbslab::Slab* slab_ = nullptr;
+  uint64_t committed_ = 0, dealt_ = 0;
+
+  struct RevEntry { uint64_t rev; bool tomb; };
+  std::unordered_map<Bytes, RevEntry> revIndex_;
+  std::unordered_set<Bytes> tombstoned_;  // keys with flagged rev-rows
+  std::unordered_set<Bytes> events_keys_; // keys containing "/events/"
+
+  struct MemRow { Bytes val; bool tomb = false; bool flag9 = false; uint64_t objrev = 0; };
+  std::map<std::pair<Bytes, uint64_t>, MemRow> memtable_;  // (userKey, rev)
+
+  bool compact_cell_set_ = false;
+  uint64_t compact_cell_ = 0;
+
+  int64_t now_ = 0;
+  struct CompactRecord { uint64_t revision; int64_t time; };
+  std::deque<CompactRecord> compact_histories_;
+
+  // watch (ring for catch-up, GPU filter for fan-out)
+  struct Ring {
+    int64_t s = 0, e = 0;
+    int l = 0;
+    std::vector<Event> arr;
+    void init(int cap) { l = cap; arr.resize(cap); }
+    void Add(const Event& ev) {
+      arr[e % l] = ev;
+      if (e == s + (int64_t)l) s++;
+      e++;
+    }
+  } ring_;
+  std::vector<Event> event_log_;
+  std::vector<Event> pending_;  // not yet fanned out
+  struct Watcher {
+    int64_t slot;
+    Bytes prefix;
+    uint64_t from_rev;
+    std::deque<Event> queue;
+    bool dropped = false;
+  };
+  std::unordered_map<int64_t, Watcher> watchers_;
+  std::vector<int64_t> free_slots_;
+  int64_t next_slot_ = 0, next_wid_ = 1;
+  static constexpr size_t kWatchQueueCap = 10000ull * 300;  // watcherhub.go:30 × eventBatchSize
+
+  // host-side perf
+  double host_write_s_ = 0, host_merge_s_ = 0;
+  int64_t ops_create_ = 0, ops_update_ = 0, ops_delete_ = 0, ops_range_ = 0;
+};
+
+Bytes EncodeObjectKey(const Bytes& userKey, uint64_t rev);  // coder/normal.go:42-50
+Bytes PrefixEnd(const Bytes& prefix);                       // util.go
+Bytes U64ToBytes(uint64_t v);
+
+}  // namespace kbstore

@@ -1,0 +1,247 @@
+"""GPU parity tests: byte-for-byte agreement between the HIP slab store and
+the CPU oracle on the full hot path (Range/Get/Count/Txn/Compact/Watch).
+All marked gpu — run on a real MI355X via gpurun."""
+import os
+import random
+import sys
+
+import pytest
+
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
+
+import parity
+
+pytestmark = pytest.mark.gpu
+
+parity.small_env()
+
+
+@pytest.fixture()
+def dual():
+    d = parity.Dual()
+    yield d
+    d.close()
+
+
+NS = [b"/registry/pods/ns-%02d" % i for i in range(8)] + \
+     [b"/registry/configmaps/ns-%02d" % i for i in range(4)] + \
+     [b"/registry/events/ns-%02d" % i for i in range(2)]
+
+
+def keyname(ns, i):
+    return ns + b"/obj-%05d" % i
+
+
+def test_golden_scenarios_on_gpu(dual):
+    """The reference's own backend_test.go tables, replayed on the GPU store
+    (same tables as tests/test_golden_oracle.py, now product vs oracle)."""
+    PFX = b"/registry/test"
+    KEY = PFX + b"/testKey"
+    VAL = b"testValue"
+    w = dual.watch(PFX + b"/", 0)
+    # create / create-twice (backend_test.go:597-630)
+    r = dual.create(KEY, VAL)
+    assert r.succeeded
+    r = dual.create(KEY, VAL + b"/2")
+    assert not r.succeeded
+    dual.poll(w)
+    # update table (backend_test.go:684-738)
+    r = dual.update(KEY, VAL, 0)          # exists -> CAS fail, returns latest
+    r = dual.update(KEY, VAL + b"3", r.kv.revision if r.kv else 0)
+    dual.poll(w)
+    # delete (backend_test.go:632-682)
+    dual.delete(PFX + b"/not/found", 0)
+    dual.delete(KEY, 0)
+    dual.poll(w)
+    # recreate over tombstone (backend_test.go:1134-1178)
+    dual.create(KEY, b"val2")
+    dual.get(KEY, 0)
+    dual.poll(w)
+    dual.diff_dump()
+    dual.diff_event_log()
+
+
+def test_range_table_on_gpu(dual):
+    # backend_test.go:740-877 range/get/count table on the GPU slab
+    KEY = b"/registry/test/testKey"
+    end_key = KEY[:-1] + bytes([KEY[-1] + 1])
+    inject = 10
+    invalid_rev = dual.p.current_rev()
+    for i in range(inject):
+        dual.create(KEY + b"/%05d" % i, b"testValue/%05d" % i)
+    init = dual.p.current_rev()
+    dual.get(KEY + b"/%05d" % (inject - 1), 0)
+    dual.get(KEY + b"/%05d" % (inject - 2), init)
+    dual.get(KEY + b"/%05d" % (inject - 1), invalid_rev)
+    dual.get(KEY + b"/none", 0)
+    dual.list(KEY, end_key, 0, 0)
+    dual.list(KEY, KEY + b"/%05d" % (inject - 2), 0, 0)
+    r = dual.list(KEY, KEY + b"/%05d" % (inject - 2), 0, inject - 4)
+    assert r.more
+    dual.list(end_key, end_key + b"/x", 0, 0)
+    dual.list(KEY + b"/%05d" % 1, KEY + b"/%05d" % (inject - 1), init - 2, inject - 5)
+    dual.list(KEY, end_key, 0, inject - 5)
+    dual.count(KEY, end_key)
+    dual.count(end_key, end_key[:-1] + bytes([end_key[-1] + 1]))
+
+
+def test_limits_and_boundaries(dual):
+    ns = NS[0]
+    for i in range(50):
+        dual.create(keyname(ns, i), b"v%d" % i)
+    end = parity.kbclient  # noqa
+    lo, hi = ns + b"/", ns + b"0"
+    for limit in (0, 1, 2, 49, 50, 51, 500):
+        dual.list(lo, hi, 0, limit)
+    # empty ranges, range at old revisions
+    dual.list(b"/registry/zzz/", b"/registry/zzz0", 0, 10)
+    cur = dual.p.current_rev()
+    for rev in (cur - 1, cur - 25, cur - 49):
+        dual.list(lo, hi, rev, 7)
+    # single-key range
+    dual.list(keyname(ns, 3), keyname(ns, 4), 0, 5)
+    # invalid args
+    dual.list(lo, b"", 0, 0)
+    dual.list(hi, lo, 0, 0)
+
+
+def test_key_validation(dual):
+    import kbclient
+    # key too long (KB_EKEYTOOLONG)
+    long_key = b"/registry/" + b"x" * 96
+    r = dual.p.create(long_key, b"v")
+    assert r.status == kbclient.KEYTOOLONG
+    # key bytes <= '$' (KB_EBADKEY)
+    r = dual.p.create(b"/registry/a\x01b", b"v")
+    assert r.status == kbclient.BADKEY
+    # max allowed length (96B total)
+    k96 = b"/registry/" + b"y" * 86
+    assert len(k96) == 96
+    dual.create(k96, b"v96")
+    dual.get(k96, 0)
+    dual.list(b"/registry/", b"/registry0", 0, 0)
+
+
+def test_mvcc_versions_and_old_revisions(dual):
+    ns = NS[1]
+    revs = {}
+    for i in range(20):
+        r = dual.create(keyname(ns, i), b"v0")
+        revs[i] = r.header_revision
+    for round_ in range(3):
+        for i in range(0, 20, 2):
+            r = dual.update(keyname(ns, i), b"v%d" % (round_ + 1), revs[i])
+            if r.succeeded:
+                revs[i] = r.header_revision
+    for i in (1, 2, 3):
+        dual.delete(keyname(ns, i), 0)
+    # reads across the whole revision history
+    cur = dual.p.current_rev()
+    for rev in range(cur - 90, cur + 2, 7):
+        dual.list(ns + b"/", ns + b"0", max(rev, 0), 9)
+        dual.get(keyname(ns, 2), max(rev, 0))
+    dual.diff_dump()
+
+
+def test_compaction_parity(dual):
+    ns = NS[2]
+    revs = {}
+    for i in range(30):
+        r = dual.create(keyname(ns, i), b"v0")
+        revs[i] = r.header_revision
+    for i in range(0, 30, 3):
+        r = dual.update(keyname(ns, i), b"v1", revs[i])
+        revs[i] = r.header_revision
+    for i in (0, 6, 12):
+        dual.delete(keyname(ns, i), 0)
+    mid = dual.p.current_rev()
+    for i in range(0, 30, 5):
+        dual.update(keyname(ns, (i + 1) % 30), b"v2", 0)  # CAS fails mostly
+    dual.compact(mid)
+    dual.diff_dump()
+    dual.list(ns + b"/", ns + b"0", 0, 0)
+    # range below the compact revision must fail identically (COMPACTED)
+    dual.list(ns + b"/", ns + b"0", max(mid - 1, 1), 5)
+    # compact to latest
+    dual.compact(0)
+    dual.diff_dump()
+    dual.count(ns + b"/", ns + b"0")
+
+
+def test_ttl_events_expiry(parity_env=None):
+    d = parity.Dual(events_ttl=1)
+    try:
+        pfx = b"/registry/events/ns-00"
+        for i in range(5):
+            d.create(pfx + b"/ev-%d" % i, b"e%d" % i)
+        d.compact(0)
+        d.list(pfx + b"/", pfx + b"0", 0, 0)
+        wr = d.create(pfx + b"/ev-0", b"again")  # consume a revision
+        d.clock_advance(2)
+        d.compact(0)
+        d.list(pfx + b"/", pfx + b"0", 0, 0)
+        d.diff_dump()
+    finally:
+        d.close()
+
+
+def test_watch_parity(dual):
+    pfx = b"/registry/pods/ns-00"
+    w_all = dual.watch(b"/registry/", 0)
+    w_ns = dual.watch(pfx, 0)
+    for i in range(10):
+        dual.create(keyname(pfx, i), b"w%d" % i)
+    dual.poll_all()
+    # catch-up from a historical revision
+    mid = dual.p.current_rev()
+    for i in range(10, 20):
+        dual.create(keyname(pfx, i), b"w%d" % i)
+    w_hist = dual.watch(pfx, mid - 3)
+    dual.poll(w_hist)
+    # watch below the ring -> identical error
+    dual.watch(pfx, 1)
+    # deletes produce DELETE events with prev value/revision
+    dual.delete(keyname(pfx, 0), 0)
+    dual.delete(keyname(pfx, 1), 0)
+    dual.poll_all()
+    dual.diff_event_log()
+
+
+def test_random_mixed_workload(dual):
+    rng = random.Random(0x6B62)
+    live = {}
+    w = dual.watch(b"/registry/", 0)
+    for step in range(1200):
+        op = rng.random()
+        ns = rng.choice(NS)
+        i = rng.randrange(60)
+        key = keyname(ns, i)
+        if op < 0.35:
+            r = dual.create(key, b"c%d" % step)
+            if r.succeeded:
+                live[key] = r.header_revision
+        elif op < 0.6:
+            prev = live.get(key, 0) if rng.random() < 0.7 else rng.randrange(1, 3000)
+            r = dual.update(key, b"u%d" % step, prev)
+            if r.succeeded:
+                live[key] = r.header_revision
+        elif op < 0.7:
+            r = dual.delete(key, live.get(key, 0) if rng.random() < 0.5 else 0)
+            if r.succeeded:
+                live.pop(key, None)
+        elif op < 0.85:
+            lo = ns + b"/"
+            hi = ns + b"0"
+            rev = 0 if rng.random() < 0.5 else max(1, dual.p.current_rev() - rng.randrange(100))
+            dual.list(lo, hi, rev, rng.choice([0, 1, 5, 17, 60]))
+        elif op < 0.95:
+            dual.get(key, 0 if rng.random() < 0.5 else max(1, dual.p.current_rev() - rng.randrange(50)))
+        else:
+            dual.count(ns + b"/", ns + b"0")
+        if step in (400, 900):
+            dual.compact(max(1, dual.p.current_rev() - 50))
+        if step % 300 == 299:
+            dual.poll(w)
+    dual.poll(w)
+    dual.diff_dump()
+    dual.diff_event_log()
