@@ -1,0 +1,396 @@
+#!/usr/bin/env python3
+"""bench.py — the measured hot path (BASELINE.json configs[1] at N=1).
+
+Workload: 1M Pod-style keys / 200k extra live revisions (zipf 1.1) / 2%
+tombstones on one MI355X; one step = 1000 ops = 900 batched Range(limit=500)
+over random namespace prefixes + 100 Txn conditional updates (90/10 mix,
+docs/benchmark.md-style 512B values, 300-client-style batching). `value` is
+whole-job ops/s with the slab resident in HBM and Range results landing in
+the device output arena; the PCIe-inclusive rate is reported separately as
+`ops_per_sec_with_d2h`. A separate watch leg measures GPU
+fan-out deliveries/s and is reported as `watch_events_per_sec`.
+
+Multi-GPU (--gpus N via torch.distributed.run): keys shard by namespace hash,
+one store per GPU; the query stream routes by namespace; weak scaling
+(per-GPU keyspace fixed as N grows). cpu_baseline: the oracle (kind "port")
+on the host cores, bounded sample, rank 0 / N=1 only.
+"""
+import argparse
+import ctypes
+import json
+import os
+import struct
+import sys
+import time
+
+REPO = os.path.dirname(os.path.abspath(__file__))
+sys.path.insert(0, REPO)
+sys.path.insert(0, os.path.join(REPO, "tests"))
+
+import numpy as np
+
+SEED = 0x6B62
+VAL_LEN = 512
+LIMIT = 500
+OPS_PER_STEP = 1000
+RANGE_SHARE = 0.9
+
+
+def log(rank, *a):
+    if rank == 0:
+        print(*a, file=sys.stderr, flush=True)
+
+
+def make_keys(nns, per_ns):
+    namespaces = [b"/registry/pods/ns-%04d" % i for i in range(nns)]
+    keys = []
+    for ns in namespaces:
+        for j in range(per_ns):
+            keys.append(ns + b"/pod-%06d" % j)
+    return namespaces, keys
+
+
+def build_store(store, namespaces, keys, rng, extra_revs, tomb_frac, my_ns=None):
+    """Load the synthetic keyspace through the normal write path."""
+    if my_ns is not None:
+        sel = [k for k in keys if k.rsplit(b"/", 1)[0] in my_ns]
+    else:
+        sel = keys
+    n = len(sel)
+    # values: seeded random bytes
+    vals = rng.integers(0, 256, size=n * VAL_LEN, dtype=np.uint8).tobytes()
+    klens = (ctypes.c_uint32 * n)(*[len(k) for k in sel])
+    vlens = (ctypes.c_uint32 * n)(*([VAL_LEN] * n))
+    kblob = b"".join(sel)
+    f = store._f("bulk_create")
+    rc = f(ctypes.c_void_p(store.h), kblob, klens, vals, vlens, ctypes.c_size_t(n))
+    assert rc == 0, "bulk_create failed"
+    revs = {k: None for k in sel}  # latest rev per key tracked client-side
+    base = store.current_rev() - n
+    for i, k in enumerate(sel):
+        revs[k] = base + i + 1
+    # extra revisions, zipf(1.1)
+    if extra_revs and n:
+        zs = (rng.zipf(1.1, size=extra_revs) - 1) % n
+        vbuf = rng.integers(0, 256, size=VAL_LEN, dtype=np.uint8).tobytes()
+        for z in zs:
+            k = sel[int(z)]
+            r = store.update(k, vbuf, revs[k])
+            assert r.succeeded, "seed update failed"
+            revs[k] = r.header_revision
+    # tombstones
+    if tomb_frac and n:
+        nt = int(n * tomb_frac)
+        idx = rng.choice(n, size=nt, replace=False)
+        for i in idx:
+            k = sel[int(i)]
+            r = store.delete(k, revs[k])
+            assert r.succeeded
+            revs[k] = None
+    store._f("flush")(ctypes.c_void_p(store.h))
+    return sel, revs
+
+
+def pack_queries(qs):
+    parts = []
+    for s, e, rev, limit in qs:
+        parts.append(struct.pack("<IIQQ", len(s), len(e), rev, limit))
+        parts.append(s)
+        parts.append(e)
+    return b"".join(parts)
+
+
+def gen_step_queries(rng, namespaces, cur_rev, n):
+    out = []
+    for _ in range(n):
+        ns = namespaces[int(rng.integers(len(namespaces)))]
+        off = int(min(rng.zipf(1.1), 500))
+        rev = max(1, cur_rev - off)
+        out.append((ns + b"/", ns + b"0", rev, LIMIT))
+    return out
+
+
+def bench_range_call(store, blob, nq, d2h):
+    f = store._f("bench_range")
+    total = ctypes.c_ulonglong()
+    secs = ctypes.c_double()
+    rc = f(ctypes.c_void_p(store.h), blob, ctypes.c_size_t(nq),
+           ctypes.c_int(1 if d2h else 0), ctypes.byref(total), ctypes.byref(secs))
+    assert rc == 0
+    return total.value, secs.value
+
+
+def perf(store):
+    buf = ctypes.create_string_buffer(4096)
+    rc = store._f("perf_json")(ctypes.c_void_p(store.h), buf, ctypes.c_size_t(4096))
+    assert rc == 0
+    return json.loads(buf.value.decode())
+
+
+def run_txns(store, live_keys, revs, rng, n):
+    vbuf = rng.integers(0, 256, size=VAL_LEN, dtype=np.uint8).tobytes()
+    done = 0
+    while done < n:
+        k = live_keys[int(rng.integers(len(live_keys)))]
+        pr = revs.get(k)
+        if pr is None:
+            continue
+        r = store.update(k, vbuf, pr)
+        if r.succeeded:
+            revs[k] = r.header_revision
+        done += 1
+
+
+def cpu_baseline_leg(namespaces, keys, sample_qs):
+    """Oracle (CPU restatement, kind 'port') on the host cores — the only
+    bench.py use of oracle/ (DESIGN.md §1)."""
+    from kbclient import open_oracle
+
+    rng = np.random.default_rng(SEED)
+    o = open_oracle()
+    o.set_current_rev(1000)
+    t0 = time.time()
+    # bounded sample: a slice of the keyspace big enough for real ranges
+    ns_n = min(len(namespaces), 200)
+    sub_ns = namespaces[:ns_n]
+    sub_keys = [k for k in keys if k.rsplit(b"/", 1)[0] in set(sub_ns)]
+    n = len(sub_keys)
+    vals = rng.integers(0, 256, size=n * VAL_LEN, dtype=np.uint8).tobytes()
+    klens = (ctypes.c_uint32 * n)(*[len(k) for k in sub_keys])
+    vlens = (ctypes.c_uint32 * n)(*([VAL_LEN] * n))
+    f = o.lib.okb_bulk_create
+    rc = f(ctypes.c_void_p(o.h), b"".join(sub_keys), klens, vals, vlens,
+           ctypes.c_size_t(n))
+    assert rc == 0
+    build_s = time.time() - t0
+    threads = os.cpu_count() or 1
+    qs = [(s, e, 0, lim) for (s, e, _r, lim) in sample_qs
+          if s.rsplit(b"/", 2)[0] + b"/" + s.rsplit(b"/", 2)[1] in set(sub_ns)
+          or True]
+    # clamp queries to the loaded namespaces
+    qs = []
+    qrng = np.random.default_rng(SEED + 7)
+    for _ in range(4000):
+        ns = sub_ns[int(qrng.integers(len(sub_ns)))]
+        qs.append((ns + b"/", ns + b"0", 0, LIMIT))
+    blob = pack_queries(qs)
+    fb = o.lib.okb_bench_range
+    total = ctypes.c_ulonglong()
+    secs = ctypes.c_double()
+    rc = fb(ctypes.c_void_p(o.h), blob, ctypes.c_size_t(len(qs)),
+            ctypes.c_int(threads), ctypes.byref(total), ctypes.byref(secs))
+    assert rc == 0
+    range_ops_s = len(qs) / secs.value
+    # txn rate (single-writer, as the reference serializes writes)
+    t0 = time.time()
+    ntx = 2000
+    rev = None
+    for i in range(ntx):
+        k = sub_keys[i % len(sub_keys)]
+        r = o.update(k, b"x" * VAL_LEN, 0)  # CAS fail path still writes a rev
+    txn_s = time.time() - t0
+    txn_ops_s = ntx / txn_s
+    o.close()
+    mix = 1.0 / (RANGE_SHARE / range_ops_s + (1 - RANGE_SHARE) / txn_ops_s)
+    return {
+        "value": round(mix, 1),
+        "unit": "ops/s",
+        "cores": threads,
+        "kind": "port",
+        "sample": f"{ns_n} namespaces/{n} keys subset; {len(qs)} Range(limit=500) "
+                  f"on {threads} threads ({secs.value:.1f}s) + {ntx} serial txns "
+                  f"({txn_s:.1f}s); mix = harmonic 90/10; build {build_s:.1f}s",
+        "range_ops_per_sec": round(range_ops_s, 1),
+        "txn_ops_per_sec": round(txn_ops_s, 1),
+    }
+
+
+def watch_leg(store, namespaces, live_keys, revs, rng, n_watchers, n_events):
+    import kbclient
+    wids = []
+    for i in range(n_watchers):
+        if i % 5 == 0:
+            pfx = b"/registry/pods/"
+        else:
+            pfx = namespaces[int(rng.integers(len(namespaces)))] + b"/"
+        st, wid = store.watch(pfx, 0)
+        assert st == kbclient.OK
+        wids.append(wid)
+    p0 = perf(store)
+    t0 = time.time()
+    run_txns(store, live_keys, revs, rng, n_events)
+    # final pump via a poll
+    store.watch_poll(wids[0])
+    dt = time.time() - t0
+    p1 = perf(store)
+    delivered = p1.get("delivered", 0) - p0.get("delivered", 0)
+    for wid in wids:
+        store.watch_cancel(wid)
+    return delivered / dt if dt > 0 else 0.0, delivered
+
+
+def main():
+    ap = argparse.ArgumentParser()
+    ap.add_argument("--gpus", type=int, default=1)
+    ap.add_argument("--steps", type=int, default=24)
+    ap.add_argument("--warmup", type=int, default=6)
+    ap.add_argument("--nns", type=int, default=2000)
+    ap.add_argument("--per-ns", type=int, default=500)
+    ap.add_argument("--extra-revs", type=int, default=200000)
+    ap.add_argument("--watchers", type=int, default=1000)
+    ap.add_argument("--watch-events", type=int, default=6000)
+    ap.add_argument("--no-cpu-baseline", action="store_true")
+    args = ap.parse_args()
+
+    rank = int(os.environ.get("RANK", "0"))
+    world = int(os.environ.get("WORLD_SIZE", "1"))
+    local_rank = int(os.environ.get("LOCAL_RANK", str(rank)))
+    dist = None
+    if world > 1:
+        import torch
+        import torch.distributed as tdist
+        os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
+        tdist.init_process_group(backend="gloo", rank=rank, world_size=world)
+        dist = tdist
+        os.environ["KB_DEVICE"] = str(local_rank)
+
+    # device capacity sizing for this workload
+    total_rows = int((args.nns * args.per_ns * 2.1 + args.extra_revs) * 1.6 / world)
+    os.environ.setdefault("KB_MAX_ROWS", str(max(total_rows, 1 << 20)))
+    os.environ.setdefault("KB_HEAP_BYTES",
+                          str(int((args.nns * args.per_ns / world + args.extra_revs)
+                                  * (VAL_LEN + 16) * 1.4) + (128 << 20)))
+    os.environ.setdefault("KB_FLUSH_ROWS", "8192")
+
+    import kubebrain_amd
+    import torch
+
+    namespaces, keys = make_keys(args.nns, args.per_ns)
+    my_ns = set(ns for i, ns in enumerate(namespaces) if i % world == rank)
+
+    store = kubebrain_amd.open_store(store_prefix=b"/registry")
+    store.set_current_rev(1000)
+    rng = np.random.default_rng(SEED + rank)
+    log(rank, f"[bench] loading shard rank={rank}/{world} "
+              f"({len(my_ns)} namespaces)...")
+    t0 = time.time()
+    sel, revs = build_store(store, namespaces, keys, rng,
+                            args.extra_revs // world, 0.02,
+                            my_ns if world > 1 else None)
+    live = [k for k in sel if revs[k] is not None]
+    log(rank, f"[bench] loaded {len(sel)} keys in {time.time()-t0:.1f}s; "
+              f"slab={perf(store)['slab_rows']} rows")
+
+    my_ns_list = sorted(my_ns)
+    qrng = np.random.default_rng(SEED + 100 + rank)
+
+    def one_step(d2h=False):
+        nq = int(OPS_PER_STEP * RANGE_SHARE / 1)  # 900 ranges
+        ntx = OPS_PER_STEP - nq                   # 100 txns
+        qs = gen_step_queries(qrng, my_ns_list, store.current_rev(), nq)
+        blob = pack_queries(qs)
+        tot, _ = bench_range_call(store, blob, len(qs), d2h)
+        run_txns(store, live, revs, qrng, ntx)
+        return tot
+
+    def barrier():
+        if dist:
+            dist.barrier()
+        if torch.cuda.is_available():
+            torch.cuda.synchronize()
+
+    for _ in range(args.warmup):
+        one_step()
+    barrier()
+    store._f("perf_reset")(ctypes.c_void_p(store.h))
+    t0 = time.time()
+    for _ in range(args.steps):
+        one_step()
+    barrier()
+    elapsed = time.time() - t0
+    if dist:
+        import torch as _t
+        e = _t.tensor([elapsed])
+        dist.all_reduce(e, op=dist.ReduceOp.MAX)
+        elapsed = float(e.item())
+    p = perf(store)
+
+    # PCIe-inclusive measurement (separate, untimed-region)
+    t0 = time.time()
+    for _ in range(max(2, args.steps // 8)):
+        one_step(d2h=True)
+    d2h_elapsed = time.time() - t0
+    d2h_steps = max(2, args.steps // 8)
+
+    # watch fan-out leg
+    wrate, delivered = watch_leg(store, my_ns_list, live, revs, qrng,
+                                 max(args.watchers // world, 8),
+                                 args.watch_events // world)
+    if dist:
+        import torch as _t
+        w = _t.tensor([wrate])
+        dist.all_reduce(w, op=dist.ReduceOp.SUM)
+        wrate = float(w.item())
+
+    total_ops = OPS_PER_STEP * args.steps * world
+    value = total_ops / elapsed
+    scan_s = p["scan_ms"] / 1e3
+    achieved = (p["rows_scanned"] * 16) / scan_s if scan_s > 0 else 0.0
+    peak = 8.0e12
+    roofline = {
+        "bound": "hbm",
+        "achieved": round(achieved, 1),
+        "peak": peak,
+        "unit": "GB/s" if False else "B/s",
+        "frac": round(achieved / peak, 4),
+        "traffic": None,
+        "kernel": "k_range_scan",
+        "note": "achieved = rows_scanned x 16B (meta+rev per row, DESIGN.md "
+                "§3.3) / HIP-event scan time on the store stream",
+    }
+
+    cpu_baseline = None
+    if rank == 0 and world == 1 and not args.no_cpu_baseline:
+        log(rank, "[bench] cpu baseline (oracle, bounded sample)...")
+        cpu_baseline = cpu_baseline_leg(namespaces, keys, [])
+
+    if rank == 0:
+        out = {
+            "metric": "range+txn ops/s (90/10, Range limit=500) + watch events/s",
+            "value": round(value, 1),
+            "unit": "ops/s",
+            "n_gpus": world,
+            "steps": args.steps,
+            "warmup": args.warmup,
+            "ms_per_step": round(elapsed / args.steps * 1e3, 3),
+            "higher_is_better": True,
+            "scaling": "weak",
+            "vs_baseline": None,
+            "dtype": "u8",
+            "data": "synthetic",
+            "config": {
+                "workload": "configs[1]: 1M Pod-style keys / 200k extra revisions "
+                            "(zipf 1.1) / 2% tombstones, Range(limit=500)+Txn "
+                            "conditional-update 90/10, 512B values, single MI355X"
+                            if world == 1 else
+                            f"configs[1] sharded by namespace hash over {world} "
+                            f"GPUs (weak scaling)",
+                "n_keys": args.nns * args.per_ns,
+                "extra_revs": args.extra_revs,
+                "limit": LIMIT,
+                "ops_per_step": OPS_PER_STEP,
+            },
+            "watch_events_per_sec": round(wrate, 1),
+            "watch_delivered_rank0": delivered,
+            "ops_per_sec_with_d2h": round(OPS_PER_STEP * d2h_steps * world / d2h_elapsed, 1),
+            "roofline": roofline,
+            "cpu_baseline": cpu_baseline,
+            "perf": p,
+        }
+        print(json.dumps(out))
+    if dist:
+        dist.destroy_process_group()
+
+
+if __name__ == "__main__":
+    main()
