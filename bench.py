@@ -164,10 +164,7 @@ def cpu_baseline_leg(namespaces, keys, sample_qs):
     assert rc == 0
     build_s = time.time() - t0
     threads = os.cpu_count() or 1
-    qs = [(s, e, 0, lim) for (s, e, _r, lim) in sample_qs
-          if s.rsplit(b"/", 2)[0] + b"/" + s.rsplit(b"/", 2)[1] in set(sub_ns)
-          or True]
-    # clamp queries to the loaded namespaces
+    # queries clamped to the loaded namespaces
     qs = []
     qrng = np.random.default_rng(SEED + 7)
     for _ in range(4000):
@@ -175,19 +172,30 @@ def cpu_baseline_leg(namespaces, keys, sample_qs):
         qs.append((ns + b"/", ns + b"0", 0, LIMIT))
     blob = pack_queries(qs)
     fb = o.lib.okb_bench_range
-    total = ctypes.c_ulonglong()
-    secs = ctypes.c_double()
-    rc = fb(ctypes.c_void_p(o.h), blob, ctypes.c_size_t(len(qs)),
-            ctypes.c_int(threads), ctypes.byref(total), ctypes.byref(secs))
-    assert rc == 0
-    range_ops_s = len(qs) / secs.value
-    # txn rate (single-writer, as the reference serializes writes)
+    # bounded sample: repeat the query batch until >=5s of CPU work
+    tot_q, tot_s = 0, 0.0
+    while tot_s < 5.0:
+        total = ctypes.c_ulonglong()
+        secs = ctypes.c_double()
+        rc = fb(ctypes.c_void_p(o.h), blob, ctypes.c_size_t(len(qs)),
+                ctypes.c_int(threads), ctypes.byref(total), ctypes.byref(secs))
+        assert rc == 0
+        tot_q += len(qs)
+        tot_s += secs.value
+    range_ops_s = tot_q / tot_s
+    secs = ctypes.c_double(tot_s)
+    # txn rate (single-writer, as the reference serializes writes):
+    # successful conditional updates with the tracked revision
+    base = 1000
+    revs = {k: base + i + 1 for i, k in enumerate(sub_keys)}
     t0 = time.time()
     ntx = 2000
-    rev = None
+    vx = b"x" * VAL_LEN
     for i in range(ntx):
         k = sub_keys[i % len(sub_keys)]
-        r = o.update(k, b"x" * VAL_LEN, 0)  # CAS fail path still writes a rev
+        r = o.update(k, vx, revs[k])
+        assert r.succeeded
+        revs[k] = r.header_revision
     txn_s = time.time() - t0
     txn_ops_s = ntx / txn_s
     o.close()
@@ -197,8 +205,8 @@ def cpu_baseline_leg(namespaces, keys, sample_qs):
         "unit": "ops/s",
         "cores": threads,
         "kind": "port",
-        "sample": f"{ns_n} namespaces/{n} keys subset; {len(qs)} Range(limit=500) "
-                  f"on {threads} threads ({secs.value:.1f}s) + {ntx} serial txns "
+        "sample": f"{ns_n} namespaces/{n} keys subset; {tot_q} Range(limit=500) "
+                  f"on {threads} threads ({tot_s:.1f}s) + {ntx} serial txns "
                   f"({txn_s:.1f}s); mix = harmonic 90/10; build {build_s:.1f}s",
         "range_ops_per_sec": round(range_ops_s, 1),
         "txn_ops_per_sec": round(txn_ops_s, 1),
@@ -260,7 +268,9 @@ def main():
     os.environ.setdefault("KB_HEAP_BYTES",
                           str(int((args.nns * args.per_ns / world + args.extra_revs)
                                   * (VAL_LEN + 16) * 1.4) + (128 << 20)))
-    os.environ.setdefault("KB_FLUSH_ROWS", "8192")
+    # large threshold: loading flushes a few times; during the bench the
+    # memtable is flushed at every kb_bench_range call anyway (timed)
+    os.environ.setdefault("KB_FLUSH_ROWS", "262144")
 
     import kubebrain_amd
     import torch
@@ -334,19 +344,25 @@ def main():
 
     total_ops = OPS_PER_STEP * args.steps * world
     value = total_ops / elapsed
-    scan_s = p["scan_ms"] / 1e3
-    achieved = (p["rows_scanned"] * 16) / scan_s if scan_s > 0 else 0.0
+    # roofline over the range path (scan + winner gather), per SURVEY §8d's
+    # algorithmic bytes: 16B per row scanned (meta+rev) +
+    # gathered record bytes (key+value+header per winner)
+    rng_s = (p["scan_ms"] + p["gather_ms"]) / 1e3
+    alg_bytes = p["rows_scanned"] * 16 + p["bytes_gathered"]
+    achieved = alg_bytes / rng_s if rng_s > 0 else 0.0
     peak = 8.0e12
     roofline = {
         "bound": "hbm",
-        "achieved": round(achieved, 1),
-        "peak": peak,
-        "unit": "GB/s" if False else "B/s",
+        "achieved": round(achieved / 1e9, 2),
+        "peak": peak / 1e9,
+        "unit": "GB/s",
         "frac": round(achieved / peak, 4),
         "traffic": None,
-        "kernel": "k_range_scan",
-        "note": "achieved = rows_scanned x 16B (meta+rev per row, DESIGN.md "
-                "§3.3) / HIP-event scan time on the store stream",
+        "kernel": "k_range_scan+k_gather",
+        "scan_only_GBps": round((p["rows_scanned"] * 16) / (p["scan_ms"] / 1e3) / 1e9, 2)
+                          if p["scan_ms"] > 0 else None,
+        "note": "achieved = (rows_scanned x 16B + gathered bytes) / HIP-event "
+                "kernel time on the store stream",
     }
 
     cpu_baseline = None

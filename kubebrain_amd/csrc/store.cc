@@ -243,6 +243,7 @@ void Store::pumpEvents() {
           int j = __builtin_ctzll(bits);
           bits &= bits - 1;
           w.queue.push_back(pending_[b0 + c * 64 + j]);
+          delivered_++;
         }
       }
       if (w.queue.size() > kWatchQueueCap) {
@@ -912,6 +913,7 @@ std::string Store::PerfJson() {
            "\"bytes_gathered\":%lld,\"winners\":%lld,\"merges\":%lld,"
            "\"compacts\":%lld,\"filter_launches\":%lld,\"filter_events\":%lld,"
            "\"filter_watchers\":%lld,\"slab_rows\":%lld,\"heap_used\":%lld,"
+           "\"delivered\":%lld,"
            "\"ops\":{\"create\":%lld,\"update\":%lld,\"delete\":%lld,"
            "\"range\":%lld}}",
            p.scan_ms, p.gather_ms, p.get_ms, p.compact_ms, p.merge_ms,
@@ -920,7 +922,8 @@ std::string Store::PerfJson() {
            (long long)p.winners, (long long)p.merges, (long long)p.compacts,
            (long long)p.filter_launches, (long long)p.filter_events,
            (long long)p.filter_watchers, (long long)slab_->rows(),
-           (long long)slab_->heap_used(), (long long)ops_create_,
+           (long long)slab_->heap_used(), (long long)delivered_,
+           (long long)ops_create_,
            (long long)ops_update_, (long long)ops_delete_, (long long)ops_range_);
   return buf;
 }
