@@ -117,6 +117,10 @@ int kb_event_log(kb_store*, uint8_t* out, size_t cap, size_t* out_len);
  * arena (the `value` mode per DESIGN.md §5). */
 int kb_bench_range(kb_store*, const uint8_t* qbuf, size_t nq, int d2h,
                    unsigned long long* total_kvs, double* secs);
+/* batched conditional updates (txn.go:249-265); tbuf = n × {u32 klen;
+ * u64 prev_rev; u32 vlen; key; val}; out_revs[i] = new revision or 0 on CAS
+ * failure */
+int kb_bench_txn(kb_store*, const uint8_t* tbuf, size_t n, uint64_t* out_revs);
 /* fast bulk insert == n serial Creates of fresh keys (see okb_bulk_create) */
 int kb_bulk_create(kb_store*, const uint8_t* keys, const uint32_t* klens,
                    const uint8_t* vals, const uint32_t* vlens, size_t n);

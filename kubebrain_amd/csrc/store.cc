@@ -903,6 +903,29 @@ bool Store::BenchRange(const uint8_t* qbuf, size_t nq, bool d2h,
   return true;
 }
 
+bool Store::BenchTxn(const uint8_t* tbuf, size_t n, uint64_t* out_revs,
+                     std::string* err) {
+  // packed records: {u32 klen; u64 prev_rev; u32 vlen; key; val} x n
+  const uint8_t* p = tbuf;
+  for (size_t i = 0; i < n; ++i) {
+    uint32_t klen, vlen;
+    uint64_t prev;
+    memcpy(&klen, p, 4); p += 4;
+    memcpy(&prev, p, 8); p += 8;
+    memcpy(&vlen, p, 4); p += 4;
+    Bytes key((const char*)p, klen); p += klen;
+    Bytes val((const char*)p, vlen); p += vlen;
+    Status st;
+    auto r = Update(key, val, prev, &st);
+    if (st != OK) {
+      if (err) *err = "bench txn failed st=" + std::to_string(st);
+      return false;
+    }
+    out_revs[i] = r.succeeded ? r.header_revision : 0;
+  }
+  return true;
+}
+
 std::string Store::PerfJson() {
   const kbslab::Perf& p = slab_->perf;
   char buf[1536];

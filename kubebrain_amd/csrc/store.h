@@ -88,6 +88,10 @@ class Store {
                   const uint32_t* vlens, size_t n, std::string* err);
   bool BenchRange(const uint8_t* qbuf, size_t nq, bool d2h,
                   unsigned long long* total, double* secs, std::string* err);
+  // batched conditional updates (txn.go:249-265 per op); out_revs[i] = new
+  // revision on success, 0 on CAS failure
+  bool BenchTxn(const uint8_t* tbuf, size_t n, uint64_t* out_revs,
+                std::string* err);
   std::string PerfJson();
   void PerfReset();
 
