@@ -106,6 +106,21 @@ __global__ void k_same_next(const uint8_t* __restrict__ keys,
   meta[i] = same ? (m | M_SAME_NEXT) : (m & ~M_SAME_NEXT);
 }
 
+// first row in [lo,hi) with (key,rev) >= (qkey,qrev)
+__device__ int64_t d_lb_range(const uint8_t* keys, const uint64_t* rev,
+                              int64_t lo, int64_t hi, const uint8_t* qkey,
+                              uint64_t qrev) {
+  while (lo < hi) {
+    int64_t mid = (lo + hi) >> 1;
+    int c = keycmp96(keys + mid * KEYW, qkey);
+    if (c < 0 || (c == 0 && rev[mid] < qrev)) lo = mid + 1; else hi = mid;
+  }
+  return lo;
+}
+
+constexpr uint64_t ROW_TAG_DELTA = 1ull << 63;  // winner row lives in the delta run
+constexpr uint64_t ROW_MASK = ROW_TAG_DELTA - 1;
+
 // ---- the north-star kernel: batched MVCC range scan --------------------
 // One workgroup per query. Winner predicate (scanner.go:389-516, DESIGN.md
 // §3.1): rev>0 ∧ rev<=R ∧ ¬tomb ∧ (¬same_next ∨ next.rev>R). Reads 16B/row
@@ -166,16 +181,217 @@ __global__ void k_range_scan(const uint8_t* __restrict__ keys,
   }
 }
 
+// ---- two-run variant: base run + sorted delta run merged at scan time ---
+// The delta run holds rows strictly newer than the base run's rows of the
+// same key, so the global winner of a key is the delta
+// winner when the delta has any row of the key with rev<=R, else the base
+// winner — base winners are suppressed by a delta probe, and the two ordered
+// winner lists are merged by rank (keys never collide across lists).
+
+// one run's winner scan (ordered append); returns written, *total = seen
+__device__ int64_t scan_run_winners(
+    const uint8_t* __restrict__ keys, const uint64_t* __restrict__ meta,
+    const uint64_t* __restrict__ rev, int64_t lo, int64_t hi, uint64_t R,
+    int64_t cap, uint64_t* out, int64_t out_cap, uint64_t tagbit,
+    const uint8_t* __restrict__ skeys, const uint64_t* __restrict__ srev,
+    int64_t slo, int64_t shi,  // suppression run (null => none)
+    int64_t* total_out, int64_t* scanned_accum, int64_t* cnt_s, int* wave_cnt) {
+  int lane = threadIdx.x & 63, w = threadIdx.x >> 6;
+  if (threadIdx.x == 0) *cnt_s = 0;
+  __syncthreads();
+  int64_t scanned = 0;
+  for (int64_t t = lo; t < hi; t += blockDim.x) {
+    int64_t i = t + threadIdx.x;
+    bool win = false;
+    if (i < hi) {
+      uint64_t r = rev[i], m = meta[i];
+      if (r > 0 && r <= R && !(m & M_TOMB))
+        win = !(m & M_SAME_NEXT) || rev[i + 1] > R;
+      if (win && skeys) {
+        const uint8_t* kk = keys + i * KEYW;
+        int64_t lb = d_lb_range(skeys, srev, slo, shi, kk, 1);
+        if (lb < shi && srev[lb] <= R && keycmp96(skeys + lb * KEYW, kk) == 0)
+          win = false;  // a newer (delta) row of this key wins instead
+      }
+    }
+    uint64_t b = __ballot(win);
+    if (lane == 0) wave_cnt[w] = __popcll(b);
+    __syncthreads();
+    int64_t waveoff = 0;
+    for (int k = 0; k < w; ++k) waveoff += wave_cnt[k];
+    int tile_total = wave_cnt[0] + wave_cnt[1] + wave_cnt[2] + wave_cnt[3];
+    if (win && out) {
+      int64_t idx = *cnt_s + waveoff + __popcll(b & ((1ull << lane) - 1));
+      if (idx < cap && idx < out_cap) out[idx] = (uint64_t)i | tagbit;
+    }
+    __syncthreads();
+    if (threadIdx.x == 0) *cnt_s += tile_total;
+    scanned += min((int64_t)blockDim.x, hi - t);
+    __syncthreads();
+    if (*cnt_s >= cap) break;
+  }
+  __syncthreads();
+  int64_t tot = *cnt_s;
+  *total_out = tot;
+  if (scanned_accum) *scanned_accum += scanned;
+  int64_t written = tot < cap ? tot : cap;
+  if (written > out_cap) written = out_cap;
+  return out ? written : 0;
+}
+
+// lower_bound over a winner list via key indirection (no cross-list ties)
+__device__ int64_t d_lb_winlist(const uint64_t* rows, int64_t n,
+                                const uint8_t* bkeys, const uint8_t* dkeys,
+                                const uint8_t* qkey) {
+  int64_t lo = 0, hi = n;
+  while (lo < hi) {
+    int64_t mid = (lo + hi) >> 1;
+    uint64_t rt = rows[mid];
+    const uint8_t* mk = (rt & ROW_TAG_DELTA ? dkeys : bkeys) +
+                        (rt & ROW_MASK) * KEYW;
+    if (keycmp96(mk, qkey) < 0) lo = mid + 1; else hi = mid;
+  }
+  return lo;
+}
+
+__global__ void k_range_scan2(
+    const uint8_t* __restrict__ bkeys, const uint64_t* __restrict__ bmeta,
+    const uint64_t* __restrict__ brev, int64_t n,
+    const uint8_t* __restrict__ dkeys, const uint64_t* __restrict__ dmeta,
+    const uint64_t* __restrict__ drev, int64_t dn,
+    const DevRangeQ* __restrict__ qs, int nq, int64_t max_cap,
+    uint64_t* __restrict__ rows_b, uint64_t* __restrict__ rows_d,
+    uint64_t* __restrict__ rows_m, int64_t* __restrict__ found_out,
+    int64_t* __restrict__ total_out, unsigned long long* __restrict__ scanned_out) {
+  int q = blockIdx.x;
+  if (q >= nq) return;
+  __shared__ int64_t lo_s, hi_s, dlo_s, dhi_s, cnt_s;
+  __shared__ int wave_cnt[4];
+  const DevRangeQ& Q = qs[q];
+  if (threadIdx.x == 0) lo_s = d_lb_range(bkeys, brev, 0, n, Q.start, 0);
+  if (threadIdx.x == 64) hi_s = d_lb_range(bkeys, brev, 0, n, Q.end, 0);
+  if (threadIdx.x == 128) dlo_s = d_lb_range(dkeys, drev, 0, dn, Q.start, 0);
+  if (threadIdx.x == 192) dhi_s = d_lb_range(dkeys, drev, 0, dn, Q.end, 0);
+  __syncthreads();
+  const int64_t cap = Q.cap > 0 ? Q.cap : INT64_MAX;
+  int64_t scanned = 0;
+  int64_t dtotal = 0, btotal = 0;
+  uint64_t* outd = Q.count_only ? nullptr : rows_d + (int64_t)q * max_cap;
+  uint64_t* outb = Q.count_only ? nullptr : rows_b + (int64_t)q * max_cap;
+  int64_t nB = scan_run_winners(dkeys, dmeta, drev, dlo_s, dhi_s, Q.read_rev,
+                                cap, outd, max_cap, ROW_TAG_DELTA, nullptr,
+                                nullptr, 0, 0, &dtotal, &scanned, &cnt_s, wave_cnt);
+  int64_t nA = scan_run_winners(bkeys, bmeta, brev, lo_s, hi_s, Q.read_rev, cap,
+                                outb, max_cap, 0, dn ? dkeys : nullptr, drev,
+                                dlo_s, dhi_s, &btotal, &scanned, &cnt_s, wave_cnt);
+  // merge by rank into rows_m (keys are disjoint across the two lists)
+  int64_t cap_m = nA + nB;
+  if (cap_m > cap) cap_m = cap;
+  if (cap_m > max_cap) cap_m = max_cap;
+  if (!Q.count_only) {
+    uint64_t* outm = rows_m + (int64_t)q * max_cap;
+    for (int64_t j = threadIdx.x; j < nA + nB; j += blockDim.x) {
+      uint64_t rt;
+      int64_t pos;
+      if (j < nA) {
+        rt = outb[j];
+        const uint8_t* k = bkeys + (rt & ROW_MASK) * KEYW;
+        pos = j + d_lb_winlist(outd, nB, bkeys, dkeys, k);
+      } else {
+        rt = outd[j - nA];
+        const uint8_t* k = dkeys + (rt & ROW_MASK) * KEYW;
+        pos = (j - nA) + d_lb_winlist(outb, nA, bkeys, dkeys, k);
+      }
+      if (pos < cap_m) outm[pos] = rt;
+    }
+  }
+  if (threadIdx.x == 0) {
+    total_out[q] = btotal + dtotal;
+    found_out[q] = Q.count_only ? 0 : cap_m;
+    atomicAdd(scanned_out, (unsigned long long)scanned);
+  }
+}
+
+// two-run point read: the delta run wins when it has any row of the key <= R
+__global__ void k_get2(const uint8_t* __restrict__ bkeys,
+                       const uint64_t* __restrict__ bmeta,
+                       const uint64_t* __restrict__ brev,
+                       const uint64_t* __restrict__ bvo,
+                       const uint8_t* __restrict__ dkeys,
+                       const uint64_t* __restrict__ dmeta,
+                       const uint64_t* __restrict__ drev,
+                       const uint64_t* __restrict__ dvo, int64_t n, int64_t dn,
+                       const uint8_t* __restrict__ heap,
+                       const DevGetQ* __restrict__ qs, int nq,
+                       uint8_t* __restrict__ out, int64_t slot,
+                       uint64_t* __restrict__ orev, uint64_t* __restrict__ ometa,
+                       int32_t* __restrict__ ofound, int32_t* __restrict__ oovf) {
+  int q = blockIdx.x * (blockDim.x / 64) + (threadIdx.x >> 6);
+  int lane = threadIdx.x & 63;
+  if (q >= nq) return;
+  const DevGetQ& Q = qs[q];
+  int64_t row = -1;
+  int isdelta = 0;
+  if (lane == 0) {
+    if (dn > 0) {
+      int64_t ub = d_upper_bound(dkeys, drev, dn, Q.key, Q.read_rev);
+      if (ub > 0) {
+        int64_t c = ub - 1;
+        if (drev[c] >= 1 && keycmp96(dkeys + c * KEYW, Q.key) == 0) {
+          row = c;
+          isdelta = 1;
+        }
+      }
+    }
+    if (row < 0) {
+      int64_t ub = d_upper_bound(bkeys, brev, n, Q.key, Q.read_rev);
+      if (ub > 0) {
+        int64_t c = ub - 1;
+        if (brev[c] >= 1 && keycmp96(bkeys + c * KEYW, Q.key) == 0) row = c;
+      }
+    }
+  }
+  row = __shfl(row, 0);
+  isdelta = __shfl(isdelta, 0);
+  if (row < 0) {
+    if (lane == 0) { ofound[q] = 0; oovf[q] = 0; }
+    return;
+  }
+  const uint64_t* meta = isdelta ? dmeta : bmeta;
+  const uint64_t* rev = isdelta ? drev : brev;
+  const uint64_t* vo = isdelta ? dvo : bvo;
+  uint64_t m = meta[row];
+  uint32_t vlen = meta_vlen(m);
+  if (lane == 0) {
+    ofound[q] = 1;
+    orev[q] = rev[row];
+    ometa[q] = m;
+    oovf[q] = vlen > slot ? 1 : 0;
+  }
+  if (vlen > slot) return;
+  const uint8_t* vs = heap + vo[row];
+  uint8_t* vd = out + (int64_t)q * slot;
+  uint32_t words = vlen >> 2;
+  for (uint32_t b = lane; b < words; b += 64)
+    ((uint32_t*)vd)[b] = ((const uint32_t*)vs)[b];
+  if (lane == 0)
+    for (uint32_t b = words * 4; b < vlen; ++b) vd[b] = vs[b];
+}
+
 // ---- gather: winners -> packed records in the per-query device arena ----
 // record: u64 rev | u32 klen | u32 vlen | key (pad8) | val (pad8)
 __device__ __forceinline__ int64_t rec_bytes(uint32_t klen, uint32_t vlen) {
   return 16 + ((klen + 7) & ~7u) + ((vlen + 7) & ~7u);
 }
 
-__global__ void k_gather(const uint8_t* __restrict__ keys,
-                         const uint64_t* __restrict__ meta,
-                         const uint64_t* __restrict__ rev,
-                         const uint64_t* __restrict__ vo,
+__global__ void k_gather(const uint8_t* __restrict__ bkeys,
+                         const uint64_t* __restrict__ bmeta,
+                         const uint64_t* __restrict__ brev,
+                         const uint64_t* __restrict__ bvo,
+                         const uint8_t* __restrict__ dkeys,
+                         const uint64_t* __restrict__ dmeta,
+                         const uint64_t* __restrict__ drev,
+                         const uint64_t* __restrict__ dvo,
                          const uint8_t* __restrict__ heap,
                          const uint64_t* __restrict__ rows_out, int64_t max_cap,
                          const int64_t* __restrict__ found_out, int nq,
@@ -198,13 +414,14 @@ __global__ void k_gather(const uint8_t* __restrict__ keys,
     int64_t j = c0 + threadIdx.x;
     int64_t sz = 0;
     if (j < nwin) {
-      uint64_t m = meta[rows[j]];
+      uint64_t rt = rows[j];
+      uint64_t m = (rt & ROW_TAG_DELTA ? dmeta : bmeta)[rt & ROW_MASK];
       sz = rec_bytes(meta_klen(m), meta_vlen(m));
     }
     lds[threadIdx.x] = sz;
     __syncthreads();
     for (int off = 1; off < 256; off <<= 1) {
-      int64_t add = threadIdx.x >= off ? lds[threadIdx.x - off] : 0;
+      int64_t add = threadIdx.x >= (unsigned)off ? lds[threadIdx.x - off] : 0;
       __syncthreads();
       lds[threadIdx.x] += add;
       __syncthreads();
@@ -225,19 +442,21 @@ __global__ void k_gather(const uint8_t* __restrict__ keys,
   int lane = threadIdx.x & 63, w = threadIdx.x >> 6;
   uint8_t* qb = gbuf + (int64_t)q * qcap;
   for (int64_t j = w; j < nwin; j += blockDim.x / 64) {
-    int64_t row = (int64_t)rows[j];
-    uint64_t m = meta[row];
+    uint64_t rt = rows[j];
+    bool isd = (rt & ROW_TAG_DELTA) != 0;
+    int64_t row = (int64_t)(rt & ROW_MASK);
+    uint64_t m = (isd ? dmeta : bmeta)[row];
     uint32_t klen = meta_klen(m), vlen = meta_vlen(m);
     uint8_t* dst = qb + qoffs[j];
     if (lane == 0) {
-      *(uint64_t*)dst = rev[row];
+      *(uint64_t*)dst = (isd ? drev : brev)[row];
       ((uint32_t*)dst)[2] = klen;
       ((uint32_t*)dst)[3] = vlen;
     }
-    const uint8_t* ks = keys + row * KEYW;
+    const uint8_t* ks = (isd ? dkeys : bkeys) + row * KEYW;
     uint8_t* kd = dst + 16;
     for (uint32_t b = lane; b < klen; b += 64) kd[b] = ks[b];
-    const uint8_t* vs = heap + vo[row];
+    const uint8_t* vs = heap + (isd ? dvo : bvo)[row];
     uint8_t* vd = dst + 16 + ((klen + 7) & ~7u);
     uint32_t words = vlen >> 2;
     for (uint32_t b = lane; b < words; b += 64)
@@ -526,6 +745,9 @@ struct Slab::Impl {
 
   struct Col { uint8_t* keys = nullptr; uint64_t *meta = nullptr, *rev = nullptr, *vo = nullptr; };
   Col A, B;
+  Col DA, DB;            // delta-run ping-pong
+  int64_t dn = 0;        // delta-run rows
+  int64_t delta_cap = 0; // rows per delta buffer
   uint8_t *heapA = nullptr, *heapB = nullptr;
 
   // scan scratch (u64, shared across ops)
@@ -539,7 +761,9 @@ struct Slab::Impl {
   int64_t arena_bytes = 384ll << 20;
   DevRangeQ* d_qs = nullptr;
   DevGetQ* d_gq = nullptr;
-  uint64_t* d_rows = nullptr;   // max_q*max_cap
+  uint64_t* d_rows = nullptr;   // max_q*max_cap (base winners)
+  uint64_t* d_rows2 = nullptr;  // delta winners
+  uint64_t* d_rowsm = nullptr;  // merged winners
   int64_t* d_offs = nullptr;    // max_q*max_cap
   int64_t *d_found = nullptr, *d_total = nullptr, *d_gbytes = nullptr;
   int32_t *d_ovf = nullptr, *d_found32 = nullptr;
@@ -555,7 +779,7 @@ struct Slab::Impl {
   // delta upload scratch (grown on demand)
   uint8_t* d_dkeys = nullptr;
   uint64_t *d_dmeta = nullptr, *d_drev = nullptr, *d_dvo = nullptr;
-  int64_t delta_cap = 0;
+  int64_t upload_cap = 0;
 
   // watcher table
   int64_t wcap = 0;
@@ -577,6 +801,9 @@ struct Slab::Impl {
   ~Impl() {
     for (void* p : {(void*)A.keys, (void*)A.meta, (void*)A.rev, (void*)A.vo,
                     (void*)B.keys, (void*)B.meta, (void*)B.rev, (void*)B.vo,
+                    (void*)DA.keys, (void*)DA.meta, (void*)DA.rev, (void*)DA.vo,
+                    (void*)DB.keys, (void*)DB.meta, (void*)DB.rev, (void*)DB.vo,
+                    (void*)d_rows2, (void*)d_rowsm,
                     (void*)heapA, (void*)heapB, (void*)s_a, (void*)s_b,
                     (void*)s_c, (void*)s_d, (void*)lv1, (void*)lv1o, (void*)lv2,
                     (void*)lv2o, (void*)lv3, (void*)lv3o, (void*)d_qs,
@@ -637,7 +864,7 @@ struct Slab::Impl {
   }
 
   bool ensure_delta(int64_t m, std::string* err) {
-    if (m <= delta_cap) return true;
+    if (m <= upload_cap) return true;
     int64_t cap = m + m / 2 + 1024;
     for (void* p : {(void*)d_dkeys, (void*)d_dmeta, (void*)d_drev, (void*)d_dvo})
       if (p) (void)hipFree(p);
@@ -645,7 +872,41 @@ struct Slab::Impl {
     HIP_CHECK(hipMalloc(&d_dmeta, cap * 8));
     HIP_CHECK(hipMalloc(&d_drev, cap * 8));
     HIP_CHECK(hipMalloc(&d_dvo, cap * 8));
-    delta_cap = cap;
+    upload_cap = cap;
+    return true;
+  }
+
+  // generic two-run merge: src (n rows) + newer (m rows, device arrays,
+  // rev-rows replace) -> dst; recomputes same_next. dst must hold n+m rows.
+  bool mergeRuns(const Col& src, int64_t n, const uint8_t* nk,
+                 const uint64_t* nm, const uint64_t* nr, const uint64_t* nv,
+                 int64_t m, Col& dst, int64_t* out_n, std::string* err) {
+    uint64_t dropped = 0;
+    if (n > 0) {
+      int64_t nb = ceil_div(n, 256);
+      hipLaunchKernelGGL(k_merge_rank, dim3((uint32_t)nb), dim3(256), 0, stream,
+                         src.keys, src.rev, n, nk, nr, m, s_a, s_b);
+      HIP_CHECK(hipMemsetAsync(s_b + n, 0, 8, stream));
+      if (!scan(s_b, s_c, n + 1, &dropped, err)) return false;
+      hipLaunchKernelGGL(k_merge_scatter_base, dim3((uint32_t)nb), dim3(256), 0,
+                         stream, src.keys, src.meta, src.rev, src.vo, s_a, s_b,
+                         s_c, dst.keys, dst.meta, dst.rev, dst.vo, n);
+    } else {
+      HIP_CHECK(hipMemsetAsync(s_c, 0, 8, stream));
+    }
+    if (m > 0) {
+      int64_t mb = ceil_div(m, 256);
+      hipLaunchKernelGGL(k_merge_scatter_delta, dim3((uint32_t)mb), dim3(256),
+                         0, stream, src.keys, src.rev, n, nk, nm, nr, nv, m,
+                         s_c, dst.keys, dst.meta, dst.rev, dst.vo);
+    }
+    int64_t new_n = n - (int64_t)dropped + m;
+    if (new_n > 0) {
+      int64_t sb = ceil_div(new_n, 256);
+      hipLaunchKernelGGL(k_same_next, dim3((uint32_t)sb), dim3(256), 0, stream,
+                         dst.keys, dst.meta, new_n);
+    }
+    *out_n = new_n;
     return true;
   }
 };
@@ -687,6 +948,14 @@ Slab* Slab::Create(int64_t max_rows, int64_t heap_cap, int device,
     HIP_CHECK_NULL(hipMalloc(&c->rev, max_rows * 8));
     HIP_CHECK_NULL(hipMalloc(&c->vo, max_rows * 8));
   }
+  I->delta_cap = env_i64("KB_DELTA_CAP", 1 << 19);
+  if (I->delta_cap > max_rows) I->delta_cap = max_rows;
+  for (Impl::Col* c : {&I->DA, &I->DB}) {
+    HIP_CHECK_NULL(hipMalloc(&c->keys, I->delta_cap * KEYW));
+    HIP_CHECK_NULL(hipMalloc(&c->meta, I->delta_cap * 8));
+    HIP_CHECK_NULL(hipMalloc(&c->rev, I->delta_cap * 8));
+    HIP_CHECK_NULL(hipMalloc(&c->vo, I->delta_cap * 8));
+  }
   HIP_CHECK_NULL(hipMalloc(&I->heapA, heap_cap));
   HIP_CHECK_NULL(hipMalloc(&I->heapB, heap_cap));
   HIP_CHECK_NULL(hipMalloc(&I->s_a, (max_rows + 2) * 8));
@@ -704,6 +973,8 @@ Slab* Slab::Create(int64_t max_rows, int64_t heap_cap, int device,
   HIP_CHECK_NULL(hipMalloc(&I->d_qs, sizeof(DevRangeQ) * I->max_q));
   HIP_CHECK_NULL(hipMalloc(&I->d_gq, sizeof(DevGetQ) * I->max_q));
   HIP_CHECK_NULL(hipMalloc(&I->d_rows, (int64_t)I->max_q * I->max_cap * 8));
+  HIP_CHECK_NULL(hipMalloc(&I->d_rows2, (int64_t)I->max_q * I->max_cap * 8));
+  HIP_CHECK_NULL(hipMalloc(&I->d_rowsm, (int64_t)I->max_q * I->max_cap * 8));
   HIP_CHECK_NULL(hipMalloc(&I->d_offs, (int64_t)I->max_q * I->max_cap * 8));
   HIP_CHECK_NULL(hipMalloc(&I->d_found, I->max_q * 8));
   HIP_CHECK_NULL(hipMalloc(&I->d_total, I->max_q * 8));
@@ -726,18 +997,88 @@ Slab* Slab::Create(int64_t max_rows, int64_t heap_cap, int device,
 
 Slab::~Slab() { delete p; }
 int64_t Slab::rows() const { return p->n; }
+int64_t Slab::delta_rows() const { return p->dn; }
 int64_t Slab::heap_used() const { return p->heap_used_; }
+
+bool Slab::HeapAppend(const void* src, int64_t len, int64_t* off, std::string* err) {
+  Impl* I = p;
+  if (len == 0) { *off = I->heap_used_; return true; }
+  if (I->heap_used_ + len > I->heap_cap) {
+    if (err) *err = "heap full (KB_HEAP_BYTES)";
+    return false;
+  }
+  HIP_CHECK(hipMemcpyAsync(I->heapA + I->heap_used_, src, len,
+                           hipMemcpyHostToDevice, I->stream));
+  HIP_CHECK(hipMemcpyAsync(I->heapB + I->heap_used_, src, len,
+                           hipMemcpyHostToDevice, I->stream));
+  HIP_CHECK(hipStreamSynchronize(I->stream));
+  *off = I->heap_used_;
+  I->heap_used_ += len;
+  return true;
+}
+
+bool Slab::AppendRows(const uint8_t* keys, const uint64_t* meta,
+                      const uint64_t* rev, const uint64_t* vo, int64_t m,
+                      std::string* err) {
+  Impl* I = p;
+  if (m == 0) return true;
+  if (I->dn + m > I->delta_cap) {
+    // fold first to make room
+    if (!Fold(err)) return false;
+    if (m > I->delta_cap) { if (err) *err = "append larger than delta cap"; return false; }
+  }
+  HIP_CHECK(hipEventRecord(I->ev0, I->stream));
+  if (!I->ensure_delta(m, err)) return false;
+  HIP_CHECK(hipMemcpyAsync(I->d_dkeys, keys, m * KEYW, hipMemcpyHostToDevice, I->stream));
+  HIP_CHECK(hipMemcpyAsync(I->d_dmeta, meta, m * 8, hipMemcpyHostToDevice, I->stream));
+  HIP_CHECK(hipMemcpyAsync(I->d_drev, rev, m * 8, hipMemcpyHostToDevice, I->stream));
+  HIP_CHECK(hipMemcpyAsync(I->d_dvo, vo, m * 8, hipMemcpyHostToDevice, I->stream));
+  int64_t new_dn = 0;
+  if (!I->mergeRuns(I->DA, I->dn, I->d_dkeys, I->d_dmeta, I->d_drev, I->d_dvo,
+                    m, I->DB, &new_dn, err))
+    return false;
+  HIP_CHECK(hipEventRecord(I->ev1, I->stream));
+  HIP_CHECK(hipStreamSynchronize(I->stream));
+  float ms = 0;
+  (void)hipEventElapsedTime(&ms, I->ev0, I->ev1);
+  perf.merge_ms += ms;
+  perf.merges++;
+  std::swap(I->DA, I->DB);
+  I->dn = new_dn;
+  return true;
+}
+
+bool Slab::Fold(std::string* err) {
+  Impl* I = p;
+  if (I->dn == 0) return true;
+  if (I->n + I->dn > I->max_rows) { if (err) *err = "slab full (KB_MAX_ROWS)"; return false; }
+  HIP_CHECK(hipEventRecord(I->ev2, I->stream));
+  int64_t new_n = 0;
+  if (!I->mergeRuns(I->A, I->n, I->DA.keys, I->DA.meta, I->DA.rev, I->DA.vo,
+                    I->dn, I->B, &new_n, err))
+    return false;
+  HIP_CHECK(hipEventRecord(I->ev3, I->stream));
+  HIP_CHECK(hipStreamSynchronize(I->stream));
+  float ms = 0;
+  (void)hipEventElapsedTime(&ms, I->ev2, I->ev3);
+  perf.merge_ms += ms;
+  perf.merges++;
+  std::swap(I->A, I->B);
+  I->n = new_n;
+  I->dn = 0;
+  return true;
+}
 
 bool Slab::Merge(const DeltaRows& d, std::string* err) {
   Impl* I = p;
   if (d.m == 0 && d.heap.empty()) return true;
+  if (!Fold(err)) return false;  // keep run ordering invariants
   if (I->n + d.m > I->max_rows) { if (err) *err = "slab full (KB_MAX_ROWS)"; return false; }
   if (I->heap_used_ + (int64_t)d.heap.size() > I->heap_cap) {
     if (err) *err = "heap full (KB_HEAP_BYTES)";
     return false;
   }
   HIP_CHECK(hipEventRecord(I->ev0, I->stream));
-  // append new values to the heap
   if (!d.heap.empty()) {
     HIP_CHECK(hipMemcpyAsync(I->heapA + I->heap_used_, d.heap.data(),
                              d.heap.size(), hipMemcpyHostToDevice, I->stream));
@@ -755,32 +1096,10 @@ bool Slab::Merge(const DeltaRows& d, std::string* err) {
                              hipMemcpyHostToDevice, I->stream));
     HIP_CHECK(hipMemcpyAsync(I->d_dvo, d.vo.data(), d.m * 8,
                              hipMemcpyHostToDevice, I->stream));
-    int64_t n = I->n;
-    uint64_t dropped = 0;
-    if (n > 0) {
-      int64_t nb = ceil_div(n, 256);
-      hipLaunchKernelGGL(k_merge_rank, dim3((uint32_t)nb), dim3(256), 0,
-                         I->stream, I->A.keys, I->A.rev, n, I->d_dkeys,
-                         I->d_drev, d.m, I->s_a, I->s_b);
-      // dropx has n+1 entries (scatter_delta indexes dropx[lb], lb<=n)
-      HIP_CHECK(hipMemsetAsync(I->s_b + n, 0, 8, I->stream));
-      if (!I->scan(I->s_b, I->s_c, n + 1, &dropped, err)) return false;
-      hipLaunchKernelGGL(k_merge_scatter_base, dim3((uint32_t)nb), dim3(256), 0,
-                         I->stream, I->A.keys, I->A.meta, I->A.rev, I->A.vo,
-                         I->s_a, I->s_b, I->s_c, I->B.keys, I->B.meta, I->B.rev,
-                         I->B.vo, n);
-    } else {
-      HIP_CHECK(hipMemsetAsync(I->s_c, 0, 8, I->stream));
-    }
-    int64_t mb = ceil_div(d.m, 256);
-    hipLaunchKernelGGL(k_merge_scatter_delta, dim3((uint32_t)mb), dim3(256), 0,
-                       I->stream, I->A.keys, I->A.rev, n, I->d_dkeys, I->d_dmeta,
-                       I->d_drev, I->d_dvo, d.m, I->s_c, I->B.keys, I->B.meta,
-                       I->B.rev, I->B.vo);
-    int64_t new_n = n - (int64_t)dropped + d.m;
-    int64_t sb = ceil_div(new_n, 256);
-    hipLaunchKernelGGL(k_same_next, dim3((uint32_t)sb), dim3(256), 0, I->stream,
-                       I->B.keys, I->B.meta, new_n);
+    int64_t new_n = 0;
+    if (!I->mergeRuns(I->A, I->n, I->d_dkeys, I->d_dmeta, I->d_drev, I->d_dvo,
+                      d.m, I->B, &new_n, err))
+      return false;
     std::swap(I->A, I->B);
     I->n = new_n;
   }
@@ -807,14 +1126,17 @@ bool Slab::RangeBatch(const std::vector<DevRangeQ>& qs, bool d2h,
   HIP_CHECK(hipMemsetAsync(I->d_scanned, 0, 8, I->stream));
   HIP_CHECK(hipMemsetAsync(I->d_bytes, 0, 8, I->stream));
   HIP_CHECK(hipEventRecord(I->ev0, I->stream));
-  hipLaunchKernelGGL(k_range_scan, dim3(nq), dim3(256), 0, I->stream, I->A.keys,
-                     I->A.meta, I->A.rev, I->n, I->d_qs, nq, I->max_cap,
-                     I->d_rows, I->d_found, I->d_total, I->d_scanned);
+  hipLaunchKernelGGL(k_range_scan2, dim3(nq), dim3(256), 0, I->stream,
+                     I->A.keys, I->A.meta, I->A.rev, I->n, I->DA.keys,
+                     I->DA.meta, I->DA.rev, I->dn, I->d_qs, nq, I->max_cap,
+                     I->d_rows, I->d_rows2, I->d_rowsm, I->d_found, I->d_total,
+                     I->d_scanned);
   HIP_CHECK(hipEventRecord(I->ev1, I->stream));
   hipLaunchKernelGGL(k_gather, dim3(nq), dim3(256), 0, I->stream, I->A.keys,
-                     I->A.meta, I->A.rev, I->A.vo, I->heapA, I->d_rows,
-                     I->max_cap, I->d_found, nq, I->d_gbuf, qcap, I->d_offs,
-                     I->d_gbytes, I->d_ovf, I->d_bytes);
+                     I->A.meta, I->A.rev, I->A.vo, I->DA.keys, I->DA.meta,
+                     I->DA.rev, I->DA.vo, I->heapA, I->d_rowsm, I->max_cap,
+                     I->d_found, nq, I->d_gbuf, qcap, I->d_offs, I->d_gbytes,
+                     I->d_ovf, I->d_bytes);
   HIP_CHECK(hipEventRecord(I->ev2, I->stream));
   // small result metadata D2H
   std::vector<int64_t> found(nq), total(nq), gbytes(nq);
@@ -900,8 +1222,9 @@ bool Slab::GetBatch(const std::vector<DevGetQ>& qs, std::vector<GetResult>* outs
                            hipMemcpyHostToDevice, I->stream));
   HIP_CHECK(hipEventRecord(I->ev0, I->stream));
   int blocks = (int)ceil_div(nq, 4);
-  hipLaunchKernelGGL(k_get, dim3(blocks), dim3(256), 0, I->stream, I->A.keys,
-                     I->A.meta, I->A.rev, I->A.vo, I->heapA, I->n, I->d_gq, nq,
+  hipLaunchKernelGGL(k_get2, dim3(blocks), dim3(256), 0, I->stream, I->A.keys,
+                     I->A.meta, I->A.rev, I->A.vo, I->DA.keys, I->DA.meta,
+                     I->DA.rev, I->DA.vo, I->n, I->dn, I->heapA, I->d_gq, nq,
                      I->d_gbuf, slot, I->d_orev, I->d_ometa, I->d_found32,
                      I->d_ovf);
   HIP_CHECK(hipEventRecord(I->ev1, I->stream));
@@ -935,6 +1258,7 @@ bool Slab::GetBatch(const std::vector<DevGetQ>& qs, std::vector<GetResult>* outs
 bool Slab::Compact(const std::vector<std::pair<Bound, Bound>>& borders,
                    uint64_t compact_rev, uint64_t timeout_rev, std::string* err) {
   Impl* I = p;
+  if (!Fold(err)) return false;  // compaction sweeps the single base run
   int64_t n = I->n;
   if (n == 0) return true;
   HIP_CHECK(hipEventRecord(I->ev0, I->stream));
@@ -1003,6 +1327,7 @@ bool Slab::Compact(const std::vector<std::pair<Bound, Bound>>& borders,
 
 bool Slab::Dump(std::vector<DumpRow>* rows_out, std::string* err) {
   Impl* I = p;
+  if (!Fold(err)) return false;
   rows_out->clear();
   int64_t n = I->n;
   if (n == 0) return true;

@@ -109,11 +109,21 @@ class Slab {
   ~Slab();
 
   int64_t rows() const;
+  int64_t delta_rows() const;
   int64_t heap_used() const;
 
-  // merge sorted delta rows into the slab (GPU merge-path by ranks;
-  // delta rev-rows REPLACE base rev-rows of the same key).
+  // merge sorted delta rows straight into the BASE run (GPU merge by ranks;
+  // delta rev-rows REPLACE base rev-rows of the same key). Used by tests and
+  // bulk paths; the hot write path uses AppendRows/Fold below.
   bool Merge(const DeltaRows& d, std::string* err);
+
+  // append value bytes to the device heap; *off = absolute offset
+  bool HeapAppend(const void* p, int64_t len, int64_t* off, std::string* err);
+  // merge sorted new rows (vo already absolute) into the DELTA run
+  bool AppendRows(const uint8_t* keys, const uint64_t* meta, const uint64_t* rev,
+                  const uint64_t* vo, int64_t m, std::string* err);
+  // fold the delta run into the base run; delta becomes empty
+  bool Fold(std::string* err);
 
   // batched Range (the north-star kernel; scanner worker.run semantics,
   // scanner.go:389-516). d2h=false leaves records in the device arena

@@ -125,69 +125,88 @@ uint64_t Store::mustDeal(uint64_t prevRevision) {  // txn.go:139-142
 }
 
 void Store::putRow(const Bytes& key, uint64_t rev, const Bytes& val) {
-  MemRow r;
-  r.val = val;
-  r.tomb = (val == kTombstone);
-  memtable_[{key, rev}] = std::move(r);
+  NewRow r;
+  r.key = key;
+  r.rev = rev;
+  bool tomb = (val == kTombstone);
+  bool ev = key.find(kEvents) != Bytes::npos;
+  r.meta = kbslab::meta_make(tomb, false, ev, (uint32_t)key.size(),
+                             (uint32_t)val.size());
+  if (heap_base_ < 0) heap_base_ = slab_->heap_used();
+  r.vo = (uint64_t)(heap_base_ + (int64_t)heap_pending_.size());
+  heap_pending_ += val;
+  heap_pending_.resize((heap_pending_.size() + 3) & ~3ull, '\0');
+  newrows_.push_back(std::move(r));
 }
 
 void Store::putRevRow(const Bytes& key, uint64_t objrev, bool flag9) {
-  MemRow r;
-  r.flag9 = flag9;
-  r.objrev = objrev;
-  memtable_[{key, 0}] = std::move(r);
+  NewRow r;
+  r.key = key;
+  r.rev = 0;
+  bool ev = key.find(kEvents) != Bytes::npos;
+  r.meta = kbslab::meta_make(false, flag9, ev, (uint32_t)key.size(), flag9 ? 9 : 8);
+  r.vo = objrev;
+  auto it = nr_revrow_.find(key);
+  if (it != nr_revrow_.end()) {
+    newrows_[it->second] = std::move(r);  // replace in place (unique key@0)
+  } else {
+    nr_revrow_[key] = newrows_.size();
+    newrows_.push_back(std::move(r));
+  }
   revIndex_[key] = RevEntry{objrev, flag9};
   if (flag9) tombstoned_.insert(key); else tombstoned_.erase(key);
-  if (key.find(kEvents) != Bytes::npos) events_keys_.insert(key);
+  if (ev) events_keys_.insert(key);
 }
 
-void Store::maybeFlush() {
-  if ((int64_t)memtable_.size() >= cfg_.flush_rows) {
-    std::string err;
-    (void)flushLocked(&err);  // errors surface on the next explicit call
-  }
-}
-
-bool Store::flushLocked(std::string* err) {
-  if (memtable_.empty()) return true;
+bool Store::syncReads(std::string* err) {
   auto t0 = std::chrono::steady_clock::now();
-  kbslab::DeltaRows d;
-  d.m = (int64_t)memtable_.size();
-  d.keys.resize((size_t)d.m * KEYW);
-  d.meta.reserve(d.m);
-  d.rev.reserve(d.m);
-  d.vo.reserve(d.m);
-  int64_t heap_base = slab_->heap_used();
-  int64_t i = 0;
-  for (auto& [ik, row] : memtable_) {
-    const Bytes& key = ik.first;
-    uint64_t rev = ik.second;
-    pad96(key, d.keys.data() + (size_t)i * KEYW);
-    bool ev = key.find(kEvents) != Bytes::npos;
-    if (rev == 0) {
-      uint32_t vlen = row.flag9 ? 9 : 8;
-      d.meta.push_back(kbslab::meta_make(false, row.flag9, ev, (uint32_t)key.size(), vlen));
-      d.rev.push_back(0);
-      d.vo.push_back(row.objrev);
-    } else {
-      d.meta.push_back(kbslab::meta_make(row.tomb, false, ev, (uint32_t)key.size(),
-                                         (uint32_t)row.val.size()));
-      d.rev.push_back(rev);
-      d.vo.push_back((uint64_t)(heap_base + (int64_t)d.heap.size()));
-      d.heap.insert(d.heap.end(), row.val.begin(), row.val.end());
-      d.heap.resize((d.heap.size() + 3) & ~3ull);  // 4B-aligned heap records
-    }
-    ++i;
+  if (!heap_pending_.empty()) {
+    int64_t off = 0;
+    if (!slab_->HeapAppend(heap_pending_.data(), (int64_t)heap_pending_.size(),
+                           &off, err))
+      return false;
+    if (off != heap_base_) { if (err) *err = "heap offset drift"; return false; }
+    heap_pending_.clear();
+    heap_base_ = -1;
   }
-  if (!slab_->Merge(d, err)) return false;
-  memtable_.clear();
-  host_merge_s_ += std::chrono::duration<double>(std::chrono::steady_clock::now() - t0).count();
+  if (!newrows_.empty()) {
+    std::sort(newrows_.begin(), newrows_.end(),
+              [](const NewRow& a, const NewRow& b) {
+                int c = a.key.compare(b.key);
+                if (c != 0) return c < 0;
+                return a.rev < b.rev;
+              });
+    size_t m = newrows_.size();
+    std::vector<uint8_t> keys(m * KEYW);
+    std::vector<uint64_t> meta(m), rev(m), vo(m);
+    for (size_t i = 0; i < m; ++i) {
+      pad96(newrows_[i].key, keys.data() + i * KEYW);
+      meta[i] = newrows_[i].meta;
+      rev[i] = newrows_[i].rev;
+      vo[i] = newrows_[i].vo;
+    }
+    if (!slab_->AppendRows(keys.data(), meta.data(), rev.data(), vo.data(),
+                           (int64_t)m, err))
+      return false;
+    newrows_.clear();
+    nr_revrow_.clear();
+  }
+  if (slab_->delta_rows() >= cfg_.flush_rows) {
+    if (!slab_->Fold(err)) return false;
+  }
+  host_merge_s_ +=
+      std::chrono::duration<double>(std::chrono::steady_clock::now() - t0).count();
   return true;
+}
+
+bool Store::foldLocked(std::string* err) {
+  if (!syncReads(err)) return false;
+  return slab_->Fold(err);
 }
 
 bool Store::Flush(std::string* err) {
   std::lock_guard<std::recursive_mutex> lk(mu_);
-  return flushLocked(err);
+  return foldLocked(err);
 }
 
 // ---- events / watch -----------------------------------------------------
@@ -369,7 +388,7 @@ WriteResponse Store::Create(const Bytes& key, const Bytes& value, Status* st) {
   uint64_t revision = deal(0, &dst);
   Status err = dst == OK ? createInternal(key, value, revision) : dst;
   notify(key, value, revision, 0, err == OK, Event::CREATE);
-  maybeFlush();
+  if (newrows_.size() >= 16384) { std::string e_; (void)syncReads(&e_); }
   if (err == CAS_FAILED) {
     resp.header_revision = revision;
     *st = OK;
@@ -382,39 +401,17 @@ WriteResponse Store::Create(const Bytes& key, const Bytes& value, Status* st) {
 }
 
 Status Store::get(const Bytes& key, uint64_t revision, Bytes* val, uint64_t* modRev) {
-  // range.go:82-121: largest object row (key,rev<=R); tombstone -> NOTFOUND
+  // range.go:82-121: largest object row (key,rev<=R); tombstone -> NOTFOUND.
+  // GPU point read over base+delta runs (range.go:91-121 reverse-iter
+  // semantics); pending writes are synced to the device delta run first.
   *modRev = 0;
+  std::string err;
+  if (!syncReads(&err)) return INTERNAL;
   uint64_t R = revision == 0 ? UINT64_MAX : revision;
-  // memtable candidates first: rows of key with 1 <= rev <= R (strictly newer
-  // than any slab row of the same key)
-  bool mem_hit = false;
-  Bytes mval;
-  uint64_t mrev = 0;
-  bool mtomb = false;
-  {
-    auto ub = memtable_.upper_bound({key, R});
-    if (ub != memtable_.begin()) {
-      auto c = std::prev(ub);
-      if (c->first.first == key && c->first.second >= 1) {
-        mem_hit = true;
-        mrev = c->first.second;
-        mval = c->second.val;
-        mtomb = c->second.tomb;
-      }
-    }
-  }
-  if (mem_hit) {
-    *modRev = mrev;
-    if (mtomb) return NOTFOUND;
-    *val = mval;
-    return OK;
-  }
-  // GPU point read (range.go:91-121 reverse-iter semantics)
   DevGetQ q;
   pad96(key, q.key);
   q.read_rev = R;
   std::vector<kbslab::GetResult> outs;
-  std::string err;
   if (!slab_->GetBatch({q}, &outs, &err)) return INTERNAL;
   if (!outs[0].found) return NOTFOUND;
   *modRev = outs[0].rev;
@@ -475,7 +472,7 @@ WriteResponse Store::Update(const Bytes& key, const Bytes& value,
     }
     notify(key, value, curRev, prevRev, err == OK, Event::PUT);
   }
-  maybeFlush();
+  if (newrows_.size() >= 16384) { std::string e_; (void)syncReads(&e_); }
   resp.header_revision = curRev;
   resp.succeeded = (err == OK);
   if (err == CAS_FAILED) {
@@ -538,7 +535,7 @@ WriteResponse Store::Delete(const Bytes& key, uint64_t prevRev, Status* st) {
     }
   }
   notify(key, old.value, newRevision, old.revision, err == OK, Event::DELETE);
-  maybeFlush();
+  if (newrows_.size() >= 16384) { std::string e_; (void)syncReads(&e_); }
   resp.header_revision = newRevision;
   resp.succeeded = (err == OK);
   if (err == CAS_FAILED) {
@@ -589,93 +586,36 @@ RangeResponse Store::List(const Bytes& start, const Bytes& end,
   Status cst = checkCompactRace(reqRevision);
   if (cst != OK) { *st = cst; return resp; }
   ops_range_++;
-
-  // memtable winners over [start, end): per key, max rev<=R (tombstones are
-  // deletion markers that override slab winners)
-  struct MemWinner { Bytes key, val; uint64_t rev; bool tomb; };
-  std::vector<MemWinner> memw;
-  int64_t mem_tombs = 0;
-  {
-    auto it = memtable_.lower_bound({start, 0});
-    auto endit = memtable_.lower_bound({end, 0});
-    const Bytes* curKey = nullptr;
-    for (; it != endit; ++it) {
-      uint64_t r = it->first.second;
-      if (r == 0 || r > reqRevision) continue;
-      if (!curKey || *curKey != it->first.first) {
-        memw.push_back(MemWinner{it->first.first, it->second.val, r, it->second.tomb});
-        curKey = &memw.back().key;
-      } else {  // ascending rev within key: later row wins
-        memw.back().val = it->second.val;
-        memw.back().rev = r;
-        memw.back().tomb = it->second.tomb;
-      }
-    }
-    for (auto& m : memw) if (m.tomb) mem_tombs++;
-  }
-
-  // GPU winners, fetched in chunks; tombstone overrides may consume up to
-  // mem_tombs extra, so over-ask. Continuation bound trick: winner_key+0x01
-  // excludes only the winner itself (key bytes <= 0x24 are rejected), so the
-  // next chunk resumes exactly after it.
   std::string err;
+  if (!syncReads(&err)) { *st = INTERNAL; return resp; }
+
+  // GPU winners over base+delta (merged device-side), fetched in chunks.
+  // Continuation bound trick: winner_key+0x01 excludes only the winner itself
+  // (key bytes <= 0x24 are rejected), so the next chunk resumes after it.
   std::vector<kbslab::RangeResult> outs;
-  std::vector<kbslab::RangeResult::Rec> dev_recs;
-  size_t di = 0, mi = 0;
-  bool dev_more = true;  // more device winners may exist past dev_recs
+  std::vector<KeyValue> kvs;
+  bool dev_more = true;
   Bytes dev_frontier = start;
-  auto fetch_dev = [&](int64_t want) -> bool {
+  auto need = [&]() { return lim <= 0 || (int64_t)kvs.size() < lim; };
+  while (need() && dev_more) {
     DevRangeQ q{};
     memset(q.start, 0, KEYW);
     memcpy(q.start, dev_frontier.data(), std::min(dev_frontier.size(), (size_t)KEYW));
     pad96(end, q.end);
     q.read_rev = reqRevision;
-    q.cap = want;
+    q.cap = lim > 0 ? lim - (int64_t)kvs.size() : 0;
     q.count_only = 0;
-    if (!slab_->RangeBatch({q}, true, &outs, &err)) return false;
+    if (!slab_->RangeBatch({q}, true, &outs, &err)) { *st = INTERNAL; return resp; }
     kbslab::RangeResult& r = outs[0];
-    if (r.overflow) return false;
-    for (auto& rec : r.recs) dev_recs.push_back(std::move(rec));
-    // more winners beyond this chunk iff the scan stopped at its cap
-    dev_more = (want > 0 && r.total >= want) ||
-               (want <= 0 && r.total > r.written);
-    if (!dev_recs.empty())
-      dev_frontier = dev_recs.back().key + Bytes("\x01", 1);
-    return true;
-  };
-  if (!fetch_dev(lim > 0 ? lim + mem_tombs : 0)) {
-    *st = outs.empty() || !outs[0].overflow ? INTERNAL : NOBUF;
-    return resp;
-  }
-
-  // ordered merge: device winners × memtable winners (override semantics)
-  std::vector<KeyValue> kvs;
-  auto need = [&]() { return lim <= 0 || (int64_t)kvs.size() < lim; };
-  while (need()) {
-    if (di >= dev_recs.size() && dev_more) {
-      int64_t want = lim > 0 ? (lim - (int64_t)kvs.size()) + mem_tombs : 0;
-      size_t before = dev_recs.size();
-      if (!fetch_dev(want)) { *st = INTERNAL; return resp; }
-      if (dev_recs.size() == before && !dev_more) {}  // exhausted
-      if (dev_recs.size() == before && dev_more) break;  // defensive
+    if (r.overflow) { *st = NOBUF; return resp; }
+    for (auto& rec : r.recs) {
+      if (!need()) break;
+      kvs.push_back(KeyValue{rec.key, rec.val, rec.rev});
     }
-    bool have_d = di < dev_recs.size();
-    bool have_m = mi < memw.size();
-    if (!have_d && !have_m) break;
-    if (!have_d && dev_more) continue;  // fetch more before consuming memw tail
-    if (have_d && (!have_m || dev_recs[di].key < memw[mi].key)) {
-      kvs.push_back(KeyValue{dev_recs[di].key, dev_recs[di].val, dev_recs[di].rev});
-      di++;
-    } else if (have_m && (!have_d || memw[mi].key < dev_recs[di].key)) {
-      if (!memw[mi].tomb)
-        kvs.push_back(KeyValue{memw[mi].key, memw[mi].val, memw[mi].rev});
-      mi++;
-    } else {  // same key: memtable rows are newer -> override
-      if (!memw[mi].tomb)
-        kvs.push_back(KeyValue{memw[mi].key, memw[mi].val, memw[mi].rev});
-      di++;
-      mi++;
-    }
+    dev_more = (q.cap > 0 && r.written >= q.cap) ||
+               (q.cap <= 0 && r.total > r.written);
+    if (r.recs.empty()) break;  // no progress => exhausted
+    dev_frontier = r.recs.back().key + Bytes("\x01", 1);
   }
 
   resp.header_revision = curRevision;
@@ -701,7 +641,7 @@ CountResponse Store::Count(const Bytes& start, const Bytes& end, Status* st) {
   Status cst = checkCompactRace(rev);
   if (cst != OK) { *st = cst; return resp; }
   std::string err;
-  if (!flushLocked(&err)) { *st = INTERNAL; return resp; }
+  if (!syncReads(&err)) { *st = INTERNAL; return resp; }
   DevRangeQ q{};
   pad96(start, q.start);
   pad96(end, q.end);
@@ -741,7 +681,7 @@ uint64_t Store::Compact(uint64_t revision, Status* st) {
     compact_cell_ = revision;
   }
   std::string err;
-  if (!flushLocked(&err)) { *st = INTERNAL; return revision; }
+  if (!foldLocked(&err)) { *st = INTERNAL; return revision; }
   pumpEvents();
   // borders (compact.go:108-127)
   std::vector<Bytes> prefixes;
@@ -800,7 +740,7 @@ uint64_t Store::Compact(uint64_t revision, Status* st) {
 
 bool Store::DumpStore(std::vector<std::pair<Bytes, Bytes>>* out, std::string* err) {
   std::lock_guard<std::recursive_mutex> lk(mu_);
-  if (!flushLocked(err)) return false;
+  if (!foldLocked(err)) return false;
   std::vector<kbslab::DumpRow> rows;
   if (!slab_->Dump(&rows, err)) return false;
   out->clear();
@@ -855,7 +795,7 @@ bool Store::BenchRange(const uint8_t* qbuf, size_t nq, bool d2h,
                        unsigned long long* total, double* secs, std::string* err) {
   // the measured hot path: batched List semantics with inputs resident in HBM
   std::lock_guard<std::recursive_mutex> lk(mu_);
-  if (!flushLocked(err)) return false;
+  if (!syncReads(err)) return false;
   struct Q { Bytes s, e; uint64_t rev; int64_t limit; };
   std::vector<Q> qs;
   qs.reserve(nq);

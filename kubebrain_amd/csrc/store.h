@@ -108,9 +108,11 @@ class Store {
   Status checkCompactRace(uint64_t revision);
   void putRow(const Bytes& key, uint64_t rev, const Bytes& val);
   void putRevRow(const Bytes& key, uint64_t objrev, bool flag9);
-  void maybeFlush();
   void pumpEvents();  // fan-out pending events via the GPU filter
-  bool flushLocked(std::string* err);
+  // push pending values + new rows to the device (delta-run merge); folds the
+  // delta into the base run past the fold threshold
+  bool syncReads(std::string* err);
+  bool foldLocked(std::string* err);  // syncReads + fold delta into base
   uint64_t getTimeoutRevision();
 
   Config cfg_;
@@ -123,8 +125,13 @@ class Store {
   std::unordered_set<Bytes> tombstoned_;  // keys with flagged rev-rows
   std::unordered_set<Bytes> events_keys_; // keys containing "/events/"
 
-  struct MemRow { Bytes val; bool tomb = false; bool flag9 = false; uint64_t objrev = 0; };
-  std::map<std::pair<Bytes, uint64_t>, MemRow> memtable_;  // (userKey, rev)
+  // rows written since the last device sync: values are
+  // staged in heap_pending_ with pre-assigned absolute heap offsets
+  struct NewRow { Bytes key; uint64_t rev, meta, vo; };
+  std::vector<NewRow> newrows_;
+  std::unordered_map<Bytes, size_t> nr_revrow_;  // key -> index of rev-row
+  Bytes heap_pending_;
+  int64_t heap_base_ = -1;
 
   bool compact_cell_set_ = false;
   uint64_t compact_cell_ = 0;
