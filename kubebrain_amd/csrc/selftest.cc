@@ -81,7 +81,7 @@ int main() {
         d.rev.push_back(rows[i].rev);
         d.vo.push_back((uint64_t)d.heap.size());
         d.heap.insert(d.heap.end(), rows[i].v.begin(), rows[i].v.end());
-        d.heap.resize((d.heap.size() + 3) & ~3ull);
+        d.heap.resize((d.heap.size() + 15) & ~15ull);
       }
     }
     STEP("merge_initial", s->Merge(d, &err));
@@ -162,7 +162,7 @@ int main() {
     d.meta.push_back(meta_make(false, false, false, 16, 3));
     d.rev.push_back(9);
     d.vo.push_back((uint64_t)s->heap_used());
-    d.heap = {'v', 'a', '9', 0};
+    d.heap = {'v', 'a', '9', 0, 0, 0, 0, 0, 0, 0, 0, 0, 0, 0, 0, 0};
     STEP("merge_replace", s->Merge(d, &err));
     std::vector<DumpRow> rows;
     STEP("dump", s->Dump(&rows, &err));

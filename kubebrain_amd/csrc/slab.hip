@@ -371,17 +371,18 @@ __global__ void k_get2(const uint8_t* __restrict__ bkeys,
   if (vlen > slot) return;
   const uint8_t* vs = heap + vo[row];
   uint8_t* vd = out + (int64_t)q * slot;
-  uint32_t words = vlen >> 2;
-  for (uint32_t b = lane; b < words; b += 64)
-    ((uint32_t*)vd)[b] = ((const uint32_t*)vs)[b];
+  uint32_t w16 = vlen >> 4;
+  for (uint32_t b = lane; b < w16; b += 64)
+    ((uint4*)vd)[b] = ((const uint4*)vs)[b];
   if (lane == 0)
-    for (uint32_t b = words * 4; b < vlen; ++b) vd[b] = vs[b];
+    for (uint32_t b = w16 * 16; b < vlen; ++b) vd[b] = vs[b];
 }
 
 // ---- gather: winners -> packed records in the per-query device arena ----
 // record: u64 rev | u32 klen | u32 vlen | key (pad8) | val (pad8)
 __device__ __forceinline__ int64_t rec_bytes(uint32_t klen, uint32_t vlen) {
-  return 16 + ((klen + 7) & ~7u) + ((vlen + 7) & ~7u);
+  // 16B-aligned record: header | key pad16 | value pad16 (uint4 copies)
+  return 16 + ((klen + 15) & ~15u) + ((vlen + 15) & ~15u);
 }
 
 __global__ void k_gather(const uint8_t* __restrict__ bkeys,
@@ -457,12 +458,12 @@ __global__ void k_gather(const uint8_t* __restrict__ bkeys,
     uint8_t* kd = dst + 16;
     for (uint32_t b = lane; b < klen; b += 64) kd[b] = ks[b];
     const uint8_t* vs = heap + (isd ? dvo : bvo)[row];
-    uint8_t* vd = dst + 16 + ((klen + 7) & ~7u);
-    uint32_t words = vlen >> 2;
-    for (uint32_t b = lane; b < words; b += 64)
-      ((uint32_t*)vd)[b] = ((const uint32_t*)vs)[b];
+    uint8_t* vd = dst + 16 + ((klen + 15) & ~15u);
+    uint32_t w16 = vlen >> 4;  // heap records and dst are 16B-aligned
+    for (uint32_t b = lane; b < w16; b += 64)
+      ((uint4*)vd)[b] = ((const uint4*)vs)[b];
     if (lane == 0)
-      for (uint32_t b = words * 4; b < vlen; ++b) vd[b] = vs[b];
+      for (uint32_t b = w16 * 16; b < vlen; ++b) vd[b] = vs[b];
   }
 }
 
@@ -474,9 +475,9 @@ __global__ void k_pack(const uint8_t* __restrict__ gbuf, int64_t qcap,
   int q = blockIdx.x;
   if (q >= nq) return;
   int64_t bytes = gbytes[q];
-  const uint32_t* src = (const uint32_t*)(gbuf + (int64_t)q * qcap);
-  uint32_t* dst = (uint32_t*)(out + goffs[q]);
-  int64_t words = bytes / 4;  // records are 8B-aligned so bytes%4==0... bytes%8==0
+  const uint4* src = (const uint4*)(gbuf + (int64_t)q * qcap);
+  uint4* dst = (uint4*)(out + goffs[q]);
+  int64_t words = bytes / 16;  // records are 16B-aligned (goffs too)
   for (int64_t w = threadIdx.x; w < words; w += blockDim.x) dst[w] = src[w];
 }
 
@@ -596,7 +597,7 @@ __global__ void k_heap_sizes(const uint64_t* __restrict__ keep,
                              uint64_t* __restrict__ sz, int64_t n) {
   int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
   if (i >= n) return;
-  sz[i] = (keep[i] && rev[i] > 0) ? ((meta_vlen(meta[i]) + 3) & ~3ull) : 0;
+  sz[i] = (keep[i] && rev[i] > 0) ? ((meta_vlen(meta[i]) + 15) & ~15ull) : 0;
 }
 
 __global__ void k_compact_scatter(const uint8_t* __restrict__ keysA,
@@ -632,10 +633,10 @@ __global__ void k_heap_scatter(const uint64_t* __restrict__ keep,
   int64_t i = (int64_t)blockIdx.x * (blockDim.x / 64) + (threadIdx.x >> 6);
   int lane = threadIdx.x & 63;
   if (i >= n || !keep[i] || rev[i] == 0) return;
-  uint64_t bytes = (meta_vlen(meta[i]) + 3) & ~3ull;
-  const uint32_t* src = (const uint32_t*)(heapA + voA[i]);
-  uint32_t* dst = (uint32_t*)(heapB + heap_off[i]);
-  for (uint64_t w = lane; w < bytes / 4; w += 64) dst[w] = src[w];
+  uint64_t bytes = (meta_vlen(meta[i]) + 15) & ~15ull;
+  const uint4* src = (const uint4*)(heapA + voA[i]);
+  uint4* dst = (uint4*)(heapB + heap_off[i]);
+  for (uint64_t w = lane; w < bytes / 16; w += 64) dst[w] = src[w];
 }
 
 // ---- merge (memtable flush): delta ranks + scatter ----------------------
@@ -795,8 +796,17 @@ struct Slab::Impl {
 
   hipEvent_t ev0 = nullptr, ev1 = nullptr, ev2 = nullptr, ev3 = nullptr;
 
-  // host staging
-  std::vector<uint8_t> h_pack;
+  // host staging (pinned for fast D2H)
+  uint8_t* h_pack = nullptr;
+  int64_t h_pack_cap = 0;
+  bool ensure_hpack(int64_t need, std::string* err) {
+    if (need <= h_pack_cap) return true;
+    if (h_pack) (void)hipHostFree(h_pack);
+    int64_t cap = need + need / 2;
+    HIP_CHECK(hipHostMalloc(&h_pack, cap));
+    h_pack_cap = cap;
+    return true;
+  }
 
   ~Impl() {
     for (void* p : {(void*)A.keys, (void*)A.meta, (void*)A.rev, (void*)A.vo,
@@ -818,6 +828,7 @@ struct Slab::Impl {
                     (void*)d_ekeys, (void*)d_erev, (void*)d_bitmap}) {
       if (p) (void)hipFree(p);
     }
+    if (h_pack) (void)hipHostFree(h_pack);
     if (ev0) (void)hipEventDestroy(ev0);
     if (ev1) (void)hipEventDestroy(ev1);
     if (ev2) (void)hipEventDestroy(ev2);
@@ -1120,7 +1131,7 @@ bool Slab::RangeBatch(const std::vector<DevRangeQ>& qs, bool d2h,
   if (nq > I->max_q) { if (err) *err = "too many queries per batch (KB_MAX_Q)"; return false; }
   outs->assign(nq, RangeResult());
   int64_t qcap = I->arena_bytes / nq;
-  qcap &= ~7ll;
+  qcap &= ~15ll;
   HIP_CHECK(hipMemcpyAsync(I->d_qs, qs.data(), sizeof(DevRangeQ) * nq,
                            hipMemcpyHostToDevice, I->stream));
   HIP_CHECK(hipMemsetAsync(I->d_scanned, 0, 8, I->stream));
@@ -1179,9 +1190,9 @@ bool Slab::RangeBatch(const std::vector<DevRangeQ>& qs, bool d2h,
                            hipMemcpyHostToDevice, I->stream));
   hipLaunchKernelGGL(k_pack, dim3(nq), dim3(256), 0, I->stream, I->d_gbuf, qcap,
                      I->d_gbytes, I->d_goffs, I->d_pack, nq);
-  if ((int64_t)I->h_pack.size() < acc) I->h_pack.resize(acc);
+  if (!I->ensure_hpack(acc, err)) return false;
   if (acc > 0)
-    HIP_CHECK(hipMemcpyAsync(I->h_pack.data(), I->d_pack, acc,
+    HIP_CHECK(hipMemcpyAsync(I->h_pack, I->d_pack, acc,
                              hipMemcpyDeviceToHost, I->stream));
   HIP_CHECK(hipEventRecord(I->ev1, I->stream));
   HIP_CHECK(hipStreamSynchronize(I->stream));
@@ -1190,7 +1201,7 @@ bool Slab::RangeBatch(const std::vector<DevRangeQ>& qs, bool d2h,
   for (int q = 0; q < nq; ++q) {
     RangeResult& r = (*outs)[q];
     if (r.overflow) continue;
-    const uint8_t* pp = I->h_pack.data() + goffs[q];
+    const uint8_t* pp = I->h_pack + goffs[q];
     r.recs.reserve(r.written);
     for (int64_t j = 0; j < r.written; ++j) {
       uint64_t rv;
@@ -1201,9 +1212,9 @@ bool Slab::RangeBatch(const std::vector<DevRangeQ>& qs, bool d2h,
       RangeResult::Rec rec;
       rec.rev = rv;
       rec.key.assign((const char*)pp + 16, klen);
-      rec.val.assign((const char*)pp + 16 + pad8(klen), vlen);
+      rec.val.assign((const char*)pp + 16 + ((klen + 15) & ~15u), vlen);
       r.recs.push_back(std::move(rec));
-      pp += 16 + pad8(klen) + pad8(vlen);
+      pp += 16 + ((klen + 15) & ~15u) + ((vlen + 15) & ~15u);
     }
   }
   return true;
@@ -1217,7 +1228,7 @@ bool Slab::GetBatch(const std::vector<DevGetQ>& qs, std::vector<GetResult>* outs
   if (nq == 0) return true;
   if (nq > I->max_q) { if (err) *err = "too many gets per batch"; return false; }
   int64_t slot = I->arena_bytes / nq;
-  slot &= ~7ll;
+  slot &= ~15ll;
   HIP_CHECK(hipMemcpyAsync(I->d_gq, qs.data(), sizeof(DevGetQ) * nq,
                            hipMemcpyHostToDevice, I->stream));
   HIP_CHECK(hipEventRecord(I->ev0, I->stream));

@@ -135,7 +135,7 @@ void Store::putRow(const Bytes& key, uint64_t rev, const Bytes& val) {
   if (heap_base_ < 0) heap_base_ = slab_->heap_used();
   r.vo = (uint64_t)(heap_base_ + (int64_t)heap_pending_.size());
   heap_pending_ += val;
-  heap_pending_.resize((heap_pending_.size() + 3) & ~3ull, '\0');
+  heap_pending_.resize((heap_pending_.size() + 15) & ~15ull, '\0');
   newrows_.push_back(std::move(r));
 }
 
