@@ -121,6 +121,55 @@ __device__ int64_t d_lb_range(const uint8_t* keys, const uint64_t* rev,
 constexpr uint64_t ROW_TAG_DELTA = 1ull << 63;  // winner row lives in the delta run
 constexpr uint64_t ROW_MASK = ROW_TAG_DELTA - 1;
 
+// two-level search: fence[i] = (key,rev) of row i*256; the fence (~96B per
+// 256 rows) stays L2/L3-resident, so a cold search costs ~1 HBM window
+// instead of log2(n) dependent misses.
+constexpr int64_t FENCE_STRIDE = 256;
+
+__global__ void k_build_fence(const uint8_t* __restrict__ keys,
+                              const uint64_t* __restrict__ rev, int64_t n,
+                              uint8_t* __restrict__ fkeys,
+                              uint64_t* __restrict__ frev, int64_t nf) {
+  int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  if (i >= nf) return;
+  int64_t row = i * FENCE_STRIDE;
+  const uint64_t* ks = (const uint64_t*)(keys + row * KEYW);
+  uint64_t* kd = (uint64_t*)(fkeys + i * KEYW);
+#pragma unroll
+  for (int k = 0; k < KEYW / 8; ++k) kd[k] = ks[k];
+  frev[i] = rev[row];
+}
+
+__device__ int64_t d_lb_fenced(const uint8_t* keys, const uint64_t* rev,
+                               int64_t n, const uint8_t* fkeys,
+                               const uint64_t* frev, int64_t nf,
+                               const uint8_t* qkey, uint64_t qrev) {
+  if (nf <= 0) {
+    int64_t lo = 0, hi = n;
+    while (lo < hi) {
+      int64_t mid = (lo + hi) >> 1;
+      int c = keycmp96(keys + mid * KEYW, qkey);
+      if (c < 0 || (c == 0 && rev[mid] < qrev)) lo = mid + 1; else hi = mid;
+    }
+    return lo;
+  }
+  int64_t lo = 0, hi = nf;
+  while (lo < hi) {  // first fence entry >= q
+    int64_t mid = (lo + hi) >> 1;
+    int c = keycmp96(fkeys + mid * KEYW, qkey);
+    if (c < 0 || (c == 0 && frev[mid] < qrev)) lo = mid + 1; else hi = mid;
+  }
+  int64_t rlo = lo == 0 ? 0 : (lo - 1) * FENCE_STRIDE + 1;
+  int64_t rhi = lo < nf ? lo * FENCE_STRIDE + 1 : n;  // result may be fence row itself
+  if (rhi > n) rhi = n;
+  while (rlo < rhi) {
+    int64_t mid = (rlo + rhi) >> 1;
+    int c = keycmp96(keys + mid * KEYW, qkey);
+    if (c < 0 || (c == 0 && rev[mid] < qrev)) rlo = mid + 1; else rhi = mid;
+  }
+  return rlo;
+}
+
 // ---- the north-star kernel: batched MVCC range scan --------------------
 // One workgroup per query. Winner predicate (scanner.go:389-516, DESIGN.md
 // §3.1): rev>0 ∧ rev<=R ∧ ¬tomb ∧ (¬same_next ∨ next.rev>R). Reads 16B/row
@@ -259,7 +308,8 @@ __global__ void k_range_scan2(
     const uint64_t* __restrict__ brev, int64_t n,
     const uint8_t* __restrict__ dkeys, const uint64_t* __restrict__ dmeta,
     const uint64_t* __restrict__ drev, int64_t dn,
-    const DevRangeQ* __restrict__ qs, int nq, int64_t max_cap,
+    const uint8_t* __restrict__ fkeys, const uint64_t* __restrict__ frev,
+    int64_t nf, const DevRangeQ* __restrict__ qs, int nq, int64_t max_cap,
     uint64_t* __restrict__ rows_b, uint64_t* __restrict__ rows_d,
     uint64_t* __restrict__ rows_m, int64_t* __restrict__ found_out,
     int64_t* __restrict__ total_out, unsigned long long* __restrict__ scanned_out) {
@@ -268,8 +318,8 @@ __global__ void k_range_scan2(
   __shared__ int64_t lo_s, hi_s, dlo_s, dhi_s, cnt_s;
   __shared__ int wave_cnt[4];
   const DevRangeQ& Q = qs[q];
-  if (threadIdx.x == 0) lo_s = d_lb_range(bkeys, brev, 0, n, Q.start, 0);
-  if (threadIdx.x == 64) hi_s = d_lb_range(bkeys, brev, 0, n, Q.end, 0);
+  if (threadIdx.x == 0) lo_s = d_lb_fenced(bkeys, brev, n, fkeys, frev, nf, Q.start, 0);
+  if (threadIdx.x == 64) hi_s = d_lb_fenced(bkeys, brev, n, fkeys, frev, nf, Q.end, 0);
   if (threadIdx.x == 128) dlo_s = d_lb_range(dkeys, drev, 0, dn, Q.start, 0);
   if (threadIdx.x == 192) dhi_s = d_lb_range(dkeys, drev, 0, dn, Q.end, 0);
   __syncthreads();
@@ -321,6 +371,8 @@ __global__ void k_get2(const uint8_t* __restrict__ bkeys,
                        const uint64_t* __restrict__ dmeta,
                        const uint64_t* __restrict__ drev,
                        const uint64_t* __restrict__ dvo, int64_t n, int64_t dn,
+                       const uint8_t* __restrict__ fkeys,
+                       const uint64_t* __restrict__ frev, int64_t nf,
                        const uint8_t* __restrict__ heap,
                        const DevGetQ* __restrict__ qs, int nq,
                        uint8_t* __restrict__ out, int64_t slot,
@@ -344,7 +396,11 @@ __global__ void k_get2(const uint8_t* __restrict__ bkeys,
       }
     }
     if (row < 0) {
-      int64_t ub = d_upper_bound(bkeys, brev, n, Q.key, Q.read_rev);
+      // upper_bound(key,R) == lower_bound(key,R+1) (R < UINT64_MAX-1 here or
+      // saturates safely: rev values are far below UINT64_MAX)
+      uint64_t nr = Q.read_rev == UINT64_MAX ? UINT64_MAX : Q.read_rev + 1;
+      int64_t ub = nr == UINT64_MAX ? d_upper_bound(bkeys, brev, n, Q.key, Q.read_rev)
+                                    : d_lb_fenced(bkeys, brev, n, fkeys, frev, nf, Q.key, nr);
       if (ub > 0) {
         int64_t c = ub - 1;
         if (brev[c] >= 1 && keycmp96(bkeys + c * KEYW, Q.key) == 0) row = c;
@@ -762,6 +818,9 @@ struct Slab::Impl {
   int64_t arena_bytes = 384ll << 20;
   DevRangeQ* d_qs = nullptr;
   DevGetQ* d_gq = nullptr;
+  uint8_t* d_fkeys = nullptr;   // fence index over the base run
+  uint64_t* d_frev = nullptr;
+  int64_t nf = 0;
   uint64_t* d_rows = nullptr;   // max_q*max_cap (base winners)
   uint64_t* d_rows2 = nullptr;  // delta winners
   uint64_t* d_rowsm = nullptr;  // merged winners
@@ -813,7 +872,7 @@ struct Slab::Impl {
                     (void*)B.keys, (void*)B.meta, (void*)B.rev, (void*)B.vo,
                     (void*)DA.keys, (void*)DA.meta, (void*)DA.rev, (void*)DA.vo,
                     (void*)DB.keys, (void*)DB.meta, (void*)DB.rev, (void*)DB.vo,
-                    (void*)d_rows2, (void*)d_rowsm,
+                    (void*)d_rows2, (void*)d_rowsm, (void*)d_fkeys, (void*)d_frev,
                     (void*)heapA, (void*)heapB, (void*)s_a, (void*)s_b,
                     (void*)s_c, (void*)s_d, (void*)lv1, (void*)lv1o, (void*)lv2,
                     (void*)lv2o, (void*)lv3, (void*)lv3o, (void*)d_qs,
@@ -885,6 +944,14 @@ struct Slab::Impl {
     HIP_CHECK(hipMalloc(&d_dvo, cap * 8));
     upload_cap = cap;
     return true;
+  }
+
+  void rebuildFence(const Col& base, int64_t n) {
+    nf = n / FENCE_STRIDE + (n % FENCE_STRIDE ? 1 : 0);
+    if (nf == 0) return;
+    hipLaunchKernelGGL(k_build_fence, dim3((uint32_t)ceil_div(nf, 256)),
+                       dim3(256), 0, stream, base.keys, base.rev, n, d_fkeys,
+                       d_frev, nf);
   }
 
   // generic two-run merge: src (n rows) + newer (m rows, device arrays,
@@ -984,6 +1051,8 @@ Slab* Slab::Create(int64_t max_rows, int64_t heap_cap, int device,
   HIP_CHECK_NULL(hipMalloc(&I->d_qs, sizeof(DevRangeQ) * I->max_q));
   HIP_CHECK_NULL(hipMalloc(&I->d_gq, sizeof(DevGetQ) * I->max_q));
   HIP_CHECK_NULL(hipMalloc(&I->d_rows, (int64_t)I->max_q * I->max_cap * 8));
+  HIP_CHECK_NULL(hipMalloc(&I->d_fkeys, (max_rows / FENCE_STRIDE + 2) * KEYW));
+  HIP_CHECK_NULL(hipMalloc(&I->d_frev, (max_rows / FENCE_STRIDE + 2) * 8));
   HIP_CHECK_NULL(hipMalloc(&I->d_rows2, (int64_t)I->max_q * I->max_cap * 8));
   HIP_CHECK_NULL(hipMalloc(&I->d_rowsm, (int64_t)I->max_q * I->max_cap * 8));
   HIP_CHECK_NULL(hipMalloc(&I->d_offs, (int64_t)I->max_q * I->max_cap * 8));
@@ -1077,6 +1146,8 @@ bool Slab::Fold(std::string* err) {
   std::swap(I->A, I->B);
   I->n = new_n;
   I->dn = 0;
+  I->rebuildFence(I->A, I->n);
+  HIP_CHECK(hipStreamSynchronize(I->stream));
   return true;
 }
 
@@ -1113,6 +1184,7 @@ bool Slab::Merge(const DeltaRows& d, std::string* err) {
       return false;
     std::swap(I->A, I->B);
     I->n = new_n;
+    I->rebuildFence(I->A, I->n);
   }
   HIP_CHECK(hipEventRecord(I->ev1, I->stream));
   HIP_CHECK(hipStreamSynchronize(I->stream));
@@ -1139,7 +1211,8 @@ bool Slab::RangeBatch(const std::vector<DevRangeQ>& qs, bool d2h,
   HIP_CHECK(hipEventRecord(I->ev0, I->stream));
   hipLaunchKernelGGL(k_range_scan2, dim3(nq), dim3(256), 0, I->stream,
                      I->A.keys, I->A.meta, I->A.rev, I->n, I->DA.keys,
-                     I->DA.meta, I->DA.rev, I->dn, I->d_qs, nq, I->max_cap,
+                     I->DA.meta, I->DA.rev, I->dn, I->d_fkeys, I->d_frev,
+                     I->nf, I->d_qs, nq, I->max_cap,
                      I->d_rows, I->d_rows2, I->d_rowsm, I->d_found, I->d_total,
                      I->d_scanned);
   HIP_CHECK(hipEventRecord(I->ev1, I->stream));
@@ -1235,7 +1308,8 @@ bool Slab::GetBatch(const std::vector<DevGetQ>& qs, std::vector<GetResult>* outs
   int blocks = (int)ceil_div(nq, 4);
   hipLaunchKernelGGL(k_get2, dim3(blocks), dim3(256), 0, I->stream, I->A.keys,
                      I->A.meta, I->A.rev, I->A.vo, I->DA.keys, I->DA.meta,
-                     I->DA.rev, I->DA.vo, I->n, I->dn, I->heapA, I->d_gq, nq,
+                     I->DA.rev, I->DA.vo, I->n, I->dn, I->d_fkeys, I->d_frev,
+                     I->nf, I->heapA, I->d_gq, nq,
                      I->d_gbuf, slot, I->d_orev, I->d_ometa, I->d_found32,
                      I->d_ovf);
   HIP_CHECK(hipEventRecord(I->ev1, I->stream));
@@ -1333,6 +1407,8 @@ bool Slab::Compact(const std::vector<std::pair<Bound, Bound>>& borders,
   std::swap(I->heapA, I->heapB);
   I->n = (int64_t)kept;
   I->heap_used_ = (int64_t)new_heap;
+  I->rebuildFence(I->A, I->n);
+  HIP_CHECK(hipStreamSynchronize(I->stream));
   return true;
 }
 
