@@ -52,6 +52,13 @@ def make_keys(nns, per_ns):
 
 def build_store(store, namespaces, keys, rng, extra_revs, tomb_frac, my_ns=None):
     """Load the synthetic keyspace through the normal write path."""
+    t_phase = time.time()
+
+    def phase(name):
+        nonlocal t_phase
+        print(f"[load] {name}: {time.time()-t_phase:.1f}s", file=sys.stderr, flush=True)
+        t_phase = time.time()
+
     if my_ns is not None:
         sel = [k for k in keys if k.rsplit(b"/", 1)[0] in my_ns]
     else:
@@ -62,9 +69,11 @@ def build_store(store, namespaces, keys, rng, extra_revs, tomb_frac, my_ns=None)
     klens = np.array([len(k) for k in sel], dtype=np.uint32).tobytes()
     vlens = np.full(n, VAL_LEN, dtype=np.uint32).tobytes()
     kblob = b"".join(sel)
+    phase("gen values/keys")
     f = store._f("bulk_create")
     rc = f(ctypes.c_void_p(store.h), kblob, klens, vals, vlens, ctypes.c_size_t(n))
     assert rc == 0, "bulk_create failed"
+    phase("bulk_create")
     revs = {k: None for k in sel}  # latest rev per key tracked client-side
     base = store.current_rev() - n
     for i, k in enumerate(sel):
@@ -86,6 +95,7 @@ def build_store(store, namespaces, keys, rng, extra_revs, tomb_frac, my_ns=None)
                 done += 1
             txn_batch(store, [(k, revs[k], vbuf) for k in uniq], revs,
                       must_succeed=True)
+        phase("extra revisions")
     # tombstones
     if tomb_frac and n:
         nt = int(n * tomb_frac)
@@ -95,7 +105,9 @@ def build_store(store, namespaces, keys, rng, extra_revs, tomb_frac, my_ns=None)
             r = store.delete(k, revs[k])
             assert r.succeeded
             revs[k] = None
+        phase("tombstones")
     store._f("flush")(ctypes.c_void_p(store.h))
+    phase("final fold")
     return sel, revs
 
 
@@ -295,9 +307,10 @@ def main():
     os.environ.setdefault("KB_HEAP_BYTES",
                           str(int((args.nns * args.per_ns / world + args.extra_revs)
                                   * (VAL_LEN + 16) * 1.4) + (128 << 20)))
-    # large threshold: loading flushes a few times; during the bench the
-    # memtable is flushed at every kb_bench_range call anyway (timed)
+    # delta-run fold threshold; syncs happen at every read batch (timed)
     os.environ.setdefault("KB_FLUSH_ROWS", "262144")
+    # the unbounded host event log is for parity tests, not the bench
+    os.environ.setdefault("KB_EVENT_LOG", "0")
 
     import kubebrain_amd
     import torch

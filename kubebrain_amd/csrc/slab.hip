@@ -495,30 +495,33 @@ __global__ void k_gather(const uint8_t* __restrict__ bkeys,
   }
   __syncthreads();
   if (base_s > qcap) return;
-  // phase B: one wave per record
+  // phase B: 16-lane record groups — 4 records per wave issue their load
+  // chains together (memory-level parallelism across records)
   int lane = threadIdx.x & 63, w = threadIdx.x >> 6;
+  int grp = lane >> 4, gl = lane & 15;
   uint8_t* qb = gbuf + (int64_t)q * qcap;
-  for (int64_t j = w; j < nwin; j += blockDim.x / 64) {
+  int64_t stride = (blockDim.x / 64) * 4;
+  for (int64_t j = (int64_t)w * 4 + grp; j < nwin; j += stride) {
     uint64_t rt = rows[j];
     bool isd = (rt & ROW_TAG_DELTA) != 0;
     int64_t row = (int64_t)(rt & ROW_MASK);
     uint64_t m = (isd ? dmeta : bmeta)[row];
     uint32_t klen = meta_klen(m), vlen = meta_vlen(m);
     uint8_t* dst = qb + qoffs[j];
-    if (lane == 0) {
+    if (gl == 0) {
       *(uint64_t*)dst = (isd ? drev : brev)[row];
       ((uint32_t*)dst)[2] = klen;
       ((uint32_t*)dst)[3] = vlen;
     }
     const uint8_t* ks = (isd ? dkeys : bkeys) + row * KEYW;
     uint8_t* kd = dst + 16;
-    for (uint32_t b = lane; b < klen; b += 64) kd[b] = ks[b];
+    for (uint32_t b = gl; b < klen; b += 16) kd[b] = ks[b];
     const uint8_t* vs = heap + (isd ? dvo : bvo)[row];
     uint8_t* vd = dst + 16 + ((klen + 15) & ~15u);
-    uint32_t w16 = vlen >> 4;  // heap records and dst are 16B-aligned
-    for (uint32_t b = lane; b < w16; b += 64)
+    uint32_t w16 = vlen >> 4;
+    for (uint32_t b = gl; b < w16; b += 16)
       ((uint4*)vd)[b] = ((const uint4*)vs)[b];
-    if (lane == 0)
+    if (gl == 0)
       for (uint32_t b = w16 * 16; b < vlen; ++b) vd[b] = vs[b];
   }
 }
@@ -1197,6 +1200,11 @@ bool Slab::Merge(const DeltaRows& d, std::string* err) {
 
 bool Slab::RangeBatch(const std::vector<DevRangeQ>& qs, bool d2h,
                       std::vector<RangeResult>* outs, std::string* err) {
+  return RangeBatchEx(qs, d2h, true, outs, err);
+}
+
+bool Slab::RangeBatchEx(const std::vector<DevRangeQ>& qs, bool d2h, bool parse,
+                        std::vector<RangeResult>* outs, std::string* err) {
   Impl* I = p;
   int nq = (int)qs.size();
   if (nq == 0) return true;
@@ -1271,6 +1279,7 @@ bool Slab::RangeBatch(const std::vector<DevRangeQ>& qs, bool d2h,
   HIP_CHECK(hipStreamSynchronize(I->stream));
   (void)hipEventElapsedTime(&ms, I->ev0, I->ev1);
   perf.pack_d2h_ms += ms;
+  if (!parse) return true;  // raw mode: bytes landed in pinned host memory
   for (int q = 0; q < nq; ++q) {
     RangeResult& r = (*outs)[q];
     if (r.overflow) continue;

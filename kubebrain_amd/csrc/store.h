@@ -170,6 +170,7 @@ class Store {
   double host_write_s_ = 0, host_merge_s_ = 0;
   int64_t ops_create_ = 0, ops_update_ = 0, ops_delete_ = 0, ops_range_ = 0;
   int64_t delivered_ = 0;  // watch events enqueued to watchers (fan-out)
+  bool keep_event_log_ = true;
 };
 
 Bytes EncodeObjectKey(const Bytes& userKey, uint64_t rev);  // coder/normal.go:42-50
