@@ -301,6 +301,14 @@ def main():
         dist = tdist
         os.environ["KB_DEVICE"] = str(local_rank)
 
+    # weak scaling (SURVEY §8e / configs[3]): per-GPU keyspace fixed, total
+    # keys grow with N (N x 1M at the defaults)
+    if world > 1:
+        args.nns *= world
+        args.extra_revs *= world
+        args.watchers *= world
+        args.watch_events *= world
+
     # device capacity sizing for this workload
     total_rows = int((args.nns * args.per_ns * 2.1 + args.extra_revs) * 1.6 / world)
     os.environ.setdefault("KB_MAX_ROWS", str(max(total_rows, 1 << 20)))
@@ -315,8 +323,10 @@ def main():
     import kubebrain_amd
     import torch
 
+    import zlib
     namespaces, keys = make_keys(args.nns, args.per_ns)
-    my_ns = set(ns for i, ns in enumerate(namespaces) if i % world == rank)
+    # namespace-hash sharding (same routing as tests/test_gloo_shard)
+    my_ns = set(ns for ns in namespaces if zlib.crc32(ns) % world == rank)
 
     store = kubebrain_amd.open_store(store_prefix=b"/registry")
     store.set_current_rev(1000)
