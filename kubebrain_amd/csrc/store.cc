@@ -160,6 +160,7 @@ void Store::putRevRow(const Bytes& key, uint64_t objrev, bool flag9) {
 }
 
 bool Store::syncReads(std::string* err) {
+  sync_n_++;
   auto t0 = std::chrono::steady_clock::now();
   if (!heap_pending_.empty()) {
     int64_t off = 0;
@@ -195,7 +196,7 @@ bool Store::syncReads(std::string* err) {
   if (slab_->delta_rows() >= cfg_.flush_rows) {
     if (!slab_->Fold(err)) return false;
   }
-  host_merge_s_ +=
+  sync_s_ +=
       std::chrono::duration<double>(std::chrono::steady_clock::now() - t0).count();
   return true;
 }
@@ -877,7 +878,7 @@ std::string Store::PerfJson() {
            "\"bytes_gathered\":%lld,\"winners\":%lld,\"merges\":%lld,"
            "\"compacts\":%lld,\"filter_launches\":%lld,\"filter_events\":%lld,"
            "\"filter_watchers\":%lld,\"slab_rows\":%lld,\"heap_used\":%lld,"
-           "\"delivered\":%lld,"
+           "\"delivered\":%lld,\"sync_s\":%.3f,\"syncs\":%lld,"
            "\"ops\":{\"create\":%lld,\"update\":%lld,\"delete\":%lld,"
            "\"range\":%lld}}",
            p.scan_ms, p.gather_ms, p.get_ms, p.compact_ms, p.merge_ms,
@@ -887,6 +888,7 @@ std::string Store::PerfJson() {
            (long long)p.filter_launches, (long long)p.filter_events,
            (long long)p.filter_watchers, (long long)slab_->rows(),
            (long long)slab_->heap_used(), (long long)delivered_,
+           sync_s_, (long long)sync_n_,
            (long long)ops_create_,
            (long long)ops_update_, (long long)ops_delete_, (long long)ops_range_);
   return buf;

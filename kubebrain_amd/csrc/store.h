@@ -171,6 +171,8 @@ class Store {
   int64_t ops_create_ = 0, ops_update_ = 0, ops_delete_ = 0, ops_range_ = 0;
   int64_t delivered_ = 0;  // watch events enqueued to watchers (fan-out)
   bool keep_event_log_ = true;
+  double sync_s_ = 0;
+  int64_t sync_n_ = 0;
 };
 
 Bytes EncodeObjectKey(const Bytes& userKey, uint64_t rev);  // coder/normal.go:42-50
