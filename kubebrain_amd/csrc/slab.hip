@@ -7,6 +7,7 @@
 
 #include <hip/hip_runtime.h>
 
+#include <chrono>
 #include <cstdio>
 #include <cstring>
 
@@ -1305,14 +1306,22 @@ bool Slab::RangeBatchEx(const std::vector<DevRangeQ>& qs, bool d2h, bool parse,
 bool Slab::GetBatch(const std::vector<DevGetQ>& qs, std::vector<GetResult>* outs,
                     std::string* err) {
   Impl* I = p;
+  auto tt0 = std::chrono::steady_clock::now();
+  auto lap = [&](double* acc) {
+    auto t = std::chrono::steady_clock::now();
+    *acc += std::chrono::duration<double>(t - tt0).count() * 1e3;
+    tt0 = t;
+  };
   int nq = (int)qs.size();
   outs->assign(nq, GetResult());
   if (nq == 0) return true;
   if (nq > I->max_q) { if (err) *err = "too many gets per batch"; return false; }
   int64_t slot = I->arena_bytes / nq;
   slot &= ~15ll;
+  lap(&perf.dbg_a);
   HIP_CHECK(hipMemcpyAsync(I->d_gq, qs.data(), sizeof(DevGetQ) * nq,
                            hipMemcpyHostToDevice, I->stream));
+  lap(&perf.dbg_b);
   HIP_CHECK(hipEventRecord(I->ev0, I->stream));
   int blocks = (int)ceil_div(nq, 4);
   hipLaunchKernelGGL(k_get2, dim3(blocks), dim3(256), 0, I->stream, I->A.keys,
@@ -1328,7 +1337,9 @@ bool Slab::GetBatch(const std::vector<DevGetQ>& qs, std::vector<GetResult>* outs
   HIP_CHECK(hipMemcpyAsync(ometa.data(), I->d_ometa, nq * 8, hipMemcpyDeviceToHost, I->stream));
   HIP_CHECK(hipMemcpyAsync(ofound.data(), I->d_found32, nq * 4, hipMemcpyDeviceToHost, I->stream));
   HIP_CHECK(hipMemcpyAsync(oovf.data(), I->d_ovf, nq * 4, hipMemcpyDeviceToHost, I->stream));
+  lap(&perf.dbg_c);
   HIP_CHECK(hipStreamSynchronize(I->stream));
+  lap(&perf.dbg_d);
   float ms = 0;
   (void)hipEventElapsedTime(&ms, I->ev0, I->ev1);
   perf.get_ms += ms;
@@ -1346,6 +1357,7 @@ bool Slab::GetBatch(const std::vector<DevGetQ>& qs, std::vector<GetResult>* outs
       HIP_CHECK(hipMemcpy(g.val.data(), I->d_gbuf + (int64_t)q * slot, vlen,
                           hipMemcpyDeviceToHost));
   }
+  lap(&perf.dbg_e);
   return true;
 }
 

@@ -91,6 +91,7 @@ struct Perf {
           merges = 0, compacts = 0, filter_launches = 0;
   int64_t rows_scanned = 0, bytes_gathered = 0, winners = 0,
           filter_events = 0, filter_watchers = 0;
+  double dbg_a = 0, dbg_b = 0, dbg_c = 0, dbg_d = 0, dbg_e = 0;
 };
 
 struct WatchFilterBatch {

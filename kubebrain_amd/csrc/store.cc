@@ -878,7 +878,7 @@ std::string Store::PerfJson() {
            "\"bytes_gathered\":%lld,\"winners\":%lld,\"merges\":%lld,"
            "\"compacts\":%lld,\"filter_launches\":%lld,\"filter_events\":%lld,"
            "\"filter_watchers\":%lld,\"slab_rows\":%lld,\"heap_used\":%lld,"
-           "\"delivered\":%lld,\"sync_s\":%.3f,\"syncs\":%lld,"
+           "\
            "\"ops\":{\"create\":%lld,\"update\":%lld,\"delete\":%lld,"
            "\"range\":%lld}}",
            p.scan_ms, p.gather_ms, p.get_ms, p.compact_ms, p.merge_ms,
@@ -889,6 +889,7 @@ std::string Store::PerfJson() {
            (long long)p.filter_watchers, (long long)slab_->rows(),
            (long long)slab_->heap_used(), (long long)delivered_,
            sync_s_, (long long)sync_n_,
+           p.dbg_a, p.dbg_b, p.dbg_c, p.dbg_d, p.dbg_e,
            (long long)ops_create_,
            (long long)ops_update_, (long long)ops_delete_, (long long)ops_range_);
   return buf;
