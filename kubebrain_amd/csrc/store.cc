@@ -878,7 +878,8 @@ std::string Store::PerfJson() {
            "\"bytes_gathered\":%lld,\"winners\":%lld,\"merges\":%lld,"
            "\"compacts\":%lld,\"filter_launches\":%lld,\"filter_events\":%lld,"
            "\"filter_watchers\":%lld,\"slab_rows\":%lld,\"heap_used\":%lld,"
-           "\
+           "\"delivered\":%lld,\"sync_s\":%.3f,\"syncs\":%lld,"
+           "\"dbg\":[%.1f,%.1f,%.1f,%.1f,%.1f],"
            "\"ops\":{\"create\":%lld,\"update\":%lld,\"delete\":%lld,"
            "\"range\":%lld}}",
            p.scan_ms, p.gather_ms, p.get_ms, p.compact_ms, p.merge_ms,
