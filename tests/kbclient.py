@@ -150,7 +150,7 @@ class Store:
                              C.c_uint64(prev_rev), C.byref(hr), C.byref(succ),
                              C.byref(has), self.buf, C.c_size_t(self.BUF),
                              C.byref(vlen), C.byref(kvrev))
-        kv = Kv(key, self.buf.raw[:vlen.value], kvrev.value) if has.value else None
+        kv = Kv(key, self.buf[:vlen.value], kvrev.value) if has.value else None
         return WriteResp(rc, hr.value, bool(succ.value), kv)
 
     def update(self, key: bytes, val: bytes, prev_rev: int = 0) -> WriteResp:
@@ -164,7 +164,7 @@ class Store:
         rc = self._f("get")(C.c_void_p(self.h), key, C.c_size_t(len(key)),
                             C.c_uint64(rev), C.byref(hr), C.byref(has), self.buf,
                             C.c_size_t(self.BUF), C.byref(vlen), C.byref(mod))
-        kv = Kv(key, self.buf.raw[:vlen.value], mod.value) if has.value else None
+        kv = Kv(key, self.buf[:vlen.value], mod.value) if has.value else None
         return rc, hr.value, kv
 
     def list(self, start: bytes, end: bytes, rev: int = 0, limit: int = 0) -> RangeResp:
@@ -175,7 +175,7 @@ class Store:
                              C.byref(hr), C.byref(more))
         if rc not in (OK,):
             return RangeResp(rc, hr.value)
-        return RangeResp(rc, hr.value, _parse_kvs(self.buf.raw[:out_len.value]),
+        return RangeResp(rc, hr.value, _parse_kvs(self.buf[:out_len.value]),
                          bool(more.value))
 
     def count(self, start: bytes, end: bytes):
@@ -203,7 +203,7 @@ class Store:
                                    C.c_size_t(self.BUF), C.byref(out_len))
         if rc != OK:
             return rc, []
-        return rc, _parse_events(self.buf.raw[:out_len.value])
+        return rc, _parse_events(self.buf[:out_len.value])
 
     def watch_cancel(self, wid: int):
         self._f("watch_cancel")(C.c_void_p(self.h), C.c_longlong(wid))
@@ -225,7 +225,7 @@ class Store:
         rc = self._f("dump")(C.c_void_p(self.h), self.buf, C.c_size_t(self.BUF),
                              C.byref(out_len), C.byref(n))
         assert rc == 0, rc
-        buf = self.buf.raw[:out_len.value]
+        buf = self.buf[:out_len.value]
         (cnt,) = struct.unpack_from("<I", buf, 0)
         off = 4
         rows = []
@@ -242,7 +242,7 @@ class Store:
         rc = self._f("event_log")(C.c_void_p(self.h), self.buf, C.c_size_t(self.BUF),
                                   C.byref(out_len))
         assert rc == 0, rc
-        return _parse_events(self.buf.raw[:out_len.value])
+        return _parse_events(self.buf[:out_len.value])
 
 
 def open_oracle(**kw) -> Store:
