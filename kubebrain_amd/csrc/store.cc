@@ -798,43 +798,40 @@ bool Store::BenchRange(const uint8_t* qbuf, size_t nq, bool d2h,
   // the measured hot path: batched List semantics with inputs resident in HBM
   std::lock_guard<std::recursive_mutex> lk(mu_);
   if (!syncReads(err)) return false;
-  struct Q { Bytes s, e; uint64_t rev; int64_t limit; };
-  std::vector<Q> qs;
-  qs.reserve(nq);
-  const uint8_t* p = qbuf;
-  for (size_t i = 0; i < nq; ++i) {
-    uint32_t slen, elen;
-    uint64_t rev, limit;
-    memcpy(&slen, p, 4); p += 4;
-    memcpy(&elen, p, 4); p += 4;
-    memcpy(&rev, p, 8); p += 8;
-    memcpy(&limit, p, 8); p += 8;
-    Q q;
-    q.s.assign((const char*)p, slen); p += slen;
-    q.e.assign((const char*)p, elen); p += elen;
-    q.rev = rev;
-    q.limit = (int64_t)limit;
-    qs.push_back(std::move(q));
+  // parse straight into device query structs (no per-query allocations)
+  std::vector<DevRangeQ> qall(nq);
+  std::vector<int64_t> limits(nq);
+  {
+    const uint8_t* p = qbuf;
+    for (size_t i = 0; i < nq; ++i) {
+      uint32_t slen, elen;
+      uint64_t rev, limit;
+      memcpy(&slen, p, 4); p += 4;
+      memcpy(&elen, p, 4); p += 4;
+      memcpy(&rev, p, 8); p += 8;
+      memcpy(&limit, p, 8); p += 8;
+      DevRangeQ& q = qall[i];
+      memset(q.start, 0, KEYW);
+      memset(q.end, 0, KEYW);
+      memcpy(q.start, p, std::min((size_t)slen, (size_t)KEYW)); p += slen;
+      memcpy(q.end, p, std::min((size_t)elen, (size_t)KEYW)); p += elen;
+      q.read_rev = rev == 0 ? committed_ : rev;
+      q.cap = (int64_t)limit > 0 ? (int64_t)limit + 1 : 0;
+      q.count_only = 0;
+      q._pad = 0;
+      limits[i] = (int64_t)limit;
+    }
   }
   const int64_t kMax = 1024;
   unsigned long long tot = 0;
   auto t0 = std::chrono::steady_clock::now();
-  for (size_t b0 = 0; b0 < qs.size(); b0 += kMax) {
-    size_t bn = std::min((size_t)kMax, qs.size() - b0);
-    std::vector<DevRangeQ> dq(bn);
-    for (size_t j = 0; j < bn; ++j) {
-      DevRangeQ& q = dq[j];
-      memset(&q, 0, sizeof(q));
-      pad96(qs[b0 + j].s, q.start);
-      pad96(qs[b0 + j].e, q.end);
-      q.read_rev = qs[b0 + j].rev == 0 ? committed_ : qs[b0 + j].rev;
-      q.cap = qs[b0 + j].limit > 0 ? qs[b0 + j].limit + 1 : 0;
-      q.count_only = 0;
-    }
-    std::vector<kbslab::RangeResult> outs;
+  std::vector<kbslab::RangeResult> outs;
+  for (size_t b0 = 0; b0 < (size_t)nq; b0 += kMax) {
+    size_t bn = std::min((size_t)kMax, (size_t)nq - b0);
+    std::vector<DevRangeQ> dq(qall.begin() + b0, qall.begin() + b0 + bn);
     if (!slab_->RangeBatchEx(dq, d2h, /*parse=*/false, &outs, err)) return false;
     for (size_t j = 0; j < bn; ++j) {
-      int64_t lim = qs[b0 + j].limit;
+      int64_t lim = limits[b0 + j];
       int64_t w = outs[j].written;
       tot += (unsigned long long)(lim > 0 && w > lim ? lim : w);
     }
