@@ -354,11 +354,16 @@ def main():
                  for _ in range(n_pre)]
     step_i = [0]
 
+    t_split = {"range_s": 0.0, "txn_s": 0.0}
+
     def one_step(d2h=False):
         blob = pre_blobs[step_i[0] % n_pre]
         step_i[0] += 1
-        tot, _ = bench_range_call(store, blob, nq, d2h)
+        tot, rsecs = bench_range_call(store, blob, nq, d2h)
+        t_split["range_s"] += rsecs
+        t0 = time.time()
         run_txns(store, live, revs, qrng, ntx)
+        t_split["txn_s"] += time.time() - t0
         return tot
 
     def barrier():
@@ -371,6 +376,7 @@ def main():
         one_step()
     barrier()
     store._f("perf_reset")(ctypes.c_void_p(store.h))
+    t_split["range_s"] = t_split["txn_s"] = 0.0
     t0 = time.time()
     for _ in range(args.steps):
         one_step()
@@ -457,6 +463,8 @@ def main():
             "watch_events_per_sec": round(wrate, 1),
             "watch_delivered_rank0": delivered,
             "ops_per_sec_with_d2h": round(OPS_PER_STEP * d2h_steps * world / d2h_elapsed, 1),
+            "step_split_ms": {"range": round(t_split["range_s"] / args.steps * 1e3, 3),
+                               "txn": round(t_split["txn_s"] / args.steps * 1e3, 3)},
             "roofline": roofline,
             "cpu_baseline": cpu_baseline,
             "perf": p,

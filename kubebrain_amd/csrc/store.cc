@@ -896,6 +896,7 @@ std::string Store::PerfJson() {
 void Store::PerfReset() {
   slab_->perf = kbslab::Perf();
   ops_create_ = ops_update_ = ops_delete_ = ops_range_ = 0;
+  sync_s_ = 0; sync_n_ = 0; delivered_ = 0;
 }
 
 }  // namespace kbstore
