@@ -382,6 +382,7 @@ def main():
         one_step()
     barrier()
     elapsed = time.time() - t0
+    split_snapshot = dict(t_split)  # before the untimed d2h/watch phases
     if dist:
         import torch as _t
         e = _t.tensor([elapsed])
@@ -421,13 +422,24 @@ def main():
         "peak": peak / 1e9,
         "unit": "GB/s",
         "frac": round(achieved / peak, 4),
-        "traffic": None,
+        "traffic": None,  # filled below from the committed PMC sidecar
         "kernel": "k_range_scan+k_gather",
         "scan_only_GBps": round((p["rows_scanned"] * 16) / (p["scan_ms"] / 1e3) / 1e9, 2)
                           if p["scan_ms"] > 0 else None,
         "note": "achieved = (rows_scanned x 16B + gathered bytes) / HIP-event "
                 "kernel time on the store stream",
     }
+
+    # PMC-measured HBM traffic per launch (profiles/pmc_traffic.json,
+    # collected with rocprofv3 --pmc FETCH_SIZE on this same workload)
+    try:
+        tj = json.load(open(os.path.join(REPO, "profiles", "pmc_traffic.json")))
+        per_launch = (tj["scan_bytes_per_launch"] * tj.get("scan_correction", 1)
+                      + tj["gather_bytes_per_launch"])
+        roofline["traffic"] = int(per_launch)
+        roofline["traffic_note"] = tj["_source"]
+    except Exception:
+        pass
 
     cpu_baseline = None
     if rank == 0 and world == 1 and not args.no_cpu_baseline:
@@ -463,8 +475,8 @@ def main():
             "watch_events_per_sec": round(wrate, 1),
             "watch_delivered_rank0": delivered,
             "ops_per_sec_with_d2h": round(OPS_PER_STEP * d2h_steps * world / d2h_elapsed, 1),
-            "step_split_ms": {"range": round(t_split["range_s"] / args.steps * 1e3, 3),
-                               "txn": round(t_split["txn_s"] / args.steps * 1e3, 3)},
+            "step_split_ms": {"range": round(split_snapshot["range_s"] / args.steps * 1e3, 3),
+                               "txn": round(split_snapshot["txn_s"] / args.steps * 1e3, 3)},
             "roofline": roofline,
             "cpu_baseline": cpu_baseline,
             "perf": p,

@@ -1231,17 +1231,27 @@ bool Slab::RangeBatchEx(const std::vector<DevRangeQ>& qs, bool d2h, bool parse,
                      I->d_found, nq, I->d_gbuf, qcap, I->d_offs, I->d_gbytes,
                      I->d_ovf, I->d_bytes);
   HIP_CHECK(hipEventRecord(I->ev2, I->stream));
-  // small result metadata D2H
+  // small result metadata: d_found/d_total/d_gbytes are CONTIGUOUS slices of
+  // one allocation (see Create), so one D2H covers them; ovf+counters ride
+  // two more copies
   std::vector<int64_t> found(nq), total(nq), gbytes(nq);
   std::vector<int32_t> ovf(nq);
   unsigned long long scanned = 0, bytes = 0;
-  HIP_CHECK(hipMemcpyAsync(found.data(), I->d_found, nq * 8, hipMemcpyDeviceToHost, I->stream));
-  HIP_CHECK(hipMemcpyAsync(total.data(), I->d_total, nq * 8, hipMemcpyDeviceToHost, I->stream));
-  HIP_CHECK(hipMemcpyAsync(gbytes.data(), I->d_gbytes, nq * 8, hipMemcpyDeviceToHost, I->stream));
-  HIP_CHECK(hipMemcpyAsync(ovf.data(), I->d_ovf, nq * 4, hipMemcpyDeviceToHost, I->stream));
-  HIP_CHECK(hipMemcpyAsync(&scanned, I->d_scanned, 8, hipMemcpyDeviceToHost, I->stream));
-  HIP_CHECK(hipMemcpyAsync(&bytes, I->d_bytes, 8, hipMemcpyDeviceToHost, I->stream));
-  HIP_CHECK(hipStreamSynchronize(I->stream));
+  {
+    std::vector<int64_t> meta3(3 * nq);
+    unsigned long long cnt2[2];
+    HIP_CHECK(hipMemcpyAsync(meta3.data(), I->d_found, nq * 8, hipMemcpyDeviceToHost, I->stream));
+    HIP_CHECK(hipMemcpyAsync(meta3.data() + nq, I->d_total, nq * 8, hipMemcpyDeviceToHost, I->stream));
+    HIP_CHECK(hipMemcpyAsync(meta3.data() + 2 * nq, I->d_gbytes, nq * 8, hipMemcpyDeviceToHost, I->stream));
+    HIP_CHECK(hipMemcpyAsync(ovf.data(), I->d_ovf, nq * 4, hipMemcpyDeviceToHost, I->stream));
+    HIP_CHECK(hipMemcpyAsync(&scanned, I->d_scanned, 8, hipMemcpyDeviceToHost, I->stream));
+    HIP_CHECK(hipMemcpyAsync(&bytes, I->d_bytes, 8, hipMemcpyDeviceToHost, I->stream));
+    HIP_CHECK(hipStreamSynchronize(I->stream));
+    memcpy(found.data(), meta3.data(), nq * 8);
+    memcpy(total.data(), meta3.data() + nq, nq * 8);
+    memcpy(gbytes.data(), meta3.data() + 2 * nq, nq * 8);
+    (void)cnt2;
+  }
   float ms = 0;
   (void)hipEventElapsedTime(&ms, I->ev0, I->ev1);
   perf.scan_ms += ms;

@@ -225,10 +225,10 @@ void Store::notify(const Bytes& key, const Bytes& val, uint64_t revision,
   e.type = type;
   e.revision = revision;
   e.kv_key = key;
-  if (type == Event::DELETE) { e.kv_value = val; e.kv_revision = prevRevision; }
-  else { e.kv_value = val; e.kv_revision = revision; }
-  ring_.Add(e);  // backend.go:263
+  e.kv_value = val;
+  e.kv_revision = type == Event::DELETE ? prevRevision : revision;
   if (keep_event_log_) event_log_.push_back(e);
+  ring_.Add(e);  // backend.go:263 (ring keeps its own copy)
   pending_.push_back(std::move(e));
   if (pending_.size() >= 300) pumpEvents();  // eventBatchSize (backend.go:41)
 }
