@@ -287,6 +287,8 @@ def main():
     ap.add_argument("--watchers", type=int, default=1000)
     ap.add_argument("--watch-events", type=int, default=6000)
     ap.add_argument("--no-cpu-baseline", action="store_true")
+    ap.add_argument("--compact-bench", action="store_true",
+                    help="also time a full compaction sweep (configs[2] shape)")
     args = ap.parse_args()
 
     rank = int(os.environ.get("RANK", "0"))
@@ -397,6 +399,33 @@ def main():
     d2h_elapsed = time.time() - t0
     d2h_steps = max(2, args.steps // 8)
 
+    # compaction sweep leg (configs[2]: drop revisions < compactRev; the
+    # sweep reads the whole slab and stream-compacts rows + value heap)
+    compact_stats = None
+    if args.compact_bench:
+        p_pre = perf(store)
+        rows_pre = p_pre["slab_rows"]
+        heap_pre = p_pre["heap_used"]
+        t0 = time.time()
+        rc, crev = store.compact(0)
+        csecs = time.time() - t0
+        assert rc == 0
+        p_post = perf(store)
+        sweep_bytes = rows_pre * 112 + p_post["slab_rows"] * 112 \
+            + heap_pre + p_post["heap_used"]
+        compact_stats = {
+            "seconds": round(csecs, 4),
+            "rows_before": rows_pre,
+            "rows_after": p_post["slab_rows"],
+            "heap_before": heap_pre,
+            "heap_after": p_post["heap_used"],
+            "device_ms": round(p_post["compact_ms"] - p_pre["compact_ms"], 3),
+            "algorithmic_GBps": round(sweep_bytes / max(
+                (p_post["compact_ms"] - p_pre["compact_ms"]) / 1e3, 1e-9) / 1e9, 2),
+            "compact_rev": crev,
+        }
+        log(rank, f"[bench] compact sweep: {compact_stats}")
+
     # watch fan-out leg
     wrate, delivered = watch_leg(store, my_ns_list, live, revs, qrng,
                                  max(args.watchers // world, 8),
@@ -472,6 +501,7 @@ def main():
                 "limit": LIMIT,
                 "ops_per_step": OPS_PER_STEP,
             },
+            "compact_sweep": compact_stats,
             "watch_events_per_sec": round(wrate, 1),
             "watch_delivered_rank0": delivered,
             "ops_per_sec_with_d2h": round(OPS_PER_STEP * d2h_steps * world / d2h_elapsed, 1),
