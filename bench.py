@@ -100,11 +100,22 @@ def build_store(store, namespaces, keys, rng, extra_revs, tomb_frac, my_ns=None)
     if tomb_frac and n:
         nt = int(n * tomb_frac)
         idx = rng.choice(n, size=nt, replace=False)
-        for i in idx:
-            k = sel[int(i)]
-            r = store.delete(k, revs[k])
-            assert r.succeeded
-            revs[k] = None
+        fdel = store._f("bench_del")
+        batch = 8192
+        for b0 in range(0, nt, batch):
+            ks = [sel[int(i)] for i in idx[b0:b0 + batch]]
+            parts = []
+            for k in ks:
+                parts.append(struct.pack("<IQ", len(k), revs[k]))
+                parts.append(k)
+            out = np.empty(len(ks), dtype=np.uint64)
+            rc = fdel(ctypes.c_void_p(store.h), b"".join(parts),
+                      ctypes.c_size_t(len(ks)),
+                      out.ctypes.data_as(ctypes.POINTER(ctypes.c_uint64)))
+            assert rc == 0
+            for k, nr in zip(ks, out):
+                assert nr != 0
+                revs[k] = None
         phase("tombstones")
     store._f("flush")(ctypes.c_void_p(store.h))
     phase("final fold")

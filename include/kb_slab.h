@@ -121,6 +121,8 @@ int kb_bench_range(kb_store*, const uint8_t* qbuf, size_t nq, int d2h,
  * u64 prev_rev; u32 vlen; key; val}; out_revs[i] = new revision or 0 on CAS
  * failure */
 int kb_bench_txn(kb_store*, const uint8_t* tbuf, size_t n, uint64_t* out_revs);
+/* batched deletes: dbuf = n × {u32 klen; u64 prev_rev; key} */
+int kb_bench_del(kb_store*, const uint8_t* dbuf, size_t n, uint64_t* out_revs);
 /* fast bulk insert == n serial Creates of fresh keys (see okb_bulk_create) */
 int kb_bulk_create(kb_store*, const uint8_t* keys, const uint32_t* klens,
                    const uint8_t* vals, const uint32_t* vlens, size_t n);

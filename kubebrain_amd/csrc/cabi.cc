@@ -265,6 +265,15 @@ int kb_bench_txn(kb_store* h, const uint8_t* tbuf, size_t n, uint64_t* out_revs)
   return 0;
 }
 
+int kb_bench_del(kb_store* h, const uint8_t* dbuf, size_t n, uint64_t* out_revs) {
+  std::string err;
+  if (!((Store*)h)->BenchDel(dbuf, n, out_revs, &err)) {
+    set_err(KB_EINTERNAL, err);
+    return KB_EINTERNAL;
+  }
+  return 0;
+}
+
 int kb_bulk_create(kb_store* h, const uint8_t* keys, const uint32_t* klens,
                    const uint8_t* vals, const uint32_t* vlens, size_t n) {
   std::string err;

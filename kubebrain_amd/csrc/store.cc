@@ -865,6 +865,24 @@ bool Store::BenchTxn(const uint8_t* tbuf, size_t n, uint64_t* out_revs,
   return true;
 }
 
+bool Store::BenchDel(const uint8_t* dbuf, size_t n, uint64_t* out_revs,
+                     std::string* err) {
+  // packed: {u32 klen; u64 prev_rev; key} x n ; out_revs[i]=rev or 0
+  const uint8_t* p = dbuf;
+  for (size_t i = 0; i < n; ++i) {
+    uint32_t klen;
+    uint64_t prev;
+    memcpy(&klen, p, 4); p += 4;
+    memcpy(&prev, p, 8); p += 8;
+    Bytes key((const char*)p, klen); p += klen;
+    Status st;
+    auto r = Delete(key, prev, &st);
+    if (st != OK) { if (err) *err = "bench del st=" + std::to_string(st); return false; }
+    out_revs[i] = r.succeeded ? r.header_revision : 0;
+  }
+  return true;
+}
+
 std::string Store::PerfJson() {
   const kbslab::Perf& p = slab_->perf;
   char buf[1536];

@@ -92,6 +92,8 @@ class Store {
   // revision on success, 0 on CAS failure
   bool BenchTxn(const uint8_t* tbuf, size_t n, uint64_t* out_revs,
                 std::string* err);
+  bool BenchDel(const uint8_t* dbuf, size_t n, uint64_t* out_revs,
+                std::string* err);
   std::string PerfJson();
   void PerfReset();
 

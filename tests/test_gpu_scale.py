@@ -26,7 +26,7 @@ def bulk(store, keys, vals_blob):
     klens = np.array([len(k) for k in keys], dtype=np.uint32).tobytes()
     vlens = np.full(n, VAL, dtype=np.uint32).tobytes()
     f = store._f("bulk_create")
-    rc = f(ctypes.c_void_p(store.h), b"".join(keys), klens, vals_blob,
+    rc = f(ctypes.c_void_p(store.h), b"".join(keys), klens, vals_blob, vlens,
            ctypes.c_size_t(n))
     assert rc == 0
 
