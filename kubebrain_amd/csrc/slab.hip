@@ -914,10 +914,12 @@ struct Slab::Impl {
         hipLaunchKernelGGL(k_block_scan, dim3((uint32_t)nb3), dim3(256), 0,
                            stream, lv2, lv2o, lv3, nb2);
         if (nb3 > 1) {
-          // max_rows <= 256^3*... (16.7M*256): one more level unnecessary for
-          // any supported size; guard anyway
-          if (err) *err = "scan: size too large";
-          return false;
+          // 4th level covers up to 256^4 = 4.3G rows (nb3 <= 256 by max_rows)
+          if (nb3 > 256) { if (err) *err = "scan: size too large"; return false; }
+          hipLaunchKernelGGL(k_block_scan, dim3(1), dim3(256), 0, stream, lv3,
+                             lv3o, lv3 + 256, nb3);
+          hipLaunchKernelGGL(k_add_offsets, dim3((uint32_t)nb3), dim3(256), 0,
+                             stream, lv2o, lv3o, nb2);
         }
         hipLaunchKernelGGL(k_add_offsets, dim3((uint32_t)nb2), dim3(256), 0,
                            stream, lv1o, lv2o, nb1);
@@ -1050,7 +1052,7 @@ Slab* Slab::Create(int64_t max_rows, int64_t heap_cap, int device,
   HIP_CHECK_NULL(hipMalloc(&I->lv1o, nb1 * 8));
   HIP_CHECK_NULL(hipMalloc(&I->lv2, nb2 * 8));
   HIP_CHECK_NULL(hipMalloc(&I->lv2o, nb2 * 8));
-  HIP_CHECK_NULL(hipMalloc(&I->lv3, 256 * 8));
+  HIP_CHECK_NULL(hipMalloc(&I->lv3, 512 * 8));  // +256 spill slot for level 4
   HIP_CHECK_NULL(hipMalloc(&I->lv3o, 256 * 8));
   HIP_CHECK_NULL(hipMalloc(&I->d_qs, sizeof(DevRangeQ) * I->max_q));
   HIP_CHECK_NULL(hipMalloc(&I->d_gq, sizeof(DevGetQ) * I->max_q));

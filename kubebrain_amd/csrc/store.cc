@@ -160,6 +160,7 @@ void Store::putRevRow(const Bytes& key, uint64_t objrev, bool flag9) {
 }
 
 bool Store::syncReads(std::string* err) {
+  if (!fatal_.empty()) { if (err) *err = "store failed: " + fatal_; return false; }
   sync_n_++;
   auto t0 = std::chrono::steady_clock::now();
   if (!heap_pending_.empty()) {
@@ -390,7 +391,10 @@ WriteResponse Store::Create(const Bytes& key, const Bytes& value, Status* st) {
   uint64_t revision = deal(0, &dst);
   Status err = dst == OK ? createInternal(key, value, revision) : dst;
   notify(key, value, revision, 0, err == OK, Event::CREATE);
-  if (newrows_.size() >= 16384) { std::string e_; (void)syncReads(&e_); }
+  if (newrows_.size() >= 16384) {
+    std::string e_;
+    if (!syncReads(&e_)) fatal_ = e_;  // surfaces on every later op
+  }
   if (err == CAS_FAILED) {
     resp.header_revision = revision;
     *st = OK;
@@ -474,7 +478,10 @@ WriteResponse Store::Update(const Bytes& key, const Bytes& value,
     }
     notify(key, value, curRev, prevRev, err == OK, Event::PUT);
   }
-  if (newrows_.size() >= 16384) { std::string e_; (void)syncReads(&e_); }
+  if (newrows_.size() >= 16384) {
+    std::string e_;
+    if (!syncReads(&e_)) fatal_ = e_;  // surfaces on every later op
+  }
   resp.header_revision = curRev;
   resp.succeeded = (err == OK);
   if (err == CAS_FAILED) {
@@ -537,7 +544,10 @@ WriteResponse Store::Delete(const Bytes& key, uint64_t prevRev, Status* st) {
     }
   }
   notify(key, old.value, newRevision, old.revision, err == OK, Event::DELETE);
-  if (newrows_.size() >= 16384) { std::string e_; (void)syncReads(&e_); }
+  if (newrows_.size() >= 16384) {
+    std::string e_;
+    if (!syncReads(&e_)) fatal_ = e_;  // surfaces on every later op
+  }
   resp.header_revision = newRevision;
   resp.succeeded = (err == OK);
   if (err == CAS_FAILED) {

@@ -175,6 +175,7 @@ class Store {
   bool keep_event_log_ = true;
   double sync_s_ = 0;
   int64_t sync_n_ = 0;
+  std::string fatal_;  // first unrecoverable device error (e.g. slab full)
 };
 
 Bytes EncodeObjectKey(const Bytes& userKey, uint64_t rev);  // coder/normal.go:42-50
