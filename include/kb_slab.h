@@ -99,6 +99,23 @@ long long kb_watch(kb_store*, const uint8_t* prefix, size_t plen, uint64_t rev,
 int kb_watch_poll(kb_store*, long long wid, uint8_t* out, size_t cap, size_t* out_len);
 void kb_watch_cancel(kb_store*, long long wid);
 
+/* ---- streaming full-range reads (backend.ListByStream range.go:247-256 +
+ * scanner.RangeStream scanner.go:129-145; batches of 300, receiver.go:118-150)
+ * and GetPartitions (range.go:208-245; single partition like badger.go:52-54,
+ * sharding lives above this ABI). Stream results are pinned at read_rev; a
+ * compaction past read_rev mid-stream returns KB_ECOMPACTED (the reference
+ * holds an engine snapshot instead). ---- */
+long long kb_stream_open(kb_store*, const uint8_t* start, size_t slen,
+                         const uint8_t* end, size_t elen, uint64_t rev,
+                         uint64_t* read_rev, int* status);
+/* out = packed kvs like kb_list; an EMPTY batch is the end marker */
+int kb_stream_next(kb_store*, long long sid, uint8_t* out, size_t cap,
+                   size_t* out_len);
+void kb_stream_close(kb_store*, long long sid);
+int kb_partitions(kb_store*, const uint8_t* start, size_t slen,
+                  const uint8_t* end, size_t elen, uint8_t* out, size_t cap,
+                  size_t* out_len, uint64_t* header_rev);
+
 /* ---- revision (tso/tso.go:41-76) ---- */
 unsigned long long kb_current_rev(kb_store*);
 void kb_set_current_rev(kb_store*, unsigned long long rev); /* leader TSO init */

@@ -200,6 +200,45 @@ int kb_watch_poll(kb_store* h, long long wid, uint8_t* out, size_t cap,
 
 void kb_watch_cancel(kb_store* h, long long wid) { ((Store*)h)->WatchCancel(wid); }
 
+long long kb_stream_open(kb_store* h, const uint8_t* start, size_t slen,
+                         const uint8_t* end, size_t elen, uint64_t rev,
+                         uint64_t* read_rev, int* status) {
+  Status st;
+  long long sid = ((Store*)h)->StreamOpen(Bytes((const char*)start, slen),
+                                          Bytes((const char*)end, elen), rev,
+                                          read_rev, &st);
+  *status = st;
+  return sid;
+}
+
+int kb_stream_next(kb_store* h, long long sid, uint8_t* out, size_t cap,
+                   size_t* out_len) {
+  std::vector<kbstore::KeyValue> kvs;
+  Status st = ((Store*)h)->StreamNext(sid, &kvs);
+  Writer w{out, cap};
+  w.u32((uint32_t)kvs.size());
+  for (auto& kv : kvs) { w.u64(kv.revision); w.str(kv.key); w.str(kv.value); }
+  *out_len = w.off;
+  if (w.overflow) return KB_ENOBUF;
+  return st;
+}
+
+void kb_stream_close(kb_store* h, long long sid) { ((Store*)h)->StreamClose(sid); }
+
+int kb_partitions(kb_store* h, const uint8_t* start, size_t slen,
+                  const uint8_t* end, size_t elen, uint8_t* out, size_t cap,
+                  size_t* out_len, uint64_t* header_rev) {
+  auto parts = ((Store*)h)->GetPartitions(Bytes((const char*)start, slen),
+                                          Bytes((const char*)end, elen),
+                                          header_rev);
+  Writer w{out, cap};
+  w.u32((uint32_t)parts.size());
+  for (auto& p : parts) w.str(p);
+  *out_len = w.off;
+  if (w.overflow) return KB_ENOBUF;
+  return 0;
+}
+
 unsigned long long kb_current_rev(kb_store* h) {
   return ((Store*)h)->GetCurrentRevision();
 }

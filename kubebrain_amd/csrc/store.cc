@@ -279,6 +279,73 @@ void Store::pumpEvents() {
   pending_.clear();
 }
 
+int64_t Store::StreamOpen(const Bytes& start, const Bytes& end,
+                          uint64_t revision, uint64_t* read_rev, Status* st) {
+  // range.go:247-256: rev defaults to current; the scan checks the compact
+  // race once at open (scanner.go:594-626 via scan())
+  std::lock_guard<std::recursive_mutex> lk(mu_);
+  Status bv = validateBound(start);
+  if (bv == OK) bv = validateBound(end);
+  if (bv != OK) { *st = bv; return -1; }
+  uint64_t rev = revision == 0 ? committed_ : revision;
+  *read_rev = rev;
+  Status cst = checkCompactRace(rev);
+  if (cst != OK) { *st = cst; return -1; }
+  StreamState ss{start, end, rev, false};
+  int64_t sid = next_sid_++;
+  streams_[sid] = std::move(ss);
+  *st = OK;
+  return sid;
+}
+
+Status Store::StreamNext(int64_t sid, std::vector<KeyValue>* kvs) {
+  // rangeStreamBatch = 300 (scanner.go:45); batches carry More=true and the
+  // end marker is an empty batch (receiver.go:118-150, scanner.go:131-143)
+  std::lock_guard<std::recursive_mutex> lk(mu_);
+  kvs->clear();
+  auto it = streams_.find(sid);
+  if (it == streams_.end()) return INVALID_ARG;
+  StreamState& ss = it->second;
+  if (ss.done) { streams_.erase(it); return OK; }
+  std::string err;
+  if (!syncReads(&err)) return INTERNAL;
+  Status cst = checkCompactRace(ss.read_rev);
+  if (cst != OK) { streams_.erase(it); return cst; }
+  DevRangeQ q{};
+  memset(q.start, 0, KEYW);
+  memcpy(q.start, ss.frontier.data(), std::min(ss.frontier.size(), (size_t)KEYW));
+  pad96(ss.end, q.end);
+  q.read_rev = ss.read_rev;
+  q.cap = 300;
+  q.count_only = 0;
+  std::vector<kbslab::RangeResult> outs;
+  if (!slab_->RangeBatch({q}, true, &outs, &err)) return INTERNAL;
+  kbslab::RangeResult& r = outs[0];
+  if (r.overflow) return NOBUF;
+  for (auto& rec : r.recs) kvs->push_back(KeyValue{rec.key, rec.val, rec.rev});
+  if (r.written < 300) {
+    ss.done = true;  // next call returns the end marker
+  } else {
+    ss.frontier = r.recs.back().key + Bytes("\x01", 1);
+  }
+  if (kvs->empty()) streams_.erase(it);  // empty == end marker now
+  return OK;
+}
+
+void Store::StreamClose(int64_t sid) {
+  std::lock_guard<std::recursive_mutex> lk(mu_);
+  streams_.erase(sid);
+}
+
+std::vector<Bytes> Store::GetPartitions(const Bytes& start, const Bytes& end,
+                                        uint64_t* header_rev) {
+  // range.go:208-245 over a single partition (badger.go:52-54):
+  // PartitionKeys = [enc(start,0), enc(end,0)]
+  std::lock_guard<std::recursive_mutex> lk(mu_);
+  *header_rev = committed_;
+  return {EncodeObjectKey(start, 0), EncodeObjectKey(end, 0)};
+}
+
 int64_t Store::Watch(const Bytes& prefix, uint64_t revision, Status* st) {
   // watch.go:37-99
   std::lock_guard<std::recursive_mutex> lk(mu_);

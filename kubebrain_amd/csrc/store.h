@@ -73,6 +73,21 @@ class Store {
   uint64_t GetCurrentRevision();
   void SetCurrentRevision(uint64_t rev);
 
+  // ListByStream (range.go:247-256 + scanner.RangeStream, receiver.go:104-166):
+  // handle-based streaming — batches of 300 winners per Next() at a pinned
+  // readRev (the reference streams one snapshot; MVCC revision pinning gives
+  // the same results unless a compaction intervenes mid-stream, which then
+  // surfaces as KB_ECOMPACTED instead of stale rows).
+  int64_t StreamOpen(const Bytes& start, const Bytes& end, uint64_t revision,
+                     uint64_t* read_rev, Status* st);
+  // fills kvs with the next batch (<=300); empty batch == end marker (stream
+  // closes itself)
+  Status StreamNext(int64_t sid, std::vector<KeyValue>* kvs);
+  void StreamClose(int64_t sid);
+  // GetPartitions (range.go:208-245; single partition like badger.go:52-54)
+  std::vector<Bytes> GetPartitions(const Bytes& start, const Bytes& end,
+                                   uint64_t* header_rev);
+
   int64_t Watch(const Bytes& prefix, uint64_t revision, Status* st);
   std::vector<Event> WatchPoll(int64_t wid, Status* st);
   void WatchCancel(int64_t wid);
@@ -176,6 +191,9 @@ class Store {
   double sync_s_ = 0;
   int64_t sync_n_ = 0;
   std::string fatal_;  // first unrecoverable device error (e.g. slab full)
+  struct StreamState { Bytes frontier, end; uint64_t read_rev; bool done; };
+  std::unordered_map<int64_t, StreamState> streams_;
+  int64_t next_sid_ = 1;
 };
 
 Bytes EncodeObjectKey(const Bytes& userKey, uint64_t rev);  // coder/normal.go:42-50

@@ -569,6 +569,37 @@ uint64_t Backend::Compact(uint64_t revision, Status* st) {
   return revision;
 }
 
+std::vector<Backend::StreamBatch> Backend::ListByStream(
+    const Bytes& start, const Bytes& end, uint64_t revision, uint64_t* read_rev,
+    Status* st) {
+  // range.go:247-256: rev defaults to current; scanner.RangeStream streams the
+  // full scan in batches of 300 (receiver.go:118-150)
+  std::vector<StreamBatch> out;
+  uint64_t rev = revision == 0 ? committed_rev_ : revision;
+  *read_rev = rev;
+  Status cst = checkCompactRace(rev, false);  // scanner.go:594-626 via scan()
+  if (cst != OK) { *st = cst; return out; }
+  std::vector<KeyValue> kvs;
+  scanRange(EncodeObjectKey(start, 0), EncodeObjectKey(end, 0), rev, 0, false,
+            0, &kvs);
+  for (size_t i = 0; i < kvs.size(); i += 300) {
+    StreamBatch b;
+    for (size_t j = i; j < std::min(i + 300, kvs.size()); ++j)
+      b.kvs.push_back(kvs[j]);
+    out.push_back(std::move(b));
+  }
+  *st = OK;
+  return out;
+}
+
+std::vector<Bytes> Backend::GetPartitions(const Bytes& start, const Bytes& end,
+                                          uint64_t* header_rev) {
+  // range.go:208-245 over the badger single partition (badger.go:52-54):
+  // PartitionKeys = [enc(start,0), enc(end,0)]
+  *header_rev = committed_rev_;
+  return {EncodeObjectKey(start, 0), EncodeObjectKey(end, 0)};
+}
+
 int64_t Backend::Watch(const Bytes& prefix, uint64_t revision, Status* st) {
   // watch.go:37-99
   Watcher w;

@@ -137,6 +137,17 @@ class Backend {
   uint64_t GetCurrentRevision() const { return committed_rev_; }  // tso.GetRevision
   void SetCurrentRevision(uint64_t rev);             // tso Init semantics
 
+  // -- ListByStream (range.go:247-256 + scanner.RangeStream scanner.go:129-145,
+  // receiver.go:104-166): batches of 300 (More=true, header=readRev) then an
+  // end marker. Returned flattened here; the C ABI chunks it. --
+  struct StreamBatch { std::vector<KeyValue> kvs; };
+  std::vector<StreamBatch> ListByStream(const Bytes& start, const Bytes& end,
+                                        uint64_t revision, uint64_t* read_rev,
+                                        Status* st);
+  // -- GetPartitions (range.go:208-245; badger single partition badger.go:52-54) --
+  std::vector<Bytes> GetPartitions(const Bytes& start, const Bytes& end,
+                                   uint64_t* header_rev);
+
   // -- watch (watch.go:37-159 + watcherhub) --
   // Registers a watcher; catch-up events are queued immediately per
   // watch.go:52-99. Returns watcher id or error in *st.

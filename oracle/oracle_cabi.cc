@@ -206,6 +206,39 @@ int okb_event_log(void* h, uint8_t* out, size_t cap, size_t* out_len) {
   return 0;
 }
 
+// ListByStream: one call returns batch #idx (re-runs the scan; test-only
+// oracle, simplicity over speed). rc: 0 with n>0; 0 with n==0 => end marker.
+int okb_stream_batch(void* h, const uint8_t* start, size_t slen,
+                     const uint8_t* end, size_t elen, uint64_t rev,
+                     uint64_t batch_idx, uint8_t* out, size_t cap,
+                     size_t* out_len, uint64_t* read_rev) {
+  Status st;
+  auto batches = ((Backend*)h)->ListByStream(Bytes((const char*)start, slen),
+                                             Bytes((const char*)end, elen), rev,
+                                             read_rev, &st);
+  if (st != oracle::OK) return st;
+  Writer w{out, cap};
+  if (batch_idx < batches.size()) writeKvs(w, batches[batch_idx].kvs);
+  else w.u32(0);
+  *out_len = w.off;
+  if (w.overflow) return kEnoBuf;
+  return 0;
+}
+
+int okb_partitions(void* h, const uint8_t* start, size_t slen,
+                   const uint8_t* end, size_t elen, uint8_t* out, size_t cap,
+                   size_t* out_len, uint64_t* header_rev) {
+  auto parts = ((Backend*)h)->GetPartitions(Bytes((const char*)start, slen),
+                                            Bytes((const char*)end, elen),
+                                            header_rev);
+  Writer w{out, cap};
+  w.u32((uint32_t)parts.size());
+  for (auto& p : parts) w.str(p);
+  *out_len = w.off;
+  if (w.overflow) return kEnoBuf;
+  return 0;
+}
+
 // ---- coder / ring / util helpers for the golden-vector tests ----
 int okb_encode_key(const uint8_t* k, size_t klen, uint64_t rev, uint8_t* out,
                    size_t cap, size_t* olen) {

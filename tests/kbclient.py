@@ -208,6 +208,66 @@ class Store:
     def watch_cancel(self, wid: int):
         self._f("watch_cancel")(C.c_void_p(self.h), C.c_longlong(wid))
 
+    def stream(self, start: bytes, end: bytes, rev: int = 0):
+        """ListByStream: list of batches (each a list of Kv); oracle and
+        product expose different mechanics (batch-indexed vs handle), unified
+        here."""
+        batches = []
+        if self.p == "okb_":
+            idx = 0
+            while True:
+                out_len = C.c_size_t(); rrev = C.c_uint64()
+                rc = self._f("stream_batch")(C.c_void_p(self.h), start,
+                                             C.c_size_t(len(start)), end,
+                                             C.c_size_t(len(end)), C.c_uint64(rev),
+                                             C.c_uint64(idx), self.buf,
+                                             C.c_size_t(self.BUF), C.byref(out_len),
+                                             C.byref(rrev))
+                if rc != OK:
+                    return rc, batches
+                kvs = _parse_kvs(self.buf[:out_len.value])
+                if not kvs:
+                    return OK, batches
+                batches.append(kvs)
+                idx += 1
+        else:
+            st = C.c_int(); rrev = C.c_uint64()
+            f = self._f("stream_open")
+            f.restype = C.c_longlong
+            sid = f(C.c_void_p(self.h), start, C.c_size_t(len(start)), end,
+                    C.c_size_t(len(end)), C.c_uint64(rev), C.byref(rrev),
+                    C.byref(st))
+            if st.value != OK:
+                return st.value, batches
+            while True:
+                out_len = C.c_size_t()
+                rc = self._f("stream_next")(C.c_void_p(self.h), C.c_longlong(sid),
+                                            self.buf, C.c_size_t(self.BUF),
+                                            C.byref(out_len))
+                if rc != OK:
+                    return rc, batches
+                kvs = _parse_kvs(self.buf[:out_len.value])
+                if not kvs:
+                    return OK, batches
+                batches.append(kvs)
+
+    def partitions(self, start: bytes, end: bytes):
+        out_len = C.c_size_t(); hr = C.c_uint64()
+        rc = self._f("partitions")(C.c_void_p(self.h), start,
+                                   C.c_size_t(len(start)), end,
+                                   C.c_size_t(len(end)), self.buf,
+                                   C.c_size_t(self.BUF), C.byref(out_len),
+                                   C.byref(hr))
+        assert rc == 0
+        buf = self.buf[:out_len.value]
+        (n,) = struct.unpack_from("<I", buf, 0)
+        off = 4
+        parts = []
+        for _ in range(n):
+            (ln,) = struct.unpack_from("<I", buf, off); off += 4
+            parts.append(buf[off:off + ln]); off += ln
+        return hr.value, parts
+
     def current_rev(self) -> int:
         f = self._f("current_rev")
         f.restype = C.c_uint64

@@ -97,6 +97,20 @@ class Dual:
         for i in range(len(self.watches)):
             self.poll(i)
 
+    def stream(self, start, end, rev=0):
+        so, bo = self.o.stream(start, end, rev)
+        sp, bp = self.p.stream(start, end, rev)
+        assert so == sp, (so, sp)
+        assert len(bo) == len(bp), (len(bo), len(bp))
+        for i, (x, y) in enumerate(zip(bo, bp)):
+            assert _fmt_kvs(x) == _fmt_kvs(y), i
+        return bp
+
+    def partitions(self, start, end):
+        po, pp = self.o.partitions(start, end), self.p.partitions(start, end)
+        assert po == pp, (po, pp)
+        return pp
+
     def diff_dump(self):
         do, dp = self.o.dump(), self.p.dump()
         assert len(do) == len(dp), (len(do), len(dp))

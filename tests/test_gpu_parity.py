@@ -245,3 +245,25 @@ def test_random_mixed_workload(dual):
     dual.poll(w)
     dual.diff_dump()
     dual.diff_event_log()
+
+
+def test_stream_and_partitions(dual):
+    # ListByStream (range.go:247-256, batches of 300) + GetPartitions parity
+    ns = NS[3]
+    for i in range(730):
+        dual.create(ns + b"/obj-%05d" % i, b"s%d" % i)
+    for i in range(0, 730, 91):
+        dual.delete(ns + b"/obj-%05d" % i, 0)
+    batches = dual.stream(ns + b"/", ns + b"0", 0)
+    assert [len(b) for b in batches[:-1]] == [300] * (len(batches) - 1)
+    assert sum(len(b) for b in batches) == 730 - len(range(0, 730, 91))
+    dual.stream(ns + b"/", ns + b"0", dual.p.current_rev() - 200)
+    dual.stream(b"/registry/zzz/", b"/registry/zzz0", 0)  # empty stream
+    dual.partitions(ns + b"/", ns + b"0")
+    dual.partitions(b"/registry/", b"/registry0")
+    # compacted stream open fails identically
+    mid = dual.p.current_rev()
+    dual.compact(mid)
+    so, _ = dual.o.stream(ns + b"/", ns + b"0", mid - 1)
+    sp, _ = dual.p.stream(ns + b"/", ns + b"0", mid - 1)
+    assert so == sp and so != 0
