@@ -171,66 +171,6 @@ __device__ int64_t d_lb_fenced(const uint8_t* keys, const uint64_t* rev,
   return rlo;
 }
 
-// ---- the north-star kernel: batched MVCC range scan --------------------
-// One workgroup per query. Winner predicate (scanner.go:389-516, DESIGN.md
-// §3.1): rev>0 ∧ rev<=R ∧ ¬tomb ∧ (¬same_next ∨ next.rev>R). Reads 16B/row
-// (meta+rev); key/value columns are untouched until gather.
-__global__ void k_range_scan(const uint8_t* __restrict__ keys,
-                             const uint64_t* __restrict__ meta,
-                             const uint64_t* __restrict__ rev, int64_t n,
-                             const DevRangeQ* __restrict__ qs, int nq,
-                             int64_t max_cap, uint64_t* __restrict__ rows_out,
-                             int64_t* __restrict__ found_out,
-                             int64_t* __restrict__ total_out,
-                             unsigned long long* __restrict__ scanned_out) {
-  int q = blockIdx.x;
-  if (q >= nq) return;
-  __shared__ int64_t lo_s, hi_s;
-  __shared__ int64_t tot_s;
-  __shared__ int wave_cnt[4];
-  const DevRangeQ& Q = qs[q];
-  if (threadIdx.x == 0) { lo_s = d_lower_bound(keys, rev, n, Q.start, 0); tot_s = 0; }
-  if (threadIdx.x == 64) hi_s = d_lower_bound(keys, rev, n, Q.end, 0);
-  __syncthreads();
-  const int64_t lo = lo_s, hi = hi_s;
-  const int64_t cap = Q.cap > 0 ? Q.cap : INT64_MAX;
-  uint64_t* out = rows_out + (int64_t)q * max_cap;
-  int lane = threadIdx.x & 63, w = threadIdx.x >> 6;
-  int64_t scanned = 0;
-  for (int64_t t = lo; t < hi; t += blockDim.x) {
-    int64_t i = t + threadIdx.x;
-    bool win = false;
-    if (i < hi) {
-      uint64_t r = rev[i];
-      uint64_t m = meta[i];
-      if (r > 0 && r <= Q.read_rev && !(m & M_TOMB))
-        win = !(m & M_SAME_NEXT) || rev[i + 1] > Q.read_rev;
-    }
-    uint64_t b = __ballot(win);
-    if (lane == 0) wave_cnt[w] = __popcll(b);
-    __syncthreads();
-    int64_t waveoff = 0;
-    for (int k = 0; k < w; ++k) waveoff += wave_cnt[k];
-    int tile_total = wave_cnt[0] + wave_cnt[1] + wave_cnt[2] + wave_cnt[3];
-    if (win && !Q.count_only) {
-      int64_t idx = tot_s + waveoff + __popcll(b & ((1ull << lane) - 1));
-      if (idx < cap && idx < max_cap) out[idx] = (uint64_t)i;
-    }
-    __syncthreads();
-    if (threadIdx.x == 0) tot_s += tile_total;
-    scanned += min((int64_t)blockDim.x, hi - t);
-    __syncthreads();
-    if (tot_s >= cap) break;  // limit satisfied (receiver.needMore)
-  }
-  if (threadIdx.x == 0) {
-    total_out[q] = tot_s;
-    int64_t written = tot_s < cap ? tot_s : cap;
-    if (written > max_cap) written = max_cap;
-    found_out[q] = Q.count_only ? 0 : written;
-    atomicAdd(scanned_out, (unsigned long long)scanned);
-  }
-}
-
 // ---- two-run variant: base run + sorted delta run merged at scan time ---
 // The delta run holds rows strictly newer than the base run's rows of the
 // same key, so the global winner of a key is the delta
@@ -539,52 +479,6 @@ __global__ void k_pack(const uint8_t* __restrict__ gbuf, int64_t qcap,
   uint4* dst = (uint4*)(out + goffs[q]);
   int64_t words = bytes / 16;  // records are 16B-aligned (goffs too)
   for (int64_t w = threadIdx.x; w < words; w += blockDim.x) dst[w] = src[w];
-}
-
-// ---- batched MVCC point read (range.go:91-121) --------------------------
-__global__ void k_get(const uint8_t* __restrict__ keys,
-                      const uint64_t* __restrict__ meta,
-                      const uint64_t* __restrict__ rev,
-                      const uint64_t* __restrict__ vo,
-                      const uint8_t* __restrict__ heap, int64_t n,
-                      const DevGetQ* __restrict__ qs, int nq,
-                      uint8_t* __restrict__ out, int64_t slot,
-                      uint64_t* __restrict__ orev, uint64_t* __restrict__ ometa,
-                      int32_t* __restrict__ ofound, int32_t* __restrict__ oovf) {
-  int q = blockIdx.x * (blockDim.x / 64) + (threadIdx.x >> 6);
-  int lane = threadIdx.x & 63;
-  if (q >= nq) return;
-  const DevGetQ& Q = qs[q];
-  int64_t row = -1;
-  if (lane == 0) {
-    // reverse iter [key@R -> key@0): largest (key,rev) <= (key,R) with rev>=1
-    int64_t ub = d_upper_bound(keys, rev, n, Q.key, Q.read_rev);
-    if (ub > 0) {
-      int64_t c = ub - 1;
-      if (rev[c] >= 1 && keycmp96(keys + c * KEYW, Q.key) == 0) row = c;
-    }
-  }
-  row = __shfl(row, 0);
-  if (row < 0) {
-    if (lane == 0) { ofound[q] = 0; oovf[q] = 0; }
-    return;
-  }
-  uint64_t m = meta[row];
-  uint32_t vlen = meta_vlen(m);
-  if (lane == 0) {
-    ofound[q] = 1;
-    orev[q] = rev[row];
-    ometa[q] = m;
-    oovf[q] = vlen > slot ? 1 : 0;
-  }
-  if (vlen > slot) return;
-  const uint8_t* vs = heap + vo[row];
-  uint8_t* vd = out + (int64_t)q * slot;
-  uint32_t words = vlen >> 2;
-  for (uint32_t b = lane; b < words; b += 64)
-    ((uint32_t*)vd)[b] = ((const uint32_t*)vs)[b];
-  if (lane == 0)
-    for (uint32_t b = words * 4; b < vlen; ++b) vd[b] = vs[b];
 }
 
 // ---- generic exclusive scan (u64), 256-wide blocks ----------------------
