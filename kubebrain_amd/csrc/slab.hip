@@ -456,7 +456,11 @@ __global__ void k_gather(const uint8_t* __restrict__ bkeys,
     }
     const uint8_t* ks = (isd ? dkeys : bkeys) + row * KEYW;
     uint8_t* kd = dst + 16;
-    for (uint32_t b = gl; b < klen; b += 16) kd[b] = ks[b];
+    // key rows are 96B and 16B-aligned per row? row*96 is 16B-aligned only for
+    // even rows; copy as u32 words (4-aligned always), padded width
+    uint32_t kw = (klen + 15) & ~15u;
+    for (uint32_t b = gl; b < kw / 4; b += 16)
+      ((uint32_t*)kd)[b] = ((const uint32_t*)ks)[b];
     const uint8_t* vs = heap + (isd ? dvo : bvo)[row];
     uint8_t* vd = dst + 16 + ((klen + 15) & ~15u);
     uint32_t w16 = vlen >> 4;

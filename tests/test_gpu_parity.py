@@ -234,10 +234,15 @@ def test_random_mixed_workload(dual):
             hi = ns + b"0"
             rev = 0 if rng.random() < 0.5 else max(1, dual.p.current_rev() - rng.randrange(100))
             dual.list(lo, hi, rev, rng.choice([0, 1, 5, 17, 60]))
-        elif op < 0.95:
+        elif op < 0.93:
             dual.get(key, 0 if rng.random() < 0.5 else max(1, dual.p.current_rev() - rng.randrange(50)))
-        else:
+        elif op < 0.96:
             dual.count(ns + b"/", ns + b"0")
+        elif op < 0.98:
+            dual.stream(ns + b"/", ns + b"0",
+                        0 if rng.random() < 0.5 else max(1, dual.p.current_rev() - rng.randrange(80)))
+        else:
+            dual.partitions(ns + b"/", ns + b"0")
         if step in (400, 900):
             dual.compact(max(1, dual.p.current_rev() - 50))
         if step % 300 == 299:
