@@ -33,7 +33,6 @@ namespace kbslab {
     }                                                                      \
   } while (0)
 
-static inline int64_t pad8(int64_t x) { return (x + 7) & ~7ll; }
 static inline int64_t ceil_div(int64_t a, int64_t b) { return (a + b - 1) / b; }
 
 // ---------------------------------------------------------------- device ---
@@ -190,33 +189,63 @@ __device__ int64_t scan_run_winners(
   if (threadIdx.x == 0) *cnt_s = 0;
   __syncthreads();
   int64_t scanned = 0;
-  for (int64_t t = lo; t < hi; t += blockDim.x) {
-    int64_t i = t + threadIdx.x;
-    bool win = false;
-    if (i < hi) {
-      uint64_t r = rev[i], m = meta[i];
-      if (r > 0 && r <= R && !(m & M_TOMB))
-        win = !(m & M_SAME_NEXT) || rev[i + 1] > R;
-      if (win && skeys) {
-        const uint8_t* kk = keys + i * KEYW;
-        int64_t lb = d_lb_range(skeys, srev, slo, shi, kk, 1);
-        if (lb < shi && srev[lb] <= R && keycmp96(skeys + lb * KEYW, kk) == 0)
-          win = false;  // a newer (delta) row of this key wins instead
+  // 4 consecutive rows per thread: 1024-row tiles quarter the number of
+  // latency-bound sync rounds per query vs 256-row tiles
+  const int64_t TILE = (int64_t)blockDim.x * 4;
+  for (int64_t t = lo; t < hi; t += TILE) {
+    int64_t i0 = t + (int64_t)threadIdx.x * 4;
+    uint32_t wins = 0, mycnt = 0;
+    if (i0 < hi) {
+      int nk = (int)min((int64_t)4, hi - i0);
+      uint64_t r[5], m[4];
+#pragma unroll
+      for (int k = 0; k < 4; ++k) {
+        r[k] = k < nk ? rev[i0 + k] : 0;
+        m[k] = k < nk ? meta[i0 + k] : 0;
+      }
+      // rev[i0+4] only needed when row i0+3's same_next is set (then in-bounds)
+      r[4] = (nk == 4 && (m[3] & M_SAME_NEXT)) ? rev[i0 + 4] : 0;
+#pragma unroll
+      for (int k = 0; k < 4; ++k) {
+        if (k >= nk) break;
+        bool win = false;
+        if (r[k] > 0 && r[k] <= R && !(m[k] & M_TOMB))
+          win = !(m[k] & M_SAME_NEXT) || r[k + 1] > R;
+        if (win && skeys) {
+          const uint8_t* kk = keys + (i0 + k) * KEYW;
+          int64_t lb = d_lb_range(skeys, srev, slo, shi, kk, 1);
+          if (lb < shi && srev[lb] <= R && keycmp96(skeys + lb * KEYW, kk) == 0)
+            win = false;  // a newer (delta) row of this key wins instead
+        }
+        if (win) { wins |= 1u << k; mycnt++; }
       }
     }
-    uint64_t b = __ballot(win);
-    if (lane == 0) wave_cnt[w] = __popcll(b);
+    // wave-inclusive scan of per-thread winner counts (ordered append)
+    uint32_t incl = mycnt;
+#pragma unroll
+    for (int o = 1; o < 64; o <<= 1) {
+      uint32_t v = __shfl_up(incl, o);
+      if (lane >= o) incl += v;
+    }
+    uint32_t excl = incl - mycnt;
+    if (lane == 63) wave_cnt[w] = (int)incl;
     __syncthreads();
     int64_t waveoff = 0;
     for (int k = 0; k < w; ++k) waveoff += wave_cnt[k];
     int tile_total = wave_cnt[0] + wave_cnt[1] + wave_cnt[2] + wave_cnt[3];
-    if (win && out) {
-      int64_t idx = *cnt_s + waveoff + __popcll(b & ((1ull << lane) - 1));
-      if (idx < cap && idx < out_cap) out[idx] = (uint64_t)i | tagbit;
+    if (out && wins) {
+      int64_t idx = *cnt_s + waveoff + excl;
+#pragma unroll
+      for (int k = 0; k < 4; ++k) {
+        if (wins & (1u << k)) {
+          if (idx < cap && idx < out_cap) out[idx] = (uint64_t)(i0 + k) | tagbit;
+          idx++;
+        }
+      }
     }
     __syncthreads();
     if (threadIdx.x == 0) *cnt_s += tile_total;
-    scanned += min((int64_t)blockDim.x, hi - t);
+    scanned += min(TILE, hi - t);
     __syncthreads();
     if (*cnt_s >= cap) break;
   }
@@ -1161,7 +1190,6 @@ bool Slab::RangeBatchEx(const std::vector<DevRangeQ>& qs, bool d2h, bool parse,
   perf.gather_launches++;
   perf.rows_scanned += (int64_t)scanned;
   perf.bytes_gathered += (int64_t)bytes;
-  int64_t tot_bytes = 0;
   for (int q = 0; q < nq; ++q) {
     RangeResult& r = (*outs)[q];
     r.written = found[q];
@@ -1169,7 +1197,6 @@ bool Slab::RangeBatchEx(const std::vector<DevRangeQ>& qs, bool d2h, bool parse,
     r.bytes = gbytes[q];
     r.overflow = ovf[q] != 0;
     perf.winners += found[q];
-    tot_bytes += ovf[q] ? 0 : gbytes[q];
   }
   if (!d2h) return true;
   // pack + one D2H + parse

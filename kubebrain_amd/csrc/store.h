@@ -184,7 +184,6 @@ class Store {
   static constexpr size_t kWatchQueueCap = 10000ull * 300;  // watcherhub.go:30 × eventBatchSize
 
   // host-side perf
-  double host_write_s_ = 0, host_merge_s_ = 0;
   int64_t ops_create_ = 0, ops_update_ = 0, ops_delete_ = 0, ops_range_ = 0;
   int64_t delivered_ = 0;  // watch events enqueued to watchers (fan-out)
   bool keep_event_log_ = true;
