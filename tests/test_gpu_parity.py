@@ -207,8 +207,9 @@ def test_watch_parity(dual):
     dual.diff_event_log()
 
 
-def test_random_mixed_workload(dual):
-    rng = random.Random(0x6B62)
+@pytest.mark.parametrize("seed", [0x6B62, 0xBEEF, 17])
+def test_random_mixed_workload(dual, seed):
+    rng = random.Random(seed)
     live = {}
     w = dual.watch(b"/registry/", 0)
     for step in range(1200):
