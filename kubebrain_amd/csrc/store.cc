@@ -676,17 +676,26 @@ RangeResponse Store::List(const Bytes& start, const Bytes& end,
   bool dev_more = true;
   Bytes dev_frontier = start;
   auto need = [&]() { return lim <= 0 || (int64_t)kvs.size() < lim; };
+  int64_t chunk_cap = 0;  // 0 = unbounded; halved on arena overflow
   while (need() && dev_more) {
     DevRangeQ q{};
     memset(q.start, 0, KEYW);
     memcpy(q.start, dev_frontier.data(), std::min(dev_frontier.size(), (size_t)KEYW));
     pad96(end, q.end);
     q.read_rev = reqRevision;
-    q.cap = lim > 0 ? lim - (int64_t)kvs.size() : 0;
+    q.cap = lim > 0 ? lim - (int64_t)kvs.size() : chunk_cap;
+    if (chunk_cap > 0 && (q.cap <= 0 || q.cap > chunk_cap)) q.cap = chunk_cap;
     q.count_only = 0;
     if (!slab_->RangeBatch({q}, true, &outs, &err)) { *st = INTERNAL; return resp; }
     kbslab::RangeResult& r = outs[0];
-    if (r.overflow) { *st = NOBUF; return resp; }
+    if (r.overflow) {
+      // results exceed the device arena: halve the chunk and continue (large
+      // values); a single record larger than the arena is a real limit
+      int64_t cur = q.cap > 0 ? q.cap : (r.written > 0 ? r.written : 4096);
+      chunk_cap = cur / 2;
+      if (chunk_cap < 1) { *st = NOBUF; return resp; }
+      continue;
+    }
     for (auto& rec : r.recs) {
       if (!need()) break;
       kvs.push_back(KeyValue{rec.key, rec.val, rec.rev});

@@ -78,3 +78,27 @@ def test_bulk_scale_parity():
         d.diff_dump()
     finally:
         d.close()
+
+
+def test_big_values():
+    """Large values: heap, gather-arena overflow handling (chunk halving),
+    point reads."""
+    d = parity.Dual()
+    try:
+        rng = np.random.default_rng(3)
+        big = rng.integers(0, 256, size=2 << 20, dtype=np.uint8).tobytes()  # 2MB
+        keys = [b"/registry/big/obj-%03d" % i for i in range(40)]
+        for i, k in enumerate(keys):
+            d.create(k, big[: (1 << 20) + i * 1024])
+        # full list: 40 x ~1MB results exceed the 64MB test arena -> the
+        # product must chunk-halve, not fail
+        d.list(b"/registry/big/", b"/registry/big0", 0, 0)
+        d.list(b"/registry/big/", b"/registry/big0", 0, 7)
+        for i in (0, 17, 39):
+            d.get(keys[i], 0)
+        d.count(b"/registry/big/", b"/registry/big0")
+        d.delete(keys[5], 0)
+        d.compact(0)
+        d.list(b"/registry/big/", b"/registry/big0", 0, 0)
+    finally:
+        d.close()
