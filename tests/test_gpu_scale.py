@@ -85,12 +85,16 @@ def test_big_values():
     point reads."""
     d = parity.Dual()
     try:
+        import ctypes as C
+        for st in (d.o, d.p):  # larger wire buffers for ~85MB of results
+            st.BUF = 192 << 20
+            st.buf = C.create_string_buffer(st.BUF)
         rng = np.random.default_rng(3)
         big = rng.integers(0, 256, size=2 << 20, dtype=np.uint8).tobytes()  # 2MB
-        keys = [b"/registry/big/obj-%03d" % i for i in range(40)]
+        keys = [b"/registry/big/obj-%03d" % i for i in range(80)]
         for i, k in enumerate(keys):
             d.create(k, big[: (1 << 20) + i * 1024])
-        # full list: 40 x ~1MB results exceed the 64MB test arena -> the
+        # full list: 80 x ~1MB results exceed the 64MB test arena -> the
         # product must chunk-halve, not fail
         d.list(b"/registry/big/", b"/registry/big0", 0, 0)
         d.list(b"/registry/big/", b"/registry/big0", 0, 7)
@@ -98,6 +102,7 @@ def test_big_values():
             d.get(keys[i], 0)
         d.count(b"/registry/big/", b"/registry/big0")
         d.delete(keys[5], 0)
+        d.get(keys[5], 0)
         d.compact(0)
         d.list(b"/registry/big/", b"/registry/big0", 0, 0)
     finally:
