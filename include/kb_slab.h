@@ -138,6 +138,12 @@ int kb_bench_range(kb_store*, const uint8_t* qbuf, size_t nq, int d2h,
  * u64 prev_rev; u32 vlen; key; val}; out_revs[i] = new revision or 0 on CAS
  * failure */
 int kb_bench_txn(kb_store*, const uint8_t* tbuf, size_t n, uint64_t* out_revs);
+/* one bench step: Range batch launched async + txn batch overlapped on the
+ * host while the kernels are in flight (results unchanged: kernels snapshot
+ * device state at launch; writes stage host-side until the next sync) */
+int kb_bench_step(kb_store*, const uint8_t* qbuf, size_t nq,
+                  const uint8_t* tbuf, size_t ntx, int d2h, uint64_t* out_revs,
+                  unsigned long long* total, double* secs);
 /* batched deletes: dbuf = n × {u32 klen; u64 prev_rev; key} */
 int kb_bench_del(kb_store*, const uint8_t* dbuf, size_t n, uint64_t* out_revs);
 /* fast bulk insert == n serial Creates of fresh keys (see okb_bulk_create) */

@@ -304,6 +304,18 @@ int kb_bench_txn(kb_store* h, const uint8_t* tbuf, size_t n, uint64_t* out_revs)
   return 0;
 }
 
+int kb_bench_step(kb_store* h, const uint8_t* qbuf, size_t nq,
+                  const uint8_t* tbuf, size_t ntx, int d2h, uint64_t* out_revs,
+                  unsigned long long* total, double* secs) {
+  std::string err;
+  if (!((Store*)h)->BenchStep(qbuf, nq, tbuf, ntx, d2h != 0, out_revs, total,
+                              secs, &err)) {
+    set_err(KB_EINTERNAL, err);
+    return KB_EINTERNAL;
+  }
+  return 0;
+}
+
 int kb_bench_del(kb_store* h, const uint8_t* dbuf, size_t n, uint64_t* out_revs) {
   std::string err;
   if (!((Store*)h)->BenchDel(dbuf, n, out_revs, &err)) {

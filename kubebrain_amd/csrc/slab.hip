@@ -1133,13 +1133,11 @@ bool Slab::RangeBatch(const std::vector<DevRangeQ>& qs, bool d2h,
   return RangeBatchEx(qs, d2h, true, outs, err);
 }
 
-bool Slab::RangeBatchEx(const std::vector<DevRangeQ>& qs, bool d2h, bool parse,
-                        std::vector<RangeResult>* outs, std::string* err) {
+bool Slab::RangeBatchStart(const std::vector<DevRangeQ>& qs, std::string* err) {
   Impl* I = p;
   int nq = (int)qs.size();
   if (nq == 0) return true;
   if (nq > I->max_q) { if (err) *err = "too many queries per batch (KB_MAX_Q)"; return false; }
-  outs->assign(nq, RangeResult());
   int64_t qcap = I->arena_bytes / nq;
   qcap &= ~15ll;
   HIP_CHECK(hipMemcpyAsync(I->d_qs, qs.data(), sizeof(DevRangeQ) * nq,
@@ -1160,6 +1158,22 @@ bool Slab::RangeBatchEx(const std::vector<DevRangeQ>& qs, bool d2h, bool parse,
                      I->d_found, nq, I->d_gbuf, qcap, I->d_offs, I->d_gbytes,
                      I->d_ovf, I->d_bytes);
   HIP_CHECK(hipEventRecord(I->ev2, I->stream));
+  return true;
+}
+
+bool Slab::RangeBatchEx(const std::vector<DevRangeQ>& qs, bool d2h, bool parse,
+                        std::vector<RangeResult>* outs, std::string* err) {
+  if (!RangeBatchStart(qs, err)) return false;
+  return RangeBatchFinish((int)qs.size(), d2h, parse, outs, err);
+}
+
+bool Slab::RangeBatchFinish(int nq, bool d2h, bool parse,
+                            std::vector<RangeResult>* outs, std::string* err) {
+  Impl* I = p;
+  outs->assign(nq, RangeResult());
+  if (nq == 0) return true;
+  int64_t qcap = I->arena_bytes / nq;
+  qcap &= ~15ll;
   // small result metadata: d_found/d_total/d_gbytes are CONTIGUOUS slices of
   // one allocation (see Create), so one D2H covers them; ovf+counters ride
   // two more copies

@@ -134,6 +134,11 @@ class Slab {
   // parse=false: D2H into pinned memory without materializing records
   bool RangeBatchEx(const std::vector<DevRangeQ>& qs, bool d2h, bool parse,
                     std::vector<RangeResult>* outs, std::string* err);
+  // async split: Start launches the scan+gather without syncing, so host
+  // work (e.g. the txn leg) overlaps the in-flight kernels; Finish collects
+  bool RangeBatchStart(const std::vector<DevRangeQ>& qs, std::string* err);
+  bool RangeBatchFinish(int nq, bool d2h, bool parse,
+                        std::vector<RangeResult>* outs, std::string* err);
 
   // batched MVCC point read (range.go:91-121 reverse-iter semantics)
   bool GetBatch(const std::vector<DevGetQ>& qs, std::vector<GetResult>* outs,

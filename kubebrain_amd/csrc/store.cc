@@ -928,6 +928,57 @@ bool Store::BenchRange(const uint8_t* qbuf, size_t nq, bool d2h,
   return true;
 }
 
+bool Store::BenchStep(const uint8_t* qbuf, size_t nq, const uint8_t* tbuf,
+                      size_t ntx, bool d2h, uint64_t* out_revs,
+                      unsigned long long* total, double* secs,
+                      std::string* err) {
+  std::lock_guard<std::recursive_mutex> lk(mu_);
+  if (!syncReads(err)) return false;
+  std::vector<DevRangeQ> qall(nq);
+  std::vector<int64_t> limits(nq);
+  {
+    const uint8_t* p = qbuf;
+    for (size_t i = 0; i < nq; ++i) {
+      uint32_t slen, elen;
+      uint64_t rev, limit;
+      memcpy(&slen, p, 4); p += 4;
+      memcpy(&elen, p, 4); p += 4;
+      memcpy(&rev, p, 8); p += 8;
+      memcpy(&limit, p, 8); p += 8;
+      DevRangeQ& q = qall[i];
+      memset(q.start, 0, KEYW);
+      memset(q.end, 0, KEYW);
+      memcpy(q.start, p, std::min((size_t)slen, (size_t)KEYW)); p += slen;
+      memcpy(q.end, p, std::min((size_t)elen, (size_t)KEYW)); p += elen;
+      q.read_rev = rev == 0 ? committed_ : rev;
+      q.cap = (int64_t)limit > 0 ? (int64_t)limit + 1 : 0;
+      q.count_only = 0;
+      q._pad = 0;
+      limits[i] = (int64_t)limit;
+    }
+  }
+  auto t0 = std::chrono::steady_clock::now();
+  if (!slab_->RangeBatchStart(qall, err)) return false;
+  // overlap: the kernels read device state snapshotted at launch; the txn
+  // batch below touches only host staging until the next syncReads, so the
+  // in-flight scan's results are unchanged (MVCC readRev pinning)
+  if (ntx > 0) {
+    if (!BenchTxn(tbuf, ntx, out_revs, err)) return false;
+  }
+  std::vector<kbslab::RangeResult> outs;
+  if (!slab_->RangeBatchFinish((int)nq, d2h, false, &outs, err)) return false;
+  *secs = std::chrono::duration<double>(std::chrono::steady_clock::now() - t0).count();
+  unsigned long long tot = 0;
+  for (size_t j = 0; j < nq; ++j) {
+    int64_t lim = limits[j];
+    int64_t w = outs[j].written;
+    tot += (unsigned long long)(lim > 0 && w > lim ? lim : w);
+  }
+  *total = tot;
+  ops_range_ += (int64_t)nq;
+  return true;
+}
+
 bool Store::BenchTxn(const uint8_t* tbuf, size_t n, uint64_t* out_revs,
                      std::string* err) {
   // packed records: {u32 klen; u64 prev_rev; u32 vlen; key; val} x n

@@ -109,6 +109,11 @@ class Store {
                 std::string* err);
   bool BenchDel(const uint8_t* dbuf, size_t n, uint64_t* out_revs,
                 std::string* err);
+  // one bench step: launch the range batch async, run the txn batch on the
+  // host while the kernels are in flight, then collect
+  bool BenchStep(const uint8_t* qbuf, size_t nq, const uint8_t* tbuf,
+                 size_t ntx, bool d2h, uint64_t* out_revs,
+                 unsigned long long* total, double* secs, std::string* err);
   std::string PerfJson();
   void PerfReset();
 

@@ -61,6 +61,15 @@ class MockStore:
             out_ptr[i] = self.rev
         return 0
 
+    def f_bench_step(self, h, qblob, nq, tblob, ntx, d2h, out_ptr, total_ref,
+                     secs_ref):
+        for i in range(ntx.value):
+            self.rev += 1
+            out_ptr[i] = self.rev
+        total_ref._obj.value = nq.value * 490
+        secs_ref._obj.value = 0.001
+        return 0
+
     def f_bench_range(self, h, blob, nq, d2h, total_ref, secs_ref):
         total_ref._obj.value = nq.value * 490
         secs_ref._obj.value = 0.001
