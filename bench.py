@@ -357,27 +357,58 @@ def main():
     my_ns_list = sorted(my_ns)
     qrng = np.random.default_rng(SEED + 100 + rank)
 
-    # pre-generate the query stream (client-side work, outside the measured
-    # server path — the reference bench's 300 clients generate requests too)
+    # pre-generate the query stream AND the txn batches (client-side work,
+    # outside the measured server path — the reference bench's 300 clients
+    # generate requests too). Txn batches use keys disjoint across steps, so
+    # each key's prev-rev is known at generation time; returned revisions
+    # update the client map after each step.
     nq = int(OPS_PER_STEP * RANGE_SHARE)  # 900 ranges
     ntx = OPS_PER_STEP - nq               # 100 txns
     n_pre = args.warmup + args.steps + max(2, args.steps // 8) + 2
     cur = store.current_rev()
     pre_blobs = [pack_queries(gen_step_queries(qrng, my_ns_list, cur, nq))
                  for _ in range(n_pre)]
+    vbuf = qrng.integers(0, 256, size=VAL_LEN, dtype=np.uint8).tobytes()
+    tx_keys, tx_blobs, seen_tx = [], [], set()
+    for _ in range(n_pre):
+        ks = []
+        while len(ks) < ntx:
+            k = live[int(qrng.integers(len(live)))]
+            if k in seen_tx or revs.get(k) is None:
+                continue
+            seen_tx.add(k)
+            ks.append(k)
+        parts = []
+        for k in ks:
+            parts.append(struct.pack("<IQI", len(k), revs[k], VAL_LEN))
+            parts.append(k)
+            parts.append(vbuf)
+        tx_keys.append(ks)
+        tx_blobs.append(b"".join(parts))
+    fstep = store._f("bench_step")
+    tx_out = np.empty(ntx, dtype=np.uint64)
+    tx_out_ptr = tx_out.ctypes.data_as(ctypes.POINTER(ctypes.c_uint64))
     step_i = [0]
 
     t_split = {"range_s": 0.0, "txn_s": 0.0}
 
     def one_step(d2h=False):
-        blob = pre_blobs[step_i[0] % n_pre]
+        # one call: range batch launched async, txn batch overlapped on the
+        # host while the kernels are in flight (kb_bench_step)
+        i = step_i[0] % n_pre
         step_i[0] += 1
-        tot, rsecs = bench_range_call(store, blob, nq, d2h)
-        t_split["range_s"] += rsecs
-        t0 = time.time()
-        run_txns(store, live, revs, qrng, ntx)
-        t_split["txn_s"] += time.time() - t0
-        return tot
+        total = ctypes.c_ulonglong()
+        secs = ctypes.c_double()
+        rc = fstep(ctypes.c_void_p(store.h), pre_blobs[i], ctypes.c_size_t(nq),
+                   tx_blobs[i], ctypes.c_size_t(ntx),
+                   ctypes.c_int(1 if d2h else 0), tx_out_ptr,
+                   ctypes.byref(total), ctypes.byref(secs))
+        assert rc == 0
+        t_split["range_s"] += secs.value
+        for k, nr in zip(tx_keys[i], tx_out):
+            if nr != 0:
+                revs[k] = int(nr)
+        return total.value
 
     def barrier():
         if dist:

@@ -504,9 +504,11 @@ __global__ void k_gather(const uint8_t* __restrict__ bkeys,
 __global__ void k_pack(const uint8_t* __restrict__ gbuf, int64_t qcap,
                        const int64_t* __restrict__ gbytes,
                        const int64_t* __restrict__ goffs,
+                       const int32_t* __restrict__ overflow,
                        uint8_t* __restrict__ out, int nq) {
   int q = blockIdx.x;
   if (q >= nq) return;
+  if (overflow[q]) return;  // nothing was gathered for this query
   int64_t bytes = gbytes[q];
   const uint4* src = (const uint4*)(gbuf + (int64_t)q * qcap);
   uint4* dst = (uint4*)(out + goffs[q]);
@@ -1222,7 +1224,7 @@ bool Slab::RangeBatchFinish(int nq, bool d2h, bool parse,
   HIP_CHECK(hipMemcpyAsync(I->d_goffs, goffs.data(), (nq + 1) * 8,
                            hipMemcpyHostToDevice, I->stream));
   hipLaunchKernelGGL(k_pack, dim3(nq), dim3(256), 0, I->stream, I->d_gbuf, qcap,
-                     I->d_gbytes, I->d_goffs, I->d_pack, nq);
+                     I->d_gbytes, I->d_goffs, I->d_ovf, I->d_pack, nq);
   if (!I->ensure_hpack(acc, err)) return false;
   if (acc > 0)
     HIP_CHECK(hipMemcpyAsync(I->h_pack, I->d_pack, acc,
