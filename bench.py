@@ -371,12 +371,18 @@ def main():
     vbuf = qrng.integers(0, 256, size=VAL_LEN, dtype=np.uint8).tobytes()
     tx_keys, tx_blobs, seen_tx = [], [], set()
     for _ in range(n_pre):
-        ks = []
+        ks, step_seen, attempts = [], set(), 0
         while len(ks) < ntx:
+            attempts += 1
+            if attempts > 20 * ntx or len(seen_tx) >= max(len(live) - ntx, 0):
+                seen_tx.clear()  # small pools: reuse keys (stale prev-revs
+                # then CAS-fail, which is legitimate mix noise; the default
+                # 1M-key config never reuses)
             k = live[int(qrng.integers(len(live)))]
-            if k in seen_tx or revs.get(k) is None:
+            if k in seen_tx or k in step_seen or revs.get(k) is None:
                 continue
             seen_tx.add(k)
+            step_seen.add(k)
             ks.append(k)
         parts = []
         for k in ks:
