@@ -463,10 +463,32 @@ __global__ void k_gather(const uint8_t* __restrict__ bkeys,
     overflow[q] = base_s > qcap ? 1 : 0;
     atomicAdd(bytes_out, (unsigned long long)(base_s <= qcap ? base_s : 0));
   }
-  __syncthreads();
-  if (base_s > qcap) return;
-  // phase B: 16-lane record groups — 4 records per wave issue their load
-  // chains together (memory-level parallelism across records)
+  // record copies happen in k_gather_copy (wider blocks, better latency
+  // hiding); this kernel only computes offsets/overflow
+}
+
+// copy winners' records into the arena; launched with 512 threads per query
+__global__ void k_gather_copy(const uint8_t* __restrict__ bkeys,
+                              const uint64_t* __restrict__ bmeta,
+                              const uint64_t* __restrict__ brev,
+                              const uint64_t* __restrict__ bvo,
+                              const uint8_t* __restrict__ dkeys,
+                              const uint64_t* __restrict__ dmeta,
+                              const uint64_t* __restrict__ drev,
+                              const uint64_t* __restrict__ dvo,
+                              const uint8_t* __restrict__ heap,
+                              const uint64_t* __restrict__ rows_out,
+                              int64_t max_cap,
+                              const int64_t* __restrict__ found_out, int nq,
+                              uint8_t* __restrict__ gbuf, int64_t qcap,
+                              const int64_t* __restrict__ offs,
+                              const int32_t* __restrict__ overflow) {
+  int q = blockIdx.x;
+  if (q >= nq || overflow[q]) return;
+  int64_t nwin = found_out[q];
+  const uint64_t* rows = rows_out + (int64_t)q * max_cap;
+  const int64_t* qoffs = offs + (int64_t)q * max_cap;
+  // 16-lane record groups — independent load chains across records
   int lane = threadIdx.x & 63, w = threadIdx.x >> 6;
   int grp = lane >> 4, gl = lane & 15;
   uint8_t* qb = gbuf + (int64_t)q * qcap;
@@ -485,9 +507,7 @@ __global__ void k_gather(const uint8_t* __restrict__ bkeys,
     }
     const uint8_t* ks = (isd ? dkeys : bkeys) + row * KEYW;
     uint8_t* kd = dst + 16;
-    // key rows are 96B and 16B-aligned per row? row*96 is 16B-aligned only for
-    // even rows; copy as u32 words (4-aligned always), padded width
-    uint32_t kw = (klen + 15) & ~15u;
+    uint32_t kw = (klen + 15) & ~15u;  // key rows are 96B: in-bounds, 4-aligned
     for (uint32_t b = gl; b < kw / 4; b += 16)
       ((uint32_t*)kd)[b] = ((const uint32_t*)ks)[b];
     const uint8_t* vs = heap + (isd ? dvo : bvo)[row];
@@ -681,6 +701,99 @@ __global__ void k_merge_scatter_delta(
   uint64_t* kd = (uint64_t*)(keysB + out * KEYW);
 #pragma unroll
   for (int kk = 0; kk < KEYW / 8; ++kk) kd[kk] = ks[kk];
+}
+
+// single-workgroup merge for small runs (n+m <= 8192): rank, drop-scan,
+// scatter and same_next in ONE launch — the multi-kernel path costs ~10
+// launches plus two host syncs, which dominates small per-step delta merges.
+__global__ void k_merge_small(const uint8_t* __restrict__ skeys,
+                              const uint64_t* __restrict__ smeta,
+                              const uint64_t* __restrict__ srev,
+                              const uint64_t* __restrict__ svo, int64_t n,
+                              const uint8_t* __restrict__ nk,
+                              const uint64_t* __restrict__ nm,
+                              const uint64_t* __restrict__ nr,
+                              const uint64_t* __restrict__ nv, int64_t m,
+                              uint8_t* __restrict__ okeys,
+                              uint64_t* __restrict__ ometa,
+                              uint64_t* __restrict__ orev,
+                              uint64_t* __restrict__ ovo,
+                              int64_t* __restrict__ out_n) {
+  __shared__ uint16_t dropx[8193];  // exclusive drop counts (n <= 8192)
+  __shared__ uint16_t partial[257];
+  int t = threadIdx.x;
+  const int T = blockDim.x;
+  // per-thread chunk of src rows: drop flags + partial sums
+  int chunk = (int)((n + T - 1) / T);
+  int i0 = t * chunk;
+  int i1 = min((int64_t)i0 + chunk, n);
+  uint16_t cnt = 0;
+  for (int i = i0; i < i1; ++i) {
+    int64_t lb = d_lb_range(nk, nr, 0, m, skeys + (int64_t)i * KEYW, srev[i]);
+    bool drop = lb < m && nr[lb] == srev[i] &&
+                keycmp96(nk + lb * KEYW, skeys + (int64_t)i * KEYW) == 0;
+    dropx[i] = drop ? 1 : 0;
+    cnt += drop;
+  }
+  partial[t] = cnt;
+  __syncthreads();
+  if (t == 0) {
+    uint16_t acc = 0;
+    for (int k = 0; k < T; ++k) { uint16_t v = partial[k]; partial[k] = acc; acc += v; }
+    partial[T] = acc;
+  }
+  __syncthreads();
+  // exclusive scan within chunk
+  {
+    uint16_t acc = partial[t];
+    for (int i = i0; i < i1; ++i) { uint16_t v = dropx[i]; dropx[i] = acc; acc += v; }
+  }
+  __syncthreads();
+  int64_t dropped = partial[T];
+  int64_t new_n = n - dropped + m;
+  if (t == 0) *out_n = new_n;
+  // scatter src rows
+  for (int64_t i = t; i < n; i += T) {
+    int64_t lb = d_lb_range(nk, nr, 0, m, skeys + i * KEYW, srev[i]);
+    bool drop = lb < m && nr[lb] == srev[i] &&
+                keycmp96(nk + lb * KEYW, skeys + i * KEYW) == 0;
+    if (drop) continue;
+    int64_t j = i - (int64_t)dropx[i] + lb;
+    orev[j] = srev[i];
+    ometa[j] = smeta[i];
+    ovo[j] = svo[i];
+    const uint64_t* ks = (const uint64_t*)(skeys + i * KEYW);
+    uint64_t* kd = (uint64_t*)(okeys + j * KEYW);
+#pragma unroll
+    for (int k = 0; k < KEYW / 8; ++k) kd[k] = ks[k];
+  }
+  // scatter new rows
+  for (int64_t j = t; j < m; j += T) {
+    int64_t lb = d_lb_range(skeys, srev, 0, n, nk + j * KEYW, nr[j]);
+    int64_t dpx = lb < n ? dropx[lb] : (int64_t)dropped;
+    int64_t o = lb - dpx + j;
+    orev[o] = nr[j];
+    ometa[o] = nm[j];
+    ovo[o] = nv[j];
+    const uint64_t* ks = (const uint64_t*)(nk + j * KEYW);
+    uint64_t* kd = (uint64_t*)(okeys + o * KEYW);
+#pragma unroll
+    for (int k = 0; k < KEYW / 8; ++k) kd[k] = ks[k];
+  }
+  __syncthreads();
+  // same_next over the merged rows
+  for (int64_t i = t; i < new_n; i += T) {
+    bool same = false;
+    if (i + 1 < new_n) {
+      const uint64_t* a = (const uint64_t*)(okeys + i * KEYW);
+      const uint64_t* b = (const uint64_t*)(okeys + (i + 1) * KEYW);
+      same = true;
+#pragma unroll
+      for (int k = 0; k < KEYW / 8; ++k)
+        if (a[k] != b[k]) { same = false; break; }
+    }
+    ometa[i] = same ? (ometa[i] | M_SAME_NEXT) : (ometa[i] & ~M_SAME_NEXT);
+  }
 }
 
 // ---- watch fan-out filter (watch.go:119-159 per-watcher predicate) ------
@@ -894,6 +1007,17 @@ struct Slab::Impl {
   bool mergeRuns(const Col& src, int64_t n, const uint8_t* nk,
                  const uint64_t* nm, const uint64_t* nr, const uint64_t* nv,
                  int64_t m, Col& dst, int64_t* out_n, std::string* err) {
+    if (n + m <= 8192) {  // single-launch path for small delta merges
+      hipLaunchKernelGGL(k_merge_small, dim3(1), dim3(256), 0, stream,
+                         src.keys, src.meta, src.rev, src.vo, n, nk, nm, nr,
+                         nv, m, dst.keys, dst.meta, dst.rev, dst.vo,
+                         (int64_t*)s_d);
+      int64_t nn = 0;
+      HIP_CHECK(hipMemcpyAsync(&nn, s_d, 8, hipMemcpyDeviceToHost, stream));
+      HIP_CHECK(hipStreamSynchronize(stream));
+      *out_n = nn;
+      return true;
+    }
     uint64_t dropped = 0;
     if (n > 0) {
       int64_t nb = ceil_div(n, 256);
@@ -1159,6 +1283,11 @@ bool Slab::RangeBatchStart(const std::vector<DevRangeQ>& qs, std::string* err) {
                      I->DA.rev, I->DA.vo, I->heapA, I->d_rowsm, I->max_cap,
                      I->d_found, nq, I->d_gbuf, qcap, I->d_offs, I->d_gbytes,
                      I->d_ovf, I->d_bytes);
+  hipLaunchKernelGGL(k_gather_copy, dim3(nq), dim3(512), 0, I->stream,
+                     I->A.keys, I->A.meta, I->A.rev, I->A.vo, I->DA.keys,
+                     I->DA.meta, I->DA.rev, I->DA.vo, I->heapA, I->d_rowsm,
+                     I->max_cap, I->d_found, nq, I->d_gbuf, qcap, I->d_offs,
+                     I->d_ovf);
   HIP_CHECK(hipEventRecord(I->ev2, I->stream));
   return true;
 }
