@@ -563,6 +563,11 @@ __global__ void k_add_offsets(uint64_t* __restrict__ out,
 }
 
 // ---- compaction (compact.go + scanner.go:444-491, 566-591) --------------
+__global__ void k_merge_newn(const uint64_t* __restrict__ dropx, int64_t n,
+                             int64_t m, int64_t* __restrict__ out_n) {
+  if (threadIdx.x == 0 && blockIdx.x == 0) *out_n = n - (int64_t)dropx[n] + m;
+}
+
 __global__ void k_fill_u64(uint64_t* a, uint64_t v, int64_t n) {
   int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
   if (i < n) a[i] = v;
@@ -1006,25 +1011,26 @@ struct Slab::Impl {
   // rev-rows replace) -> dst; recomputes same_next. dst must hold n+m rows.
   bool mergeRuns(const Col& src, int64_t n, const uint8_t* nk,
                  const uint64_t* nm, const uint64_t* nr, const uint64_t* nv,
-                 int64_t m, Col& dst, int64_t* out_n, std::string* err) {
-    if (n + m <= 8192) {  // single-launch path for small delta merges
+                 int64_t m, Col& dst, int64_t* out_n, std::string* err,
+                 bool device_newn_ok = false) {
+    if (n + m <= 2048) {  // single-launch path only where one CU wins
       hipLaunchKernelGGL(k_merge_small, dim3(1), dim3(256), 0, stream,
                          src.keys, src.meta, src.rev, src.vo, n, nk, nm, nr,
                          nv, m, dst.keys, dst.meta, dst.rev, dst.vo,
                          (int64_t*)s_d);
+      if (device_newn_ok) return true;  // caller knows new_n; stay async
       int64_t nn = 0;
       HIP_CHECK(hipMemcpyAsync(&nn, s_d, 8, hipMemcpyDeviceToHost, stream));
       HIP_CHECK(hipStreamSynchronize(stream));
       *out_n = nn;
       return true;
     }
-    uint64_t dropped = 0;
     if (n > 0) {
       int64_t nb = ceil_div(n, 256);
       hipLaunchKernelGGL(k_merge_rank, dim3((uint32_t)nb), dim3(256), 0, stream,
                          src.keys, src.rev, n, nk, nr, m, s_a, s_b);
       HIP_CHECK(hipMemsetAsync(s_b + n, 0, 8, stream));
-      if (!scan(s_b, s_c, n + 1, &dropped, err)) return false;
+      if (!scan(s_b, s_c, n + 1, nullptr, err)) return false;  // no host sync
       hipLaunchKernelGGL(k_merge_scatter_base, dim3((uint32_t)nb), dim3(256), 0,
                          stream, src.keys, src.meta, src.rev, src.vo, s_a, s_b,
                          s_c, dst.keys, dst.meta, dst.rev, dst.vo, n);
@@ -1037,7 +1043,22 @@ struct Slab::Impl {
                          0, stream, src.keys, src.rev, n, nk, nm, nr, nv, m,
                          s_c, dst.keys, dst.meta, dst.rev, dst.vo);
     }
-    int64_t new_n = n - (int64_t)dropped + m;
+    // new_n on device (s_d[0]) to avoid an extra host round trip; same_next
+    // must cover the full possible extent (n+m), which is safe: rows beyond
+    // new_n are untouched dst scratch read-only here... they ARE read, so
+    // clamp instead: compute new_n first, then same_next sized by it.
+    hipLaunchKernelGGL(k_merge_newn, dim3(1), dim3(64), 0, stream, s_c, n, m,
+                       (int64_t*)s_d);
+    int64_t new_n;
+    if (device_newn_ok) {
+      // caller tracks new_n exactly; same_next sized by the upper bound is
+      // safe only if it matches — use the caller-known count via *out_n
+      new_n = *out_n;
+    } else {
+      new_n = 0;
+      HIP_CHECK(hipMemcpyAsync(&new_n, s_d, 8, hipMemcpyDeviceToHost, stream));
+      HIP_CHECK(hipStreamSynchronize(stream));
+    }
     if (new_n > 0) {
       int64_t sb = ceil_div(new_n, 256);
       hipLaunchKernelGGL(k_same_next, dim3((uint32_t)sb), dim3(256), 0, stream,
@@ -1158,30 +1179,48 @@ bool Slab::HeapAppend(const void* src, int64_t len, int64_t* off, std::string* e
 
 bool Slab::AppendRows(const uint8_t* keys, const uint64_t* meta,
                       const uint64_t* rev, const uint64_t* vo, int64_t m,
-                      std::string* err) {
+                      std::string* err, int64_t known_new_dn) {
   Impl* I = p;
   if (m == 0) return true;
   if (I->dn + m > I->delta_cap) {
-    // fold first to make room
+    // fold first to make room (prediction no longer valid: go sync)
     if (!Fold(err)) return false;
+    known_new_dn = -1;
     if (m > I->delta_cap) { if (err) *err = "append larger than delta cap"; return false; }
   }
-  HIP_CHECK(hipEventRecord(I->ev0, I->stream));
+  static const bool validate = getenv("KB_VALIDATE_DN") &&
+                               atoi(getenv("KB_VALIDATE_DN")) != 0;
+  bool async = known_new_dn >= 0 && !validate;
+  if (!async) HIP_CHECK(hipEventRecord(I->ev0, I->stream));
   if (!I->ensure_delta(m, err)) return false;
   HIP_CHECK(hipMemcpyAsync(I->d_dkeys, keys, m * KEYW, hipMemcpyHostToDevice, I->stream));
   HIP_CHECK(hipMemcpyAsync(I->d_dmeta, meta, m * 8, hipMemcpyHostToDevice, I->stream));
   HIP_CHECK(hipMemcpyAsync(I->d_drev, rev, m * 8, hipMemcpyHostToDevice, I->stream));
   HIP_CHECK(hipMemcpyAsync(I->d_dvo, vo, m * 8, hipMemcpyHostToDevice, I->stream));
-  int64_t new_dn = 0;
+  int64_t new_dn = async ? known_new_dn : 0;  // async: mergeRuns reads it
   if (!I->mergeRuns(I->DA, I->dn, I->d_dkeys, I->d_dmeta, I->d_drev, I->d_dvo,
-                    m, I->DB, &new_dn, err))
+                    m, I->DB, &new_dn, err, /*device_newn_ok=*/async))
     return false;
+  if (async) {
+    // kernels queue on the stream; subsequent reads queue behind them.
+    // NOTE: the host `keys/meta/rev/vo` buffers were consumed by the H2D
+    // copies above, which HIP stages synchronously for pageable memory.
+    perf.merges++;
+    std::swap(I->DA, I->DB);
+    I->dn = known_new_dn;
+    return true;
+  }
   HIP_CHECK(hipEventRecord(I->ev1, I->stream));
   HIP_CHECK(hipStreamSynchronize(I->stream));
   float ms = 0;
   (void)hipEventElapsedTime(&ms, I->ev0, I->ev1);
   perf.merge_ms += ms;
   perf.merges++;
+  if (known_new_dn >= 0 && new_dn != known_new_dn) {
+    if (err) *err = "delta row-count prediction mismatch: " +
+                    std::to_string(known_new_dn) + " vs " + std::to_string(new_dn);
+    return false;
+  }
   std::swap(I->DA, I->DB);
   I->dn = new_dn;
   return true;

@@ -120,9 +120,14 @@ class Slab {
 
   // append value bytes to the device heap; *off = absolute offset
   bool HeapAppend(const void* p, int64_t len, int64_t* off, std::string* err);
-  // merge sorted new rows (vo already absolute) into the DELTA run
+  // merge sorted new rows (vo already absolute) into the DELTA run.
+  // known_new_dn >= 0: the caller knows the exact post-merge row count
+  // (drops = rev-row replacements it tracked), so the merge runs fully
+  // async — no host sync; the next kernel on the stream queues behind it.
+  // KB_VALIDATE_DN=1 re-checks the prediction (enabled by the test env).
   bool AppendRows(const uint8_t* keys, const uint64_t* meta, const uint64_t* rev,
-                  const uint64_t* vo, int64_t m, std::string* err);
+                  const uint64_t* vo, int64_t m, std::string* err,
+                  int64_t known_new_dn = -1);
   // fold the delta run into the base run; delta becomes empty
   bool Fold(std::string* err);
 

@@ -188,14 +188,29 @@ bool Store::syncReads(std::string* err) {
       rev[i] = newrows_[i].rev;
       vo[i] = newrows_[i].vo;
     }
+    // delta-merge drops happen exactly where an uploaded rev-row's key
+    // already has a rev-row in the delta run — tracked host-side, so the
+    // merge can run fully async with an exact predicted row count
+    if (slab_->delta_rows() + (int64_t)m > (int64_t)env_i64("KB_DELTA_CAP", 1 << 19)) {
+      if (!slab_->Fold(err)) return false;
+      delta_revkeys_.clear();
+    }
+    int64_t drops = 0;
+    for (size_t i = 0; i < m; ++i) {
+      if (rev[i] != 0) continue;
+      auto ins = delta_revkeys_.insert(newrows_[i].key);
+      if (!ins.second) drops++;
+    }
+    int64_t predicted = slab_->delta_rows() + (int64_t)m - drops;
     if (!slab_->AppendRows(keys.data(), meta.data(), rev.data(), vo.data(),
-                           (int64_t)m, err))
+                           (int64_t)m, err, predicted))
       return false;
     newrows_.clear();
     nr_revrow_.clear();
   }
   if (slab_->delta_rows() >= cfg_.flush_rows) {
     if (!slab_->Fold(err)) return false;
+    delta_revkeys_.clear();
   }
   sync_s_ +=
       std::chrono::duration<double>(std::chrono::steady_clock::now() - t0).count();

@@ -195,6 +195,7 @@ class Store {
   double sync_s_ = 0;
   int64_t sync_n_ = 0;
   std::string fatal_;  // first unrecoverable device error (e.g. slab full)
+  std::unordered_set<Bytes> delta_revkeys_;  // keys with rev-rows in the delta run
   struct StreamState { Bytes frontier, end; uint64_t read_rev; bool done; };
   std::unordered_map<int64_t, StreamState> streams_;
   int64_t next_sid_ = 1;

@@ -127,3 +127,4 @@ def small_env():
     os.environ.setdefault("KB_HEAP_BYTES", str(256 << 20))
     os.environ.setdefault("KB_ARENA_BYTES", str(64 << 20))
     os.environ.setdefault("KB_FLUSH_ROWS", "512")  # exercise merges often
+    os.environ.setdefault("KB_VALIDATE_DN", "1")   # verify async-merge row-count predictions
