@@ -1158,6 +1158,7 @@ Slab* Slab::Create(int64_t max_rows, int64_t heap_cap, int device,
 Slab::~Slab() { delete p; }
 int64_t Slab::rows() const { return p->n; }
 int64_t Slab::delta_rows() const { return p->dn; }
+int64_t Slab::delta_capacity() const { return p->delta_cap; }
 int64_t Slab::heap_used() const { return p->heap_used_; }
 
 bool Slab::HeapAppend(const void* src, int64_t len, int64_t* off, std::string* err) {

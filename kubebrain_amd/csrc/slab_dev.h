@@ -111,6 +111,7 @@ class Slab {
 
   int64_t rows() const;
   int64_t delta_rows() const;
+  int64_t delta_capacity() const;
   int64_t heap_used() const;
 
   // merge sorted delta rows straight into the BASE run (GPU merge by ranks;

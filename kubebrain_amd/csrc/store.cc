@@ -191,7 +191,7 @@ bool Store::syncReads(std::string* err) {
     // delta-merge drops happen exactly where an uploaded rev-row's key
     // already has a rev-row in the delta run — tracked host-side, so the
     // merge can run fully async with an exact predicted row count
-    if (slab_->delta_rows() + (int64_t)m > (int64_t)env_i64("KB_DELTA_CAP", 1 << 19)) {
+    if (slab_->delta_rows() + (int64_t)m > slab_->delta_capacity()) {
       if (!slab_->Fold(err)) return false;
       delta_revkeys_.clear();
     }
@@ -219,7 +219,9 @@ bool Store::syncReads(std::string* err) {
 
 bool Store::foldLocked(std::string* err) {
   if (!syncReads(err)) return false;
-  return slab_->Fold(err);
+  if (!slab_->Fold(err)) return false;
+  delta_revkeys_.clear();  // the delta-run mirror must track every fold
+  return true;
 }
 
 bool Store::Flush(std::string* err) {
