@@ -191,6 +191,71 @@ def run_txns(store, live_keys, revs, rng, n):
     txn_batch(store, ops, revs)
 
 
+def cross_shard_leg_wrapper(store, dist, pg, rank, world, all_namespaces):
+    out = cross_shard_leg(store, dist, pg, rank, world, all_namespaces)
+    return out
+
+
+def cross_shard_leg(store, dist, pg, rank, world, all_namespaces, n_queries=64,
+                    limit=LIMIT):
+    """configs[3]: cross-shard Range — every shard scans its sub-slab, winner
+    payloads are exchanged with one collective (RCCL over xGMI when CUDA is
+    up, the gloo group otherwise), and rank 0 k-way-merges the sorted runs
+    with the global limit+1 cut (semantics pinned by
+    tests/test_gloo_shard.py)."""
+    import heapq
+
+    import torch
+
+    use_cuda = torch.cuda.is_available()
+    dev = torch.device("cuda", int(os.environ.get("LOCAL_RANK", 0))) \
+        if use_cuda else torch.device("cpu")
+    t0 = time.time()
+    merged_counts = []
+    for qi in range(n_queries):
+        # a cross-namespace range spanning every shard
+        lo = b"/registry/pods/ns-%04d" % (qi % max(len(all_namespaces) // 2, 1))
+        r = store.list(lo, b"/registry/pods0", 0, limit + 1)
+        parts = [struct.pack("<QII", kv.revision, len(kv.key), len(kv.value))
+                 + kv.key + kv.value for kv in r.kvs]
+        blob = b"".join(parts)
+        # exchange: lengths then padded payload (allgatherv emulation)
+        ln = torch.tensor([len(blob)], dtype=torch.int64, device=dev)
+        lens = [torch.zeros(1, dtype=torch.int64, device=dev) for _ in range(world)]
+        dist.all_gather(lens, ln, group=pg)
+        mx = int(max(x.item() for x in lens))
+        buf = torch.zeros(mx, dtype=torch.uint8, device=dev)
+        if blob:
+            buf[:len(blob)] = torch.frombuffer(bytearray(blob), dtype=torch.uint8).to(dev)
+        outs = [torch.zeros(mx, dtype=torch.uint8, device=dev) for _ in range(world)]
+        dist.all_gather(outs, buf, group=pg)
+        if rank == 0:
+            runs = []
+            for rr in range(world):
+                data = bytes(outs[rr][: int(lens[rr].item())].cpu().numpy().tobytes())
+                run, off = [], 0
+                while off < len(data):
+                    rev, klen, vlen = struct.unpack_from("<QII", data, off)
+                    off += 16
+                    k = data[off:off + klen]; off += klen
+                    v = data[off:off + vlen]; off += vlen
+                    run.append((k, rev, v))
+                runs.append(run)
+            merged = list(heapq.merge(*runs, key=lambda kv: kv[0]))[: limit]
+            assert all(merged[i][0] < merged[i + 1][0]
+                       for i in range(len(merged) - 1)), "merge order"
+            merged_counts.append(len(merged))
+    dt = time.time() - t0
+    return {
+        "queries": n_queries * 1,
+        "ops_per_sec": round(n_queries / dt, 1),
+        "limit": limit,
+        "transport": "nccl(RCCL/xGMI)" if use_cuda else "gloo",
+        "merged_counts_min_max": [min(merged_counts), max(merged_counts)]
+        if rank == 0 and merged_counts else None,
+    }
+
+
 def cpu_baseline_leg(namespaces, keys, sample_qs):
     """Oracle (CPU restatement, kind 'port') on the host cores — the only
     bench.py use of oracle/ (DESIGN.md §1)."""
@@ -447,6 +512,20 @@ def main():
     d2h_elapsed = time.time() - t0
     d2h_steps = max(2, args.steps // 8)
 
+    # cross-shard Range leg (configs[3]) at N>1: RCCL/xGMI exchange + merge.
+    # Exception-guarded: a failure degrades to a JSON note, never the run.
+    cross_shard = None
+    if world > 1:
+        try:
+            pg = None
+            import torch as _t
+            if _t.cuda.is_available():
+                pg = dist.new_group(backend="nccl")
+            cross_shard = cross_shard_leg_wrapper(store, dist, pg, rank, world,
+                                                  namespaces)
+        except Exception as e:  # noqa: BLE001
+            cross_shard = {"error": repr(e)[:300]}
+
     # compaction sweep leg (configs[2]: drop revisions < compactRev; the
     # sweep reads the whole slab and stream-compacts rows + value heap)
     compact_stats = None
@@ -549,6 +628,7 @@ def main():
                 "limit": LIMIT,
                 "ops_per_step": OPS_PER_STEP,
             },
+            "cross_shard_range": cross_shard,
             "compact_sweep": compact_stats,
             "watch_events_per_sec": round(wrate, 1),
             "watch_delivered_rank0": delivered,

@@ -37,21 +37,12 @@ def load_library() -> ctypes.CDLL:
 
 def open_store(store_prefix: bytes = b"/registry", **kw):
     """Open a GPU-backed store (raises RuntimeError without a GPU)."""
-    import sys
-
-    sys.path.insert(0, os.path.join(_HERE, "..", "tests"))
-    from kbclient import Store
-
-    class _KbStore(Store):
-        def __init__(self, **kw2):
-            super().__init__(LIB_PATH, "kb_", **kw2)
-            if not self.h:
-                raise RuntimeError("kb_new failed")
+    from .client import Store
 
     if not os.path.exists(LIB_PATH):
         build()
     try:
-        return _KbStore(store_prefix=store_prefix, **kw)
+        return Store(LIB_PATH, "kb_", store_prefix=store_prefix, **kw)
     except RuntimeError as e:
         lib = ctypes.CDLL(LIB_PATH)
         buf = ctypes.create_string_buffer(512)

@@ -38,6 +38,13 @@ class MockStore:
     def watch_cancel(self, wid):
         pass
 
+    def list(self, start, end, rev=0, limit=0):
+        from kubebrain_amd.client import Kv, RangeResp
+        n = min(limit if limit > 0 else 5, 5)
+        kvs = [Kv(start + b"k%04d-r%d" % (i, os.environ.get("RANK", "0") != "0"),
+                  b"v" * 8, 100 + i) for i in range(n)]
+        return RangeResp(0, self.rev, kvs, False)
+
     def compact(self, rev=0):
         return 0, self.rev
 
@@ -119,6 +126,8 @@ def _worker(rank, world, port, outdir):
         assert data["n_gpus"] == world
         assert data["scaling"] == "weak"
         assert data["config"]["n_keys"] == 16 * world * 10  # weak scaling
+        cs = data.get("cross_shard_range")
+        assert cs and "error" not in cs, cs
         with open(os.path.join(outdir, "bench_mock.json"), "w") as f:
             f.write(out)
 
