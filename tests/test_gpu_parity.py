@@ -273,3 +273,29 @@ def test_stream_and_partitions(dual):
     so, _ = dual.o.stream(ns + b"/", ns + b"0", mid - 1)
     sp, _ = dual.p.stream(ns + b"/", ns + b"0", mid - 1)
     assert so == sp and so != 0
+
+
+def test_watch_cancel_slot_reuse(dual):
+    # cancel/re-register cycles reuse device watcher slots (WatcherClear)
+    pfx = b"/registry/pods/ns-05"
+    w1 = dual.watch(pfx + b"/a", 0)
+    w2 = dual.watch(pfx + b"/b", 0)
+    w3 = dual.watch(pfx + b"/", 0)
+    for i in range(6):
+        dual.create(pfx + b"/a-%d" % i if i % 2 else pfx + b"/b-%d" % i, b"x")
+    dual.poll_all()
+    # cancel the middle watcher on both sides, then register another (the
+    # product reuses its device slot)
+    wo, wp = dual.watches[w2]
+    dual.o.watch_cancel(wo)
+    dual.p.watch_cancel(wp)
+    w4 = dual.watch(pfx + b"/a", 0)
+    for i in range(6, 12):
+        dual.create(pfx + b"/a-%d" % i, b"y")
+    dual.poll(w1)
+    dual.poll(w3)
+    dual.poll(w4)
+    # the canceled watcher no longer exists on the product side
+    import kbclient
+    rc, _ = dual.p.watch_poll(wp)
+    assert rc == kbclient.WATCH_DROPPED
