@@ -81,7 +81,8 @@ __device__ int64_t d_upper_bound(const uint8_t* keys, const uint64_t* rev,
 // ---- same_next: adjacent-key equality bits, LDS-tiled coalesced loads ----
 __global__ void k_same_next(const uint8_t* __restrict__ keys,
                             uint64_t* __restrict__ meta, int64_t n) {
-  __shared__ uint8_t tile[(256 + 1) * KEYW];
+  __shared__ uint64_t tile64[(256 + 1) * KEYW / 8];  // u64 storage: aligned
+  uint8_t* tile = (uint8_t*)tile64;
   int64_t base = (int64_t)blockIdx.x * 256;
   if (base >= n) return;
   int64_t count = min((int64_t)256, n - base);
