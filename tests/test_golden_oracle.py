@@ -90,6 +90,18 @@ def test_ring_table():
         assert (bool(high.value), bool(low.value)) == (case["high"], case["low"]), case
         assert (oldest.value, newest.value) == (case["oldest"], case["newest"]), case
         assert list(evs[:n.value]) == case["events"], case
+    # ring_test.go:99-107 (NewRingFindEvents): under-filled ring, cap 10, add 1..7
+    revs7 = (C.c_uint64 * 7)(*range(1, 8))
+    empty = C.c_int(); high = C.c_int(); low = C.c_int()
+    oldest = C.c_uint64(); newest = C.c_uint64()
+    evs = (C.c_uint64 * 16)()
+    n = C.c_size_t()
+    rc = lib.okb_ring_test(10, revs7, C.c_size_t(7), C.c_uint64(7),
+                           C.byref(empty), C.byref(high), C.byref(low),
+                           C.byref(oldest), C.byref(newest), evs, C.c_size_t(16),
+                           C.byref(n))
+    assert rc == 0 and newest.value == 7
+    assert list(evs[:n.value]) == [7]
 
 
 # ---- backend table tests ----
