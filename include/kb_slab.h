@@ -120,6 +120,11 @@ int kb_partitions(kb_store*, const uint8_t* start, size_t slen,
 unsigned long long kb_current_rev(kb_store*);
 void kb_set_current_rev(kb_store*, unsigned long long rev); /* leader TSO init */
 
+/* backend.Config.SkippedPrefixes (compact.go:108-127): comma-separated */
+int kb_set_skipped_prefixes(kb_store*, const char* csv);
+/* the encoded compact borders (golden-pinned by compact_test.go:36-79) */
+int kb_compact_borders(kb_store*, uint8_t* out, size_t cap, size_t* out_len);
+
 /* ---- test/ops hooks ---- */
 void kb_clock_advance(kb_store*, long long seconds); /* TTL clock (scanner.go:147-177) */
 int kb_flush(kb_store*); /* memtable -> HBM slab merge (normally automatic) */

@@ -335,6 +335,30 @@ int kb_bulk_create(kb_store* h, const uint8_t* keys, const uint32_t* klens,
   return 0;
 }
 
+int kb_set_skipped_prefixes(kb_store* h, const char* csv) {
+  std::vector<Bytes> sp;
+  Bytes cur;
+  for (const char* p = csv;; ++p) {
+    if (*p == ',' || *p == 0) {
+      if (!cur.empty()) sp.push_back(cur);
+      cur.clear();
+      if (*p == 0) break;
+    } else cur += *p;
+  }
+  ((Store*)h)->SetSkippedPrefixes(sp);
+  return 0;
+}
+
+int kb_compact_borders(kb_store* h, uint8_t* out, size_t cap, size_t* out_len) {
+  auto bs = ((Store*)h)->CompactBorders();
+  Writer w{out, cap};
+  w.u32((uint32_t)bs.size());
+  for (auto& b : bs) w.str(b);
+  *out_len = w.off;
+  if (w.overflow) return KB_ENOBUF;
+  return 0;
+}
+
 int kb_perf_json(kb_store* h, char* out, size_t cap) {
   std::string j = ((Store*)h)->PerfJson();
   if (j.size() + 1 > cap) return KB_ENOBUF;

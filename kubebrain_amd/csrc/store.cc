@@ -774,6 +774,21 @@ uint64_t Store::getTimeoutRevision() {
   return prev.revision;
 }
 
+std::vector<Bytes> Store::CompactBorders() const {
+  // compact.go:108-127
+  std::vector<Bytes> prefixes;
+  prefixes.push_back(cfg_.prefix);
+  for (auto& p : cfg_.skipped_prefixes) prefixes.push_back(p);
+  std::vector<Bytes> borders;
+  for (auto key : prefixes) {
+    if (key.empty() || key.back() != '/') key += '/';
+    borders.push_back(EncodeObjectKey(key, 0));
+    borders.push_back(EncodeObjectKey(PrefixEnd(key), 0));
+  }
+  std::sort(borders.begin(), borders.end());
+  return borders;
+}
+
 uint64_t Store::Compact(uint64_t revision, Status* st) {
   // compact.go:31-127 (retry-queue MinRevision stub == 0)
   std::lock_guard<std::recursive_mutex> lk(mu_);
@@ -788,7 +803,8 @@ uint64_t Store::Compact(uint64_t revision, Status* st) {
   std::string err;
   if (!foldLocked(&err)) { *st = INTERNAL; return revision; }
   pumpEvents();
-  // borders (compact.go:108-127)
+  // borders (compact.go:108-127) — user-key-space bounds for the device
+  // kernels (CompactBorders() above returns the encoded form for parity)
   std::vector<Bytes> prefixes;
   prefixes.push_back(cfg_.prefix);
   for (auto& p : cfg_.skipped_prefixes) prefixes.push_back(p);

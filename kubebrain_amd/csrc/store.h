@@ -93,6 +93,8 @@ class Store {
   void WatchCancel(int64_t wid);
 
   void ClockAdvance(int64_t secs);
+  void SetSkippedPrefixes(const std::vector<Bytes>& sp) { cfg_.skipped_prefixes = sp; }
+  std::vector<Bytes> CompactBorders() const;
   bool Flush(std::string* err);
   // sorted (internal key, value) pairs, byte-diffable vs the oracle dump
   bool DumpStore(std::vector<std::pair<Bytes, Bytes>>* out, std::string* err);

@@ -156,7 +156,9 @@ class Backend {
   std::vector<Event> WatchPoll(int64_t wid, Status* st);
   void WatchCancel(int64_t wid);
 
-  // -- test hooks --
+  // -- config/test hooks --
+  void SetSkippedPrefixes(const std::vector<Bytes>& sp) { cfg_.skipped_prefixes = sp; }
+  std::vector<Bytes> CompactBorders() const { return getCompactBorders(); }
   // Advance the TTL clock (scanner compactHistories, scanner.go:147-177).
   void ClockAdvance(int64_t seconds) { now_ += seconds; }
   // Dump the full internal store (sorted internal key -> value) for slab diff.

@@ -239,6 +239,30 @@ int okb_partitions(void* h, const uint8_t* start, size_t slen,
   return 0;
 }
 
+int okb_set_skipped_prefixes(void* h, const char* csv) {
+  std::vector<Bytes> sp;
+  Bytes cur;
+  for (const char* p = csv; ; ++p) {
+    if (*p == ',' || *p == 0) {
+      if (!cur.empty()) sp.push_back(cur);
+      cur.clear();
+      if (*p == 0) break;
+    } else cur += *p;
+  }
+  ((Backend*)h)->SetSkippedPrefixes(sp);
+  return 0;
+}
+
+int okb_compact_borders(void* h, uint8_t* out, size_t cap, size_t* out_len) {
+  auto bs = ((Backend*)h)->CompactBorders();
+  Writer w{out, cap};
+  w.u32((uint32_t)bs.size());
+  for (auto& b : bs) w.str(b);
+  *out_len = w.off;
+  if (w.overflow) return kEnoBuf;
+  return 0;
+}
+
 // ---- coder / ring / util helpers for the golden-vector tests ----
 int okb_encode_key(const uint8_t* k, size_t klen, uint64_t rev, uint8_t* out,
                    size_t cap, size_t* olen) {

@@ -358,3 +358,43 @@ def test_compacted_range_errors(oracle):
     assert r.status == COMPACTED
     r = oracle.list(KEY, KEY + b"0", 0, 10)
     assert r.status == OK
+
+
+def test_compact_borders_with_skipped_prefixes():
+    # compact_test.go:36-79 (TestConstructCompactBordersWithSkippedPrefixOption)
+    import struct as _s
+
+    def borders(prefix, skipped):
+        s = open_oracle(store_prefix=prefix)
+        try:
+            if skipped:
+                s.lib.okb_set_skipped_prefixes(C.c_void_p(s.h), b",".join(skipped))
+            out_len = C.c_size_t()
+            rc = s.lib.okb_compact_borders(C.c_void_p(s.h), s.buf,
+                                           C.c_size_t(s.BUF), C.byref(out_len))
+            assert rc == 0
+            buf = s.buf[:out_len.value]
+            (n,) = _s.unpack_from("<I", buf, 0)
+            off, res = 4, []
+            for _ in range(n):
+                (ln,) = _s.unpack_from("<I", buf, off); off += 4
+                res.append(buf[off:off + ln]); off += ln
+            return res
+        finally:
+            s.close()
+
+    def enc_rev_key(k):
+        return b"\x57\xfb\x80\x8b" + k + b"$" + (0).to_bytes(8, "big")
+
+    got = borders(b"/registry/test",
+                  [b"/registry/test/pods", b"/registry/test/events"])
+    assert got == [
+        enc_rev_key(b"/registry/test/"),
+        enc_rev_key(b"/registry/test/events/"),
+        enc_rev_key(b"/registry/test/events0"),
+        enc_rev_key(b"/registry/test/pods/"),
+        enc_rev_key(b"/registry/test/pods0"),
+        enc_rev_key(b"/registry/test0"),
+    ]
+    got = borders(b"/registry/test", [])
+    assert got == [enc_rev_key(b"/registry/test/"), enc_rev_key(b"/registry/test0")]
