@@ -95,7 +95,9 @@ long long kb_watch(kb_store*, const uint8_t* prefix, size_t plen, uint64_t rev,
                    int* status);
 /* out = packed events {u32 n; n × {i32 type; u64 rev; u64 kv_rev; u32 klen;
  * key; u32 vlen; val}}. Returns KB_EWATCH_DROPPED once a slow consumer was
- * dropped (watcherhub.go:84-94: per-watcher buffer 10000). */
+ * dropped (watcherhub.go:84-94: per-watcher buffer 10000). On KB_ENOBUF the
+ * queue is left INTACT (*out_len = required size): retry with a larger
+ * buffer and no event is lost. */
 int kb_watch_poll(kb_store*, long long wid, uint8_t* out, size_t cap, size_t* out_len);
 void kb_watch_cancel(kb_store*, long long wid);
 

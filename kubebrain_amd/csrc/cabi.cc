@@ -189,12 +189,18 @@ long long kb_watch(kb_store* h, const uint8_t* prefix, size_t plen, uint64_t rev
 
 int kb_watch_poll(kb_store* h, long long wid, uint8_t* out, size_t cap,
                   size_t* out_len) {
+  // overflow-safe: the queue is only drained when the packed events fit, so
+  // KB_ENOBUF can be retried with a larger buffer without losing events
+  // (contiguous-revision delivery, backend.go:214-238). *out_len carries the
+  // required size on KB_ENOBUF.
   Status st;
-  auto evs = ((Store*)h)->WatchPoll(wid, &st);
+  size_t need = 0;
+  auto evs = ((Store*)h)->WatchPollLimited(wid, cap, &need, &st);
+  if (st == kbstore::NOBUF) { *out_len = need; return KB_ENOBUF; }
   Writer w{out, cap};
   writeEvents(w, evs);
   *out_len = w.off;
-  if (w.overflow) return KB_ENOBUF;
+  if (w.overflow) return KB_ENOBUF;  // unreachable: size was pre-checked
   return st;
 }
 

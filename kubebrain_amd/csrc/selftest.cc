@@ -180,7 +180,7 @@ int main() {
     borders[0].first.rev = 0;
     pad("/registry0", borders[0].second.key);
     borders[0].second.rev = 0;
-    STEP("compact", s->Compact(borders, 6, 0, &err));
+    STEP("compact", s->Compact(borders, 6, {0}, &err));
     std::vector<DumpRow> rows;
     STEP("dump2", s->Dump(&rows, &err));
     printf("  rows=%zu:", rows.size());

@@ -289,9 +289,9 @@ __global__ void k_range_scan2(
   __shared__ int64_t lo_s, hi_s, dlo_s, dhi_s, cnt_s;
   __shared__ int wave_cnt[4];
   const DevRangeQ& Q = qs[q];
-  if (threadIdx.x == 0) lo_s = d_lb_fenced(bkeys, brev, n, fkeys, frev, nf, Q.start, 0);
+  if (threadIdx.x == 0) lo_s = d_lb_fenced(bkeys, brev, n, fkeys, frev, nf, Q.start, Q.start_rev);
   if (threadIdx.x == 64) hi_s = d_lb_fenced(bkeys, brev, n, fkeys, frev, nf, Q.end, 0);
-  if (threadIdx.x == 128) dlo_s = d_lb_range(dkeys, drev, 0, dn, Q.start, 0);
+  if (threadIdx.x == 128) dlo_s = d_lb_range(dkeys, drev, 0, dn, Q.start, Q.start_rev);
   if (threadIdx.x == 192) dhi_s = d_lb_range(dkeys, drev, 0, dn, Q.end, 0);
   __syncthreads();
   const int64_t cap = Q.cap > 0 ? Q.cap : INT64_MAX;
@@ -1161,6 +1161,7 @@ int64_t Slab::rows() const { return p->n; }
 int64_t Slab::delta_rows() const { return p->dn; }
 int64_t Slab::delta_capacity() const { return p->delta_cap; }
 int64_t Slab::heap_used() const { return p->heap_used_; }
+int64_t Slab::max_winner_cap() const { return p->max_cap; }
 
 bool Slab::HeapAppend(const void* src, int64_t len, int64_t* off, std::string* err) {
   Impl* I = p;
@@ -1485,7 +1486,12 @@ bool Slab::GetBatch(const std::vector<DevGetQ>& qs, std::vector<GetResult>* outs
 }
 
 bool Slab::Compact(const std::vector<std::pair<Bound, Bound>>& borders,
-                   uint64_t compact_rev, uint64_t timeout_rev, std::string* err) {
+                   uint64_t compact_rev, const std::vector<uint64_t>& timeout_revs,
+                   std::string* err) {
+  if (timeout_revs.size() != borders.size()) {
+    if (err) *err = "Compact: timeout_revs must be per border pair";
+    return false;
+  }
   Impl* I = p;
   if (!Fold(err)) return false;  // compaction sweeps the single base run
   int64_t n = I->n;
@@ -1520,7 +1526,7 @@ bool Slab::Compact(const std::vector<std::pair<Bound, Bound>>& borders,
     if (hi <= lo) continue;
     hipLaunchKernelGGL(k_compact_mark, dim3((uint32_t)ceil_div(hi - lo, 256)),
                        dim3(256), 0, I->stream, I->A.meta, I->A.rev, I->A.vo, lo,
-                       hi, compact_rev, timeout_rev, I->s_a);
+                       hi, compact_rev, timeout_revs[i / 2], I->s_a);
   }
   // new row index
   uint64_t kept = 0;

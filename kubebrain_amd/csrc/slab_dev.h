@@ -40,6 +40,12 @@ struct DevRangeQ {
   uint8_t start[KEYW];
   uint8_t end[KEYW];
   uint64_t read_rev;
+  // scan begins at the first row with (key,rev) >= (start, start_rev).
+  // 0 = inclusive start-of-key (the normal case); UINT64_MAX = strictly
+  // after every row of `start` — the exclusive continuation bound chunked
+  // List/Stream use (exact even for keys of the full KEYW width, where a
+  // key-suffix trick would truncate)
+  uint64_t start_rev;
   int64_t cap;        // winner-write cap per query (limit+1); <=0 => unbounded
   int32_t count_only;
   int32_t _pad;
@@ -113,6 +119,9 @@ class Slab {
   int64_t delta_rows() const;
   int64_t delta_capacity() const;
   int64_t heap_used() const;
+  // per-query winner cap of the device arena (KB_MAX_CAP); chunked List
+  // clamps each chunk's cap to this and continues the frontier loop
+  int64_t max_winner_cap() const;
 
   // merge sorted delta rows straight into the BASE run (GPU merge by ranks;
   // delta rev-rows REPLACE base rev-rows of the same key). Used by tests and
@@ -152,9 +161,13 @@ class Slab {
 
   // compaction mark+sweep over encoded borders (compact.go:55-68 +
   // scanner.go:444-491, 566-591). Bounds are (key96, rev) pairs.
+  // timeout_revs[i] is the TTL timeout revision of border pair i — the
+  // reference pops one per scanner.Compact scan (scanner.go:147-177), so
+  // with >=2 pairs the values differ per pair.
   struct Bound { uint8_t key[KEYW]; uint64_t rev; };
   bool Compact(const std::vector<std::pair<Bound, Bound>>& borders,
-               uint64_t compact_rev, uint64_t timeout_rev, std::string* err);
+               uint64_t compact_rev, const std::vector<uint64_t>& timeout_revs,
+               std::string* err);
 
   // full slab dump for parity diffs (debug; D2H of all columns + used heap)
   bool Dump(std::vector<DumpRow>* rows_out, std::string* err);

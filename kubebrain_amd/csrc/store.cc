@@ -286,10 +286,11 @@ void Store::pumpEvents() {
         }
       }
       if (w.queue.size() > kWatchQueueCap) {
-        // drop slow consumer (watcherhub.go:84-94)
+        // drop slow consumer (watcherhub.go:84-94); the device slot is
+        // recycled immediately — only WatchPoll's "dropped" notice remains
         w.dropped = true;
         w.queue.clear();
-        slab_->WatcherClear(w.slot);
+        releaseSlot(w);
       }
     }
   }
@@ -328,22 +329,31 @@ Status Store::StreamNext(int64_t sid, std::vector<KeyValue>* kvs) {
   if (!syncReads(&err)) return INTERNAL;
   Status cst = checkCompactRace(ss.read_rev);
   if (cst != OK) { streams_.erase(it); return cst; }
-  DevRangeQ q{};
-  memset(q.start, 0, KEYW);
-  memcpy(q.start, ss.frontier.data(), std::min(ss.frontier.size(), (size_t)KEYW));
-  pad96(ss.end, q.end);
-  q.read_rev = ss.read_rev;
-  q.cap = 300;
-  q.count_only = 0;
+  // assemble one 300-winner batch; chunked when the device winner arena
+  // (max_winner_cap) is smaller than the batch. Continuation = exclusive
+  // (key, rev=+inf) bound past the last chunk's final key (exact for
+  // full-width keys; see List).
+  const int64_t max_cap = slab_->max_winner_cap();
   std::vector<kbslab::RangeResult> outs;
-  if (!slab_->RangeBatch({q}, true, &outs, &err)) return INTERNAL;
-  kbslab::RangeResult& r = outs[0];
-  if (r.overflow) return NOBUF;
-  for (auto& rec : r.recs) kvs->push_back(KeyValue{rec.key, rec.val, rec.rev});
-  if (r.written < 300) {
-    ss.done = true;  // next call returns the end marker
-  } else {
-    ss.frontier = r.recs.back().key + Bytes("\x01", 1);
+  while ((int64_t)kvs->size() < 300) {
+    DevRangeQ q{};
+    memset(q.start, 0, KEYW);
+    memcpy(q.start, ss.frontier.data(), std::min(ss.frontier.size(), (size_t)KEYW));
+    pad96(ss.end, q.end);
+    q.read_rev = ss.read_rev;
+    q.start_rev = ss.started ? UINT64_MAX : 0;
+    q.cap = std::min<int64_t>(300 - (int64_t)kvs->size(), max_cap);
+    q.count_only = 0;
+    if (!slab_->RangeBatch({q}, true, &outs, &err)) return INTERNAL;
+    kbslab::RangeResult& r = outs[0];
+    if (r.overflow) return NOBUF;
+    for (auto& rec : r.recs) kvs->push_back(KeyValue{rec.key, rec.val, rec.rev});
+    bool more = r.written >= q.cap || r.total > r.written;
+    if (!r.recs.empty()) {
+      ss.frontier = r.recs.back().key;
+      ss.started = true;
+    }
+    if (!more || r.recs.empty()) { ss.done = true; break; }
   }
   if (kvs->empty()) streams_.erase(it);  // empty == end marker now
   return OK;
@@ -413,6 +423,7 @@ int64_t Store::Watch(const Bytes& prefix, uint64_t revision, Status* st) {
   pad96(prefix, p96);
   std::string err;
   if (!slab_->WatcherSet(slot, p96, (uint32_t)prefix.size(), w.from_rev, &err)) {
+    free_slots_.push_back(slot);  // the reserved slot must not leak
     *st = INTERNAL;
     return -1;
   }
@@ -422,13 +433,43 @@ int64_t Store::Watch(const Bytes& prefix, uint64_t revision, Status* st) {
   return wid;
 }
 
+void Store::releaseSlot(Watcher& w) {
+  // every drop path returns the device slot (slow-consumer drops included),
+  // so the watcher table and k_watch_filter's per-event work stay bounded
+  // under watcher churn
+  if (w.slot < 0) return;
+  slab_->WatcherClear(w.slot);
+  free_slots_.push_back(w.slot);
+  w.slot = -1;
+}
+
 std::vector<Event> Store::WatchPoll(int64_t wid, Status* st) {
+  return WatchPollLimited(wid, SIZE_MAX, nullptr, st);
+}
+
+// Drains the watcher queue only if its serialized size (4 + per event
+// 28+klen+vlen — the kb_watch_poll wire format) fits max_bytes; otherwise
+// returns NOBUF with the queue INTACT so a retry with a larger buffer still
+// sees every event (contiguous-revision delivery guarantee).
+std::vector<Event> Store::WatchPollLimited(int64_t wid, size_t max_bytes,
+                                           size_t* need_bytes, Status* st) {
   std::lock_guard<std::recursive_mutex> lk(mu_);
   pumpEvents();
+  if (need_bytes) *need_bytes = 0;
   auto it = watchers_.find(wid);
   if (it == watchers_.end()) { *st = WATCH_DROPPED; return {}; }
   Watcher& w = it->second;
-  if (w.dropped) { *st = WATCH_DROPPED; watchers_.erase(it); return {}; }
+  if (w.dropped) {
+    *st = WATCH_DROPPED;
+    releaseSlot(w);  // no-op if pumpEvents already recycled it
+    watchers_.erase(it);
+    return {};
+  }
+  size_t need = 4;
+  for (const Event& e : w.queue)
+    need += 28 + e.kv_key.size() + e.kv_value.size();
+  if (need_bytes) *need_bytes = need;
+  if (need > max_bytes) { *st = NOBUF; return {}; }
   std::vector<Event> out(w.queue.begin(), w.queue.end());
   w.queue.clear();
   *st = OK;
@@ -439,8 +480,7 @@ void Store::WatchCancel(int64_t wid) {
   std::lock_guard<std::recursive_mutex> lk(mu_);
   auto it = watchers_.find(wid);
   if (it == watchers_.end()) return;
-  slab_->WatcherClear(it->second.slot);
-  free_slots_.push_back(it->second.slot);
+  releaseSlot(it->second);
   watchers_.erase(it);
 }
 
@@ -686,12 +726,18 @@ RangeResponse Store::List(const Bytes& start, const Bytes& end,
   if (!syncReads(&err)) { *st = INTERNAL; return resp; }
 
   // GPU winners over base+delta (merged device-side), fetched in chunks.
-  // Continuation bound trick: winner_key+0x01 excludes only the winner itself
-  // (key bytes <= 0x24 are rejected), so the next chunk resumes after it.
+  // Each chunk's cap is clamped to the device winner arena (max_winner_cap);
+  // the frontier loop continues until `lim` is filled or the range is
+  // exhausted, so limits larger than the arena stay exact. Continuation is an
+  // exclusive (key, rev=+inf) bound (DevRangeQ.start_rev = UINT64_MAX):
+  // the next chunk resumes strictly after every row of the last winner's key
+  // — exact even for keys of the full 96B width.
+  const int64_t max_cap = slab_->max_winner_cap();
   std::vector<kbslab::RangeResult> outs;
   std::vector<KeyValue> kvs;
   bool dev_more = true;
   Bytes dev_frontier = start;
+  uint64_t frontier_rev = 0;  // first chunk: inclusive start-of-key
   auto need = [&]() { return lim <= 0 || (int64_t)kvs.size() < lim; };
   int64_t chunk_cap = 0;  // 0 = unbounded; halved on arena overflow
   while (need() && dev_more) {
@@ -700,8 +746,10 @@ RangeResponse Store::List(const Bytes& start, const Bytes& end,
     memcpy(q.start, dev_frontier.data(), std::min(dev_frontier.size(), (size_t)KEYW));
     pad96(end, q.end);
     q.read_rev = reqRevision;
+    q.start_rev = frontier_rev;
     q.cap = lim > 0 ? lim - (int64_t)kvs.size() : chunk_cap;
     if (chunk_cap > 0 && (q.cap <= 0 || q.cap > chunk_cap)) q.cap = chunk_cap;
+    if (q.cap <= 0 || q.cap > max_cap) q.cap = max_cap;
     q.count_only = 0;
     if (!slab_->RangeBatch({q}, true, &outs, &err)) { *st = INTERNAL; return resp; }
     kbslab::RangeResult& r = outs[0];
@@ -717,10 +765,12 @@ RangeResponse Store::List(const Bytes& start, const Bytes& end,
       if (!need()) break;
       kvs.push_back(KeyValue{rec.key, rec.val, rec.rev});
     }
-    dev_more = (q.cap > 0 && r.written >= q.cap) ||
-               (q.cap <= 0 && r.total > r.written);
+    // a chunk that filled its (clamped) cap, or saw more winners than it
+    // materialized, may have more rows past the frontier
+    dev_more = r.written >= q.cap || r.total > r.written;
     if (r.recs.empty()) break;  // no progress => exhausted
-    dev_frontier = r.recs.back().key + Bytes("\x01", 1);
+    dev_frontier = r.recs.back().key;
+    frontier_rev = UINT64_MAX;  // strictly after the last winner's key
   }
 
   resp.header_revision = curRevision;
@@ -816,22 +866,26 @@ uint64_t Store::Compact(uint64_t revision, Status* st) {
   }
   std::sort(borders.begin(), borders.end());
   std::vector<std::pair<kbslab::Slab::Bound, kbslab::Slab::Bound>> bpairs;
-  uint64_t timeout_rev = 0;
+  std::vector<uint64_t> timeout_revs;
+  std::vector<std::pair<Bytes, Bytes>> pair_keys;
   for (size_t i = 0; i + 1 < borders.size(); i += 2) {
     // scanner.Compact: logCompactHistory + checkCompactRace(compact) Put +
-    // per-scan timeout revision (scanner.go:147-198, 594-603)
+    // per-scan timeout revision (scanner.go:147-198, 594-603) — one history
+    // record is pushed AND one timeout revision popped PER pair, so the
+    // values differ across pairs and each pair's scan must use its own
     compact_histories_.push_back(CompactRecord{revision, now_});
     compact_cell_set_ = true;
     compact_cell_ = revision;
-    timeout_rev = getTimeoutRevision();  // identical across pairs in one call
+    timeout_revs.push_back(getTimeoutRevision());
     kbslab::Slab::Bound lo{}, hi{};
     pad96(borders[i], lo.key);
     lo.rev = 0;
     pad96(borders[i + 1], hi.key);
     hi.rev = 0;
     bpairs.push_back({lo, hi});
+    pair_keys.push_back({borders[i], borders[i + 1]});
   }
-  if (!slab_->Compact(bpairs, revision, timeout_rev, &err)) {
+  if (!slab_->Compact(bpairs, revision, timeout_revs, &err)) {
     *st = INTERNAL;
     return revision;
   }
@@ -843,15 +897,25 @@ uint64_t Store::Compact(uint64_t revision, Status* st) {
       it = tombstoned_.erase(it);
     } else ++it;
   }
-  if (timeout_rev != 0) {
-    for (auto it = events_keys_.begin(); it != events_keys_.end();) {
-      auto ri = revIndex_.find(*it);
-      if (ri != revIndex_.end() && ri->second.rev <= timeout_rev) {
-        tombstoned_.erase(*it);
-        revIndex_.erase(ri);
-        it = events_keys_.erase(it);
-      } else ++it;
+  // TTL'd /events/ keys, per pair with that pair's timeout revision
+  for (auto it = events_keys_.begin(); it != events_keys_.end();) {
+    bool dead = false;
+    auto ri = revIndex_.find(*it);
+    if (ri != revIndex_.end()) {
+      for (size_t i = 0; i < pair_keys.size(); ++i) {
+        if (timeout_revs[i] == 0) continue;
+        if (*it >= pair_keys[i].first && *it < pair_keys[i].second &&
+            ri->second.rev <= timeout_revs[i]) {
+          dead = true;
+          break;
+        }
+      }
     }
+    if (dead) {
+      tombstoned_.erase(*it);
+      revIndex_.erase(ri);
+      it = events_keys_.erase(it);
+    } else ++it;
   }
   *st = OK;
   return revision;

@@ -90,6 +90,10 @@ class Store {
 
   int64_t Watch(const Bytes& prefix, uint64_t revision, Status* st);
   std::vector<Event> WatchPoll(int64_t wid, Status* st);
+  // non-destructive on overflow: if the serialized size exceeds max_bytes,
+  // returns NOBUF (need_bytes = required size) with the queue intact
+  std::vector<Event> WatchPollLimited(int64_t wid, size_t max_bytes,
+                                      size_t* need_bytes, Status* st);
   void WatchCancel(int64_t wid);
 
   void ClockAdvance(int64_t secs);
@@ -133,6 +137,8 @@ class Store {
   void putRow(const Bytes& key, uint64_t rev, const Bytes& val);
   void putRevRow(const Bytes& key, uint64_t objrev, bool flag9);
   void pumpEvents();  // fan-out pending events via the GPU filter
+  struct Watcher;
+  void releaseSlot(Watcher& w);  // recycle a device watcher slot (idempotent)
   // push pending values + new rows to the device (delta-run merge); folds the
   // delta into the base run past the fold threshold
   bool syncReads(std::string* err);
@@ -179,7 +185,7 @@ class Store {
   std::vector<Event> event_log_;
   std::vector<Event> pending_;  // not yet fanned out
   struct Watcher {
-    int64_t slot;
+    int64_t slot;  // device slot; -1 once released (releaseSlot)
     Bytes prefix;
     uint64_t from_rev;
     std::deque<Event> queue;
@@ -198,7 +204,12 @@ class Store {
   int64_t sync_n_ = 0;
   std::string fatal_;  // first unrecoverable device error (e.g. slab full)
   std::unordered_set<Bytes> delta_revkeys_;  // keys with rev-rows in the delta run
-  struct StreamState { Bytes frontier, end; uint64_t read_rev; bool done; };
+  struct StreamState {
+    Bytes frontier, end;
+    uint64_t read_rev;
+    bool done;
+    bool started = false;  // frontier is a continuation (exclusive key bound)
+  };
   std::unordered_map<int64_t, StreamState> streams_;
   int64_t next_sid_ = 1;
 };

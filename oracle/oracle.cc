@@ -642,18 +642,34 @@ int64_t Backend::Watch(const Bytes& prefix, uint64_t revision, Status* st) {
 }
 
 std::vector<Event> Backend::WatchPoll(int64_t wid, Status* st) {
+  return WatchPollLimited(wid, SIZE_MAX, nullptr, st);
+}
+
+// Same overflow contract as the product's kb_watch_poll: if the serialized
+// size (4 + per event 28+klen+vlen) exceeds max_bytes, return NOBUF without
+// consuming anything so a retry sees every event.
+std::vector<Event> Backend::WatchPollLimited(int64_t wid, size_t max_bytes,
+                                             size_t* need_bytes, Status* st) {
+  if (need_bytes) *need_bytes = 0;
   auto it = watchers_.find(wid);
   if (it == watchers_.end()) { *st = WATCH_DROPPED; return {}; }
   Watcher& w = it->second;
-  std::vector<Event> out = std::move(w.pending);
-  w.pending.clear();
+  std::vector<Event> out;
+  for (const Event& e : w.pending) out.push_back(e);
   // processEvents: filterByRevision then filterByPrefix (watch.go:119-133)
-  for (; w.log_pos < event_log_.size(); ++w.log_pos) {
-    const Event& e = event_log_[w.log_pos];
+  size_t pos = w.log_pos;
+  for (; pos < event_log_.size(); ++pos) {
+    const Event& e = event_log_[pos];
     if (e.revision < w.from_rev) continue;
     if (e.kv_key.compare(0, w.prefix.size(), w.prefix) != 0) continue;
     out.push_back(e);
   }
+  size_t need = 4;
+  for (const Event& e : out) need += 28 + e.kv_key.size() + e.kv_value.size();
+  if (need_bytes) *need_bytes = need;
+  if (need > max_bytes) { *st = NOBUF; return {}; }
+  w.pending.clear();
+  w.log_pos = pos;
   *st = OK;
   return out;
 }

@@ -41,6 +41,7 @@ enum Status : int32_t {
   KEYTOOLONG = 11,     // product-only; oracle has no width limit
   BADKEY = 12,
   INTERNAL = 13,
+  NOBUF = 100,         // caller buffer too small (== KB_ENOBUF)
 };
 
 using Bytes = std::string;  // raw byte strings
@@ -154,6 +155,10 @@ class Backend {
   int64_t Watch(const Bytes& prefix, uint64_t revision, Status* st);
   // Drain pending events for watcher (delivery batches flattened).
   std::vector<Event> WatchPoll(int64_t wid, Status* st);
+  // non-destructive on overflow (same contract as kb_watch_poll): if the
+  // serialized size exceeds max_bytes, returns NOBUF with the queue intact
+  std::vector<Event> WatchPollLimited(int64_t wid, size_t max_bytes,
+                                      size_t* need_bytes, Status* st);
   void WatchCancel(int64_t wid);
 
   // -- config/test hooks --

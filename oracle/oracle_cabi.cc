@@ -171,8 +171,12 @@ long long okb_watch(void* h, const uint8_t* prefix, size_t plen, uint64_t rev,
 }
 
 int okb_watch_poll(void* h, long long wid, uint8_t* out, size_t cap, size_t* out_len) {
+  // overflow-safe like kb_watch_poll: KB_ENOBUF leaves the queue intact and
+  // *out_len carries the required size
   Status st;
-  auto evs = ((Backend*)h)->WatchPoll(wid, &st);
+  size_t need = 0;
+  auto evs = ((Backend*)h)->WatchPollLimited(wid, cap, &need, &st);
+  if (st == oracle::NOBUF) { *out_len = need; return kEnoBuf; }
   Writer w{out, cap};
   writeEvents(w, evs);
   *out_len = w.off;

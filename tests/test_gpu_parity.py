@@ -341,3 +341,124 @@ def test_skipped_prefixes_compaction():
         d.list(b"/registry/pods/ns-00/", b"/registry/pods/ns-000", cur - 25, 0)
     finally:
         d.close()
+
+
+def test_list_limit_beyond_winner_cap():
+    """Limits larger than the device winner arena (KB_MAX_CAP) stay exact:
+    List's chunked frontier loop clamps each chunk to the arena and keeps
+    going instead of silently truncating with more=false (advisor r1,
+    store.cc List)."""
+    os.environ["KB_MAX_CAP"] = "8"
+    d = parity.Dual()
+    try:
+        ns = b"/registry/pods/ns-77"
+        for i in range(40):
+            d.create(ns + b"/o-%05d" % i, b"v%d" % i)
+        r = d.list(ns + b"/", ns + b"0", 0, 25)   # limit 25 > cap 8
+        assert len(r.kvs) == 25 and r.more
+        r = d.list(ns + b"/", ns + b"0", 0, 0)    # unlimited
+        assert len(r.kvs) == 40 and not r.more
+        r = d.list(ns + b"/", ns + b"0", 0, 40)   # limit == total
+        assert len(r.kvs) == 40 and not r.more
+        r = d.list(ns + b"/", ns + b"0", 0, 60)   # limit > total
+        assert len(r.kvs) == 40 and not r.more
+    finally:
+        d.close()
+        del os.environ["KB_MAX_CAP"]
+
+
+def test_full_width_key_continuation():
+    """Keys of exactly 96 bytes (the full key-column width) across chunk and
+    stream-batch borders: the continuation is an exclusive (key, rev) bound
+    (DevRangeQ.start_rev), so no row is duplicated or re-scanned and
+    unlimited List terminates (advisor r1, store.cc List/StreamNext)."""
+    os.environ["KB_MAX_CAP"] = "4"
+    d = parity.Dual()
+    try:
+        base = b"/registry/pods/ns-96/"
+        keys = []
+        for i in range(12):
+            k = base + b"k%05d" % i
+            k += b"x" * (96 - len(k))  # exactly KEYW bytes
+            assert len(k) == 96
+            keys.append(k)
+            d.create(k, b"w%d" % i)
+        end = b"/registry/pods/ns-960"
+        r = d.list(base, end, 0, 0)
+        assert [kv.key for kv in r.kvs] == keys
+        r = d.list(base, end, 0, 7)
+        assert len(r.kvs) == 7 and r.more
+        d.update(keys[5], b"new", 0)
+        d.list(base, end, 0, 0)
+        batches = d.stream(base, end, 0)
+        assert sum(len(b) for b in batches) == 12
+        d.diff_dump()
+    finally:
+        d.close()
+        del os.environ["KB_MAX_CAP"]
+
+
+def test_watch_poll_enobuf_preserves_events():
+    """kb_watch_poll with a too-small buffer returns KB_ENOBUF without
+    consuming the queue (advisor r1, cabi.cc): the retry with a larger
+    buffer delivers every event, on both ABIs."""
+    import ctypes as C
+
+    import kbclient
+    d = parity.Dual()
+    try:
+        pfx = b"/registry/pods/ns-44"
+        w = d.watch(pfx + b"/", 0)
+        for i in range(10):
+            d.create(pfx + b"/e-%d" % i, b"val-%d" % i)
+        wo, wp = d.watches[w]
+        small = C.create_string_buffer(16)
+        need_p = C.c_size_t()
+        need_o = C.c_size_t()
+        rc_p = d.p._f("watch_poll")(C.c_void_p(d.p.h), C.c_longlong(wp),
+                                    small, C.c_size_t(16), C.byref(need_p))
+        rc_o = d.o._f("watch_poll")(C.c_void_p(d.o.h), C.c_longlong(wo),
+                                    small, C.c_size_t(16), C.byref(need_o))
+        assert rc_p == kbclient.ENOBUF and rc_o == kbclient.ENOBUF
+        assert need_p.value == need_o.value > 16  # required size reported
+        evs = d.poll(w)  # retry with the big buffer: nothing was lost
+        assert len(evs) == 10
+    finally:
+        d.close()
+
+
+def test_compact_multi_pair_ttl():
+    """TTL expiry with SkippedPrefixes (>=2 border pairs): each pair's scan
+    uses its OWN popped timeout revision — the reference pops one per
+    scanner.Compact scan (scanner.go:147-177), so /events/ TTL expiry runs
+    in the first pair even when a later pair's timeout is 0 (advisor r1,
+    store.cc Compact). Also covers >=3 TTL-spaced compaction cycles
+    (VERDICT r1 weak #7)."""
+    import ctypes as C
+    d = parity.Dual(events_ttl=1)
+    try:
+        skipped = b"/registry/zz"
+        d.o.lib.okb_set_skipped_prefixes(C.c_void_p(d.o.h), skipped)
+        d.p.lib.kb_set_skipped_prefixes(C.c_void_p(d.p.h), skipped)
+        pfx = b"/registry/events/ns-9"
+        for i in range(6):
+            d.create(pfx + b"/ev-%d" % i, b"e%d" % i)
+        d.create(b"/registry/zz/obj", b"zz")
+        d.compact(0)           # logs one history record per border pair
+        d.clock_advance(2)     # expire those records
+        d.create(b"/registry/other/x", b"x")  # consume a revision
+        d.compact(0)           # pair 1 pops its timeout -> events expire
+        d.list(pfx + b"/", pfx + b"0", 0, 0)
+        d.diff_dump()
+        # cycle 3+: history pop across repeated TTL-spaced compactions
+        for i in range(3):
+            d.create(pfx + b"/late-%d" % i, b"l%d" % i)
+        d.clock_advance(2)
+        d.compact(0)
+        d.clock_advance(2)
+        d.compact(0)
+        d.list(pfx + b"/", pfx + b"0", 0, 0)
+        d.diff_dump()
+        d.diff_event_log()
+    finally:
+        d.close()
