@@ -1,14 +1,18 @@
 #!/usr/bin/env python3
-"""bench.py — the measured hot path (BASELINE.json configs[1] at N=1).
+"""bench.py — the measured hot path (BASELINE.json's 10M-key headline config
+at N=1: the metric is quoted on a "10M-key MVCC slab").
 
-Workload: 1M Pod-style keys / 200k extra live revisions (zipf 1.1) / 2%
+Workload: 10M Pod-style keys / 1M extra live revisions (zipf 1.1) / 2%
 tombstones on one MI355X; one step = 1000 ops = 900 batched Range(limit=500)
 over random namespace prefixes + 100 Txn conditional updates (90/10 mix,
 docs/benchmark.md-style 512B values, 300-client-style batching). `value` is
 whole-job ops/s with the slab resident in HBM and Range results landing in
-the device output arena; the PCIe-inclusive rate is reported separately as
-`ops_per_sec_with_d2h`. A separate watch leg measures GPU
-fan-out deliveries/s and is reported as `watch_events_per_sec`.
+the device output arena; the PCIe-inclusive rates are reported separately as
+`ops_per_sec_with_d2h` (full records, pipelined payload copy) and
+`ops_per_sec_with_d2h_keys_only` (etcd3 KeysOnly semantics — key+mod-rev
+only; the reference's shim ignores that flag, so this is an extension and is
+never `value`). The default run also times the configs[2] compaction sweep
+and the 10k-watcher fan-out leg (configs[4] shape at N=1).
 
 Multi-GPU (--gpus N via torch.distributed.run): keys shard by namespace hash,
 one store per GPU; the query stream routes by namespace; weak scaling
@@ -64,15 +68,19 @@ def build_store(store, namespaces, keys, rng, extra_revs, tomb_frac, my_ns=None)
     else:
         sel = keys
     n = len(sel)
-    # values: seeded random bytes
-    vals = rng.integers(0, 256, size=n * VAL_LEN, dtype=np.uint8).tobytes()
-    klens = np.array([len(k) for k in sel], dtype=np.uint32).tobytes()
-    vlens = np.full(n, VAL_LEN, dtype=np.uint32).tobytes()
-    kblob = b"".join(sel)
-    phase("gen values/keys")
+    # chunked load (1M keys per call) bounds host memory at 10M-key scale
     f = store._f("bulk_create")
-    rc = f(ctypes.c_void_p(store.h), kblob, klens, vals, vlens, ctypes.c_size_t(n))
-    assert rc == 0, "bulk_create failed"
+    CH = 1 << 20
+    for c0 in range(0, n, CH):
+        ck = sel[c0:c0 + CH]
+        m = len(ck)
+        vals = rng.integers(0, 256, size=m * VAL_LEN, dtype=np.uint8).tobytes()
+        klens = np.array([len(k) for k in ck], dtype=np.uint32).tobytes()
+        vlens = np.full(m, VAL_LEN, dtype=np.uint32).tobytes()
+        kblob = b"".join(ck)
+        rc = f(ctypes.c_void_p(store.h), kblob, klens, vals, vlens,
+               ctypes.c_size_t(m))
+        assert rc == 0, "bulk_create failed"
     phase("bulk_create")
     revs = {k: None for k in sel}  # latest rev per key tracked client-side
     base = store.current_rev() - n
@@ -357,14 +365,16 @@ def main():
     ap.add_argument("--gpus", type=int, default=1)
     ap.add_argument("--steps", type=int, default=24)
     ap.add_argument("--warmup", type=int, default=6)
+    # the headline config (BASELINE.json metric): 10M keys / 1M extra revs
     ap.add_argument("--nns", type=int, default=2000)
-    ap.add_argument("--per-ns", type=int, default=500)
-    ap.add_argument("--extra-revs", type=int, default=200000)
-    ap.add_argument("--watchers", type=int, default=1000)
+    ap.add_argument("--per-ns", type=int, default=5000)
+    ap.add_argument("--extra-revs", type=int, default=1000000)
+    ap.add_argument("--watchers", type=int, default=10000)
     ap.add_argument("--watch-events", type=int, default=6000)
     ap.add_argument("--no-cpu-baseline", action="store_true")
-    ap.add_argument("--compact-bench", action="store_true",
-                    help="also time a full compaction sweep (configs[2] shape)")
+    ap.add_argument("--no-compact-bench", action="store_true",
+                    help="skip the compaction-sweep leg (configs[2] shape; "
+                         "on by default)")
     args = ap.parse_args()
 
     rank = int(os.environ.get("RANK", "0"))
@@ -463,16 +473,17 @@ def main():
 
     t_split = {"range_s": 0.0, "txn_s": 0.0}
 
-    def one_step(d2h=False):
+    def one_step(mode=0):
         # one call: range batch launched async, txn batch overlapped on the
-        # host while the kernels are in flight (kb_bench_step)
+        # host while the kernels are in flight (kb_bench_step). mode bit0 =
+        # d2h (pipelined payload copy), bit1 = keys_only.
         i = step_i[0] % n_pre
         step_i[0] += 1
         total = ctypes.c_ulonglong()
         secs = ctypes.c_double()
         rc = fstep(ctypes.c_void_p(store.h), pre_blobs[i], ctypes.c_size_t(nq),
                    tx_blobs[i], ctypes.c_size_t(ntx),
-                   ctypes.c_int(1 if d2h else 0), tx_out_ptr,
+                   ctypes.c_int(mode), tx_out_ptr,
                    ctypes.byref(total), ctypes.byref(secs))
         assert rc == 0
         t_split["range_s"] += secs.value
@@ -480,6 +491,9 @@ def main():
             if nr != 0:
                 revs[k] = int(nr)
         return total.value
+
+    def drain():
+        assert store._f("sync")(ctypes.c_void_p(store.h)) == 0
 
     def barrier():
         if dist:
@@ -505,12 +519,22 @@ def main():
         elapsed = float(e.item())
     p = perf(store)
 
-    # PCIe-inclusive measurement (separate, untimed-region)
+    # PCIe-inclusive measurement (separate, untimed-region): full records,
+    # payload copy pipelined across steps (kb_sync drains the tail before the
+    # clock stops)
+    d2h_steps = max(4, args.steps // 4)
     t0 = time.time()
-    for _ in range(max(2, args.steps // 8)):
-        one_step(d2h=True)
+    for _ in range(d2h_steps):
+        one_step(mode=1)
+    drain()
     d2h_elapsed = time.time() - t0
-    d2h_steps = max(2, args.steps // 8)
+    # keys-only variant (etcd3 KeysOnly semantics; response payload is
+    # key + mod-revision per winner — an extension, see module docstring)
+    t0 = time.time()
+    for _ in range(d2h_steps):
+        one_step(mode=3)
+    drain()
+    d2h_ko_elapsed = time.time() - t0
 
     # cross-shard Range leg (configs[3]) at N>1: RCCL/xGMI exchange + merge.
     # Exception-guarded: a failure degrades to a JSON note, never the run.
@@ -529,7 +553,7 @@ def main():
     # compaction sweep leg (configs[2]: drop revisions < compactRev; the
     # sweep reads the whole slab and stream-compacts rows + value heap)
     compact_stats = None
-    if args.compact_bench:
+    if not args.no_compact_bench:
         p_pre = perf(store)
         rows_pre = p_pre["slab_rows"]
         heap_pre = p_pre["heap_used"]
@@ -617,22 +641,27 @@ def main():
             "dtype": "u8",
             "data": "synthetic",
             "config": {
-                "workload": "configs[1]: 1M Pod-style keys / 200k extra revisions "
-                            "(zipf 1.1) / 2% tombstones, Range(limit=500)+Txn "
-                            "conditional-update 90/10, 512B values, single MI355X"
+                "workload": f"10M-key MVCC slab headline config: "
+                            f"{args.nns * args.per_ns // 1000000}M Pod-style keys / "
+                            f"{args.extra_revs} extra revisions (zipf 1.1) / 2% "
+                            f"tombstones, Range(limit=500)+Txn conditional-update "
+                            f"90/10, 512B values, single MI355X"
                             if world == 1 else
-                            f"configs[1] sharded by namespace hash over {world} "
-                            f"GPUs (weak scaling)",
+                            f"10M-key headline config sharded by namespace hash "
+                            f"over {world} GPUs (weak scaling)",
                 "n_keys": args.nns * args.per_ns,
                 "extra_revs": args.extra_revs,
                 "limit": LIMIT,
                 "ops_per_step": OPS_PER_STEP,
+                "watchers": max(args.watchers // world, 8) * world,
             },
             "cross_shard_range": cross_shard,
             "compact_sweep": compact_stats,
             "watch_events_per_sec": round(wrate, 1),
             "watch_delivered_rank0": delivered,
             "ops_per_sec_with_d2h": round(OPS_PER_STEP * d2h_steps * world / d2h_elapsed, 1),
+            "ops_per_sec_with_d2h_keys_only": round(
+                OPS_PER_STEP * d2h_steps * world / d2h_ko_elapsed, 1),
             "step_split_ms": {"range": round(split_snapshot["range_s"] / args.steps * 1e3, 3),
                                "txn": round(split_snapshot["txn_s"] / args.steps * 1e3, 3)},
             "roofline": roofline,

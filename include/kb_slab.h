@@ -136,20 +136,28 @@ int kb_dump(kb_store*, uint8_t* out, size_t cap, size_t* out_len, uint64_t* n_ro
 int kb_event_log(kb_store*, uint8_t* out, size_t cap, size_t* out_len);
 
 /* ---- bench support (the measured hot path; used by bench.py) ---- */
-/* n Range queries in one device batch; returns total winners. d2h!=0 copies
- * results to host (PCIe-inclusive mode); d2h==0 leaves them in the device
- * arena (the `value` mode per DESIGN.md §5). */
-int kb_bench_range(kb_store*, const uint8_t* qbuf, size_t nq, int d2h,
+/* n Range queries in one device batch; returns total winners. mode bits:
+ * 1 = d2h — results copied to pinned host memory, PIPELINED (the payload
+ *     copy overlaps the next batch's kernels; call kb_sync before reading
+ *     wall-clock);
+ * 2 = keys_only — etcd3 RangeRequest.KeysOnly semantics: key + mod-revision
+ *     per winner, no value bytes (an extension: the reference's etcd shim
+ *     ignores the flag, server/etcd/kv.go:48-67).
+ * mode 0 leaves results in the device arena (the `value` mode). */
+int kb_bench_range(kb_store*, const uint8_t* qbuf, size_t nq, int mode,
                    unsigned long long* total_kvs, double* secs);
+/* wait for in-flight pipelined D2H copies */
+int kb_sync(kb_store*);
 /* batched conditional updates (txn.go:249-265); tbuf = n × {u32 klen;
  * u64 prev_rev; u32 vlen; key; val}; out_revs[i] = new revision or 0 on CAS
  * failure */
 int kb_bench_txn(kb_store*, const uint8_t* tbuf, size_t n, uint64_t* out_revs);
 /* one bench step: Range batch launched async + txn batch overlapped on the
  * host while the kernels are in flight (results unchanged: kernels snapshot
- * device state at launch; writes stage host-side until the next sync) */
+ * device state at launch; writes stage host-side until the next sync).
+ * mode bits as kb_bench_range. */
 int kb_bench_step(kb_store*, const uint8_t* qbuf, size_t nq,
-                  const uint8_t* tbuf, size_t ntx, int d2h, uint64_t* out_revs,
+                  const uint8_t* tbuf, size_t ntx, int mode, uint64_t* out_revs,
                   unsigned long long* total, double* secs);
 /* batched deletes: dbuf = n × {u32 klen; u64 prev_rev; key} */
 int kb_bench_del(kb_store*, const uint8_t* dbuf, size_t n, uint64_t* out_revs);

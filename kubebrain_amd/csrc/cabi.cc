@@ -291,10 +291,19 @@ int kb_event_log(kb_store* h, uint8_t* out, size_t cap, size_t* out_len) {
   return 0;
 }
 
-int kb_bench_range(kb_store* h, const uint8_t* qbuf, size_t nq, int d2h,
+int kb_bench_range(kb_store* h, const uint8_t* qbuf, size_t nq, int mode,
                    unsigned long long* total, double* secs) {
   std::string err;
-  if (!((Store*)h)->BenchRange(qbuf, nq, d2h != 0, total, secs, &err)) {
+  if (!((Store*)h)->BenchRange(qbuf, nq, mode, total, secs, &err)) {
+    set_err(KB_EINTERNAL, err);
+    return KB_EINTERNAL;
+  }
+  return 0;
+}
+
+int kb_sync(kb_store* h) {
+  std::string err;
+  if (!((Store*)h)->Sync(&err)) {
     set_err(KB_EINTERNAL, err);
     return KB_EINTERNAL;
   }
@@ -311,10 +320,10 @@ int kb_bench_txn(kb_store* h, const uint8_t* tbuf, size_t n, uint64_t* out_revs)
 }
 
 int kb_bench_step(kb_store* h, const uint8_t* qbuf, size_t nq,
-                  const uint8_t* tbuf, size_t ntx, int d2h, uint64_t* out_revs,
+                  const uint8_t* tbuf, size_t ntx, int mode, uint64_t* out_revs,
                   unsigned long long* total, double* secs) {
   std::string err;
-  if (!((Store*)h)->BenchStep(qbuf, nq, tbuf, ntx, d2h != 0, out_revs, total,
+  if (!((Store*)h)->BenchStep(qbuf, nq, tbuf, ntx, mode, out_revs, total,
                               secs, &err)) {
     set_err(KB_EINTERNAL, err);
     return KB_EINTERNAL;

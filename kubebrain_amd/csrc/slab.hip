@@ -422,7 +422,8 @@ __global__ void k_gather(const uint8_t* __restrict__ bkeys,
                          const uint64_t* __restrict__ dvo,
                          const uint8_t* __restrict__ heap,
                          const uint64_t* __restrict__ rows_out, int64_t max_cap,
-                         const int64_t* __restrict__ found_out, int nq,
+                         const int64_t* __restrict__ found_out,
+                         const DevRangeQ* __restrict__ qs, int nq,
                          uint8_t* __restrict__ gbuf, int64_t qcap,
                          int64_t* __restrict__ offs,  // [q*max_cap+j] scratch
                          int64_t* __restrict__ gbytes_out,
@@ -431,6 +432,7 @@ __global__ void k_gather(const uint8_t* __restrict__ bkeys,
   int q = blockIdx.x;
   if (q >= nq) return;
   int64_t nwin = found_out[q];
+  const bool konly = qs[q].keys_only != 0;
   const uint64_t* rows = rows_out + (int64_t)q * max_cap;
   int64_t* qoffs = offs + (int64_t)q * max_cap;
   __shared__ int64_t lds[256];
@@ -444,7 +446,7 @@ __global__ void k_gather(const uint8_t* __restrict__ bkeys,
     if (j < nwin) {
       uint64_t rt = rows[j];
       uint64_t m = (rt & ROW_TAG_DELTA ? dmeta : bmeta)[rt & ROW_MASK];
-      sz = rec_bytes(meta_klen(m), meta_vlen(m));
+      sz = rec_bytes(meta_klen(m), konly ? 0 : meta_vlen(m));
     }
     lds[threadIdx.x] = sz;
     __syncthreads();
@@ -480,13 +482,15 @@ __global__ void k_gather_copy(const uint8_t* __restrict__ bkeys,
                               const uint8_t* __restrict__ heap,
                               const uint64_t* __restrict__ rows_out,
                               int64_t max_cap,
-                              const int64_t* __restrict__ found_out, int nq,
+                              const int64_t* __restrict__ found_out,
+                              const DevRangeQ* __restrict__ qs, int nq,
                               uint8_t* __restrict__ gbuf, int64_t qcap,
                               const int64_t* __restrict__ offs,
                               const int32_t* __restrict__ overflow) {
   int q = blockIdx.x;
   if (q >= nq || overflow[q]) return;
   int64_t nwin = found_out[q];
+  const bool konly = qs[q].keys_only != 0;
   const uint64_t* rows = rows_out + (int64_t)q * max_cap;
   const int64_t* qoffs = offs + (int64_t)q * max_cap;
   // 16-lane record groups — independent load chains across records
@@ -499,7 +503,7 @@ __global__ void k_gather_copy(const uint8_t* __restrict__ bkeys,
     bool isd = (rt & ROW_TAG_DELTA) != 0;
     int64_t row = (int64_t)(rt & ROW_MASK);
     uint64_t m = (isd ? dmeta : bmeta)[row];
-    uint32_t klen = meta_klen(m), vlen = meta_vlen(m);
+    uint32_t klen = meta_klen(m), vlen = konly ? 0 : meta_vlen(m);
     uint8_t* dst = qb + qoffs[j];
     if (gl == 0) {
       *(uint64_t*)dst = (isd ? drev : brev)[row];
@@ -511,6 +515,7 @@ __global__ void k_gather_copy(const uint8_t* __restrict__ bkeys,
     uint32_t kw = (klen + 15) & ~15u;  // key rows are 96B: in-bounds, 4-aligned
     for (uint32_t b = gl; b < kw / 4; b += 16)
       ((uint32_t*)kd)[b] = ((const uint32_t*)ks)[b];
+    if (konly) continue;
     const uint8_t* vs = heap + (isd ? dvo : bvo)[row];
     uint8_t* vd = dst + 16 + ((klen + 15) & ~15u);
     uint32_t w16 = vlen >> 4;
@@ -919,7 +924,57 @@ struct Slab::Impl {
     return true;
   }
 
+  // pipelined payload D2H (bench d2h mode): ping-pong pack arenas; each
+  // batch's payload copy runs on the copy stream and overlaps the NEXT
+  // batch's kernels on the compute stream (the PCIe copy and the scan use
+  // different engines). Slot reuse waits on that slot's prior copy.
+  hipStream_t cstream = nullptr;
+  uint8_t* d_packs[2] = {nullptr, nullptr};  // [0] aliases d_pack
+  uint8_t* h_packs[2] = {nullptr, nullptr};
+  int64_t h_packs_cap[2] = {0, 0};
+  hipEvent_t ev_pk[2] = {nullptr, nullptr}, ev_cp[2] = {nullptr, nullptr};
+  bool cp_busy[2] = {false, false};
+  int pk_idx = 0;
+  double* pack_ms_acc = nullptr;  // -> perf.pack_d2h_ms (set by Slab)
+
+  bool ensure_pipe(int idx, int64_t need, std::string* err) {
+    if (!cstream) {
+      HIP_CHECK(hipStreamCreate(&cstream));
+      HIP_CHECK(hipEventCreate(&ev_pk[0]));
+      HIP_CHECK(hipEventCreate(&ev_pk[1]));
+      HIP_CHECK(hipEventCreate(&ev_cp[0]));
+      HIP_CHECK(hipEventCreate(&ev_cp[1]));
+      d_packs[0] = d_pack;
+    }
+    if (idx == 1 && !d_packs[1]) HIP_CHECK(hipMalloc(&d_packs[1], arena_bytes));
+    if (need > h_packs_cap[idx]) {
+      if (h_packs[idx]) (void)hipHostFree(h_packs[idx]);
+      int64_t cap = need + need / 2;
+      HIP_CHECK(hipHostMalloc(&h_packs[idx], cap));
+      h_packs_cap[idx] = cap;
+    }
+    return true;
+  }
+
+  // wait for slot idx's in-flight copy (accumulating its event time)
+  void wait_slot(int idx) {
+    if (!cp_busy[idx]) return;
+    (void)hipEventSynchronize(ev_cp[idx]);
+    float ms = 0;
+    if (hipEventElapsedTime(&ms, ev_pk[idx], ev_cp[idx]) == hipSuccess &&
+        pack_ms_acc)
+      *pack_ms_acc += ms;
+    cp_busy[idx] = false;
+  }
+
   ~Impl() {
+    wait_slot(0);
+    wait_slot(1);
+    if (d_packs[1]) (void)hipFree(d_packs[1]);
+    for (int i = 0; i < 2; ++i) if (h_packs[i]) (void)hipHostFree(h_packs[i]);
+    for (hipEvent_t e : {ev_pk[0], ev_pk[1], ev_cp[0], ev_cp[1]})
+      if (e) (void)hipEventDestroy(e);
+    if (cstream) (void)hipStreamDestroy(cstream);
     for (void* p : {(void*)A.keys, (void*)A.meta, (void*)A.rev, (void*)A.vo,
                     (void*)B.keys, (void*)B.meta, (void*)B.rev, (void*)B.vo,
                     (void*)DA.keys, (void*)DA.meta, (void*)DA.rev, (void*)DA.vo,
@@ -1323,13 +1378,13 @@ bool Slab::RangeBatchStart(const std::vector<DevRangeQ>& qs, std::string* err) {
   hipLaunchKernelGGL(k_gather, dim3(nq), dim3(256), 0, I->stream, I->A.keys,
                      I->A.meta, I->A.rev, I->A.vo, I->DA.keys, I->DA.meta,
                      I->DA.rev, I->DA.vo, I->heapA, I->d_rowsm, I->max_cap,
-                     I->d_found, nq, I->d_gbuf, qcap, I->d_offs, I->d_gbytes,
-                     I->d_ovf, I->d_bytes);
+                     I->d_found, I->d_qs, nq, I->d_gbuf, qcap, I->d_offs,
+                     I->d_gbytes, I->d_ovf, I->d_bytes);
   hipLaunchKernelGGL(k_gather_copy, dim3(nq), dim3(512), 0, I->stream,
                      I->A.keys, I->A.meta, I->A.rev, I->A.vo, I->DA.keys,
                      I->DA.meta, I->DA.rev, I->DA.vo, I->heapA, I->d_rowsm,
-                     I->max_cap, I->d_found, nq, I->d_gbuf, qcap, I->d_offs,
-                     I->d_ovf);
+                     I->max_cap, I->d_found, I->d_qs, nq, I->d_gbuf, qcap,
+                     I->d_offs, I->d_ovf);
   HIP_CHECK(hipEventRecord(I->ev2, I->stream));
   return true;
 }
@@ -1386,11 +1441,36 @@ bool Slab::RangeBatchFinish(int nq, bool d2h, bool parse,
     perf.winners += found[q];
   }
   if (!d2h) return true;
-  // pack + one D2H + parse
   std::vector<int64_t> goffs(nq + 1);
   int64_t acc = 0;
   for (int q = 0; q < nq; ++q) { goffs[q] = acc; acc += ovf[q] ? 0 : gbytes[q]; }
   goffs[nq] = acc;
+  if (!parse) {
+    // raw bench mode: PIPELINED pack + D2H. The payload copy runs on the
+    // copy stream and overlaps the next batch's kernels; slot reuse (every
+    // second batch) waits on that slot's previous copy. DrainD2H() collects
+    // the tail.
+    I->pack_ms_acc = &perf.pack_d2h_ms;
+    int idx = I->pk_idx;
+    I->pk_idx ^= 1;
+    if (!I->ensure_pipe(idx, acc, err)) return false;
+    I->wait_slot(idx);
+    HIP_CHECK(hipMemcpyAsync(I->d_goffs, goffs.data(), (nq + 1) * 8,
+                             hipMemcpyHostToDevice, I->stream));
+    hipLaunchKernelGGL(k_pack, dim3(nq), dim3(256), 0, I->stream, I->d_gbuf,
+                       qcap, I->d_gbytes, I->d_goffs, I->d_ovf, I->d_packs[idx],
+                       nq);
+    HIP_CHECK(hipEventRecord(I->ev_pk[idx], I->stream));
+    HIP_CHECK(hipStreamWaitEvent(I->cstream, I->ev_pk[idx], 0));
+    if (acc > 0)
+      HIP_CHECK(hipMemcpyAsync(I->h_packs[idx], I->d_packs[idx], acc,
+                               hipMemcpyDeviceToHost, I->cstream));
+    HIP_CHECK(hipEventRecord(I->ev_cp[idx], I->cstream));
+    I->cp_busy[idx] = true;
+    return true;
+  }
+  // parse mode (List/Stream): synchronous pack + one D2H + parse
+  I->wait_slot(0);  // d_pack aliases pipeline slot 0
   HIP_CHECK(hipEventRecord(I->ev0, I->stream));
   HIP_CHECK(hipMemcpyAsync(I->d_goffs, goffs.data(), (nq + 1) * 8,
                            hipMemcpyHostToDevice, I->stream));
@@ -1404,7 +1484,6 @@ bool Slab::RangeBatchFinish(int nq, bool d2h, bool parse,
   HIP_CHECK(hipStreamSynchronize(I->stream));
   (void)hipEventElapsedTime(&ms, I->ev0, I->ev1);
   perf.pack_d2h_ms += ms;
-  if (!parse) return true;  // raw mode: bytes landed in pinned host memory
   for (int q = 0; q < nq; ++q) {
     RangeResult& r = (*outs)[q];
     if (r.overflow) continue;
@@ -1424,6 +1503,13 @@ bool Slab::RangeBatchFinish(int nq, bool d2h, bool parse,
       pp += 16 + ((klen + 15) & ~15u) + ((vlen + 15) & ~15u);
     }
   }
+  return true;
+}
+
+bool Slab::DrainD2H(std::string* err) {
+  p->pack_ms_acc = &perf.pack_d2h_ms;
+  p->wait_slot(0);
+  p->wait_slot(1);
   return true;
 }
 

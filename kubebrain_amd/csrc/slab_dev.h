@@ -48,7 +48,10 @@ struct DevRangeQ {
   uint64_t start_rev;
   int64_t cap;        // winner-write cap per query (limit+1); <=0 => unbounded
   int32_t count_only;
-  int32_t _pad;
+  // etcd3 RangeRequest.KeysOnly semantics (an extension: the reference's etcd
+  // shim ignores the flag, kv.go:48-67): gather key + mod-revision per winner,
+  // no value bytes (header vlen = 0)
+  int32_t keys_only;
 };
 struct DevGetQ {
   uint8_t key[KEYW];
@@ -150,10 +153,14 @@ class Slab {
   bool RangeBatchEx(const std::vector<DevRangeQ>& qs, bool d2h, bool parse,
                     std::vector<RangeResult>* outs, std::string* err);
   // async split: Start launches the scan+gather without syncing, so host
-  // work (e.g. the txn leg) overlaps the in-flight kernels; Finish collects
+  // work (e.g. the txn leg) overlaps the in-flight kernels; Finish collects.
+  // d2h with parse=false runs PIPELINED: the payload copy lands in pinned
+  // host memory on a copy stream, overlapping the next batch's kernels —
+  // call DrainD2H() before reading wall-clock results.
   bool RangeBatchStart(const std::vector<DevRangeQ>& qs, std::string* err);
   bool RangeBatchFinish(int nq, bool d2h, bool parse,
                         std::vector<RangeResult>* outs, std::string* err);
+  bool DrainD2H(std::string* err);
 
   // batched MVCC point read (range.go:91-121 reverse-iter semantics)
   bool GetBatch(const std::vector<DevGetQ>& qs, std::vector<GetResult>* outs,

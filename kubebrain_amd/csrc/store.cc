@@ -976,35 +976,43 @@ bool Store::BulkCreate(const uint8_t* keys, const uint32_t* klens,
   return true;
 }
 
-bool Store::BenchRange(const uint8_t* qbuf, size_t nq, bool d2h,
+// packed bench queries {u32 slen; u32 elen; u64 rev; u64 limit; start; end}
+// straight into device query structs (no per-query allocations)
+static void parseBenchQueries(const uint8_t* qbuf, size_t nq, uint64_t cur_rev,
+                              int mode, std::vector<DevRangeQ>* qall,
+                              std::vector<int64_t>* limits) {
+  qall->assign(nq, DevRangeQ{});
+  limits->resize(nq);
+  const int keys_only = (mode & 2) ? 1 : 0;
+  const uint8_t* p = qbuf;
+  for (size_t i = 0; i < nq; ++i) {
+    uint32_t slen, elen;
+    uint64_t rev, limit;
+    memcpy(&slen, p, 4); p += 4;
+    memcpy(&elen, p, 4); p += 4;
+    memcpy(&rev, p, 8); p += 8;
+    memcpy(&limit, p, 8); p += 8;
+    DevRangeQ& q = (*qall)[i];
+    memcpy(q.start, p, std::min((size_t)slen, (size_t)KEYW)); p += slen;
+    memcpy(q.end, p, std::min((size_t)elen, (size_t)KEYW)); p += elen;
+    q.read_rev = rev == 0 ? cur_rev : rev;
+    q.start_rev = 0;
+    q.cap = (int64_t)limit > 0 ? (int64_t)limit + 1 : 0;
+    q.count_only = 0;
+    q.keys_only = keys_only;
+    (*limits)[i] = (int64_t)limit;
+  }
+}
+
+bool Store::BenchRange(const uint8_t* qbuf, size_t nq, int mode,
                        unsigned long long* total, double* secs, std::string* err) {
   // the measured hot path: batched List semantics with inputs resident in HBM
   std::lock_guard<std::recursive_mutex> lk(mu_);
   if (!syncReads(err)) return false;
-  // parse straight into device query structs (no per-query allocations)
-  std::vector<DevRangeQ> qall(nq);
-  std::vector<int64_t> limits(nq);
-  {
-    const uint8_t* p = qbuf;
-    for (size_t i = 0; i < nq; ++i) {
-      uint32_t slen, elen;
-      uint64_t rev, limit;
-      memcpy(&slen, p, 4); p += 4;
-      memcpy(&elen, p, 4); p += 4;
-      memcpy(&rev, p, 8); p += 8;
-      memcpy(&limit, p, 8); p += 8;
-      DevRangeQ& q = qall[i];
-      memset(q.start, 0, KEYW);
-      memset(q.end, 0, KEYW);
-      memcpy(q.start, p, std::min((size_t)slen, (size_t)KEYW)); p += slen;
-      memcpy(q.end, p, std::min((size_t)elen, (size_t)KEYW)); p += elen;
-      q.read_rev = rev == 0 ? committed_ : rev;
-      q.cap = (int64_t)limit > 0 ? (int64_t)limit + 1 : 0;
-      q.count_only = 0;
-      q._pad = 0;
-      limits[i] = (int64_t)limit;
-    }
-  }
+  std::vector<DevRangeQ> qall;
+  std::vector<int64_t> limits;
+  parseBenchQueries(qbuf, nq, committed_, mode, &qall, &limits);
+  const bool d2h = (mode & 1) != 0;
   const int64_t kMax = 1024;
   unsigned long long tot = 0;
   auto t0 = std::chrono::steady_clock::now();
@@ -1025,35 +1033,21 @@ bool Store::BenchRange(const uint8_t* qbuf, size_t nq, bool d2h,
   return true;
 }
 
+bool Store::Sync(std::string* err) {
+  std::lock_guard<std::recursive_mutex> lk(mu_);
+  return slab_->DrainD2H(err);
+}
+
 bool Store::BenchStep(const uint8_t* qbuf, size_t nq, const uint8_t* tbuf,
-                      size_t ntx, bool d2h, uint64_t* out_revs,
+                      size_t ntx, int mode, uint64_t* out_revs,
                       unsigned long long* total, double* secs,
                       std::string* err) {
   std::lock_guard<std::recursive_mutex> lk(mu_);
   if (!syncReads(err)) return false;
-  std::vector<DevRangeQ> qall(nq);
-  std::vector<int64_t> limits(nq);
-  {
-    const uint8_t* p = qbuf;
-    for (size_t i = 0; i < nq; ++i) {
-      uint32_t slen, elen;
-      uint64_t rev, limit;
-      memcpy(&slen, p, 4); p += 4;
-      memcpy(&elen, p, 4); p += 4;
-      memcpy(&rev, p, 8); p += 8;
-      memcpy(&limit, p, 8); p += 8;
-      DevRangeQ& q = qall[i];
-      memset(q.start, 0, KEYW);
-      memset(q.end, 0, KEYW);
-      memcpy(q.start, p, std::min((size_t)slen, (size_t)KEYW)); p += slen;
-      memcpy(q.end, p, std::min((size_t)elen, (size_t)KEYW)); p += elen;
-      q.read_rev = rev == 0 ? committed_ : rev;
-      q.cap = (int64_t)limit > 0 ? (int64_t)limit + 1 : 0;
-      q.count_only = 0;
-      q._pad = 0;
-      limits[i] = (int64_t)limit;
-    }
-  }
+  std::vector<DevRangeQ> qall;
+  std::vector<int64_t> limits;
+  parseBenchQueries(qbuf, nq, committed_, mode, &qall, &limits);
+  const bool d2h = (mode & 1) != 0;
   auto t0 = std::chrono::steady_clock::now();
   if (!slab_->RangeBatchStart(qall, err)) return false;
   // overlap: the kernels read device state snapshotted at launch; the txn

@@ -107,7 +107,9 @@ class Store {
   // bench support
   bool BulkCreate(const uint8_t* keys, const uint32_t* klens, const uint8_t* vals,
                   const uint32_t* vlens, size_t n, std::string* err);
-  bool BenchRange(const uint8_t* qbuf, size_t nq, bool d2h,
+  // mode bits: 1 = d2h (pipelined payload copy to pinned host memory),
+  // 2 = keys_only (etcd3 KeysOnly semantics: no value bytes gathered)
+  bool BenchRange(const uint8_t* qbuf, size_t nq, int mode,
                   unsigned long long* total, double* secs, std::string* err);
   // batched conditional updates (txn.go:249-265 per op); out_revs[i] = new
   // revision on success, 0 on CAS failure
@@ -118,8 +120,10 @@ class Store {
   // one bench step: launch the range batch async, run the txn batch on the
   // host while the kernels are in flight, then collect
   bool BenchStep(const uint8_t* qbuf, size_t nq, const uint8_t* tbuf,
-                 size_t ntx, bool d2h, uint64_t* out_revs,
+                 size_t ntx, int mode, uint64_t* out_revs,
                  unsigned long long* total, double* secs, std::string* err);
+  // wait for in-flight pipelined D2H copies (call before reading wall-clock)
+  bool Sync(std::string* err);
   std::string PerfJson();
   void PerfReset();
 

@@ -407,10 +407,16 @@ int Backend::scanRange(const Bytes& start, const Bytes& end, uint64_t revision,
                        int64_t limit, bool compact, uint64_t timeoutRevision,
                        std::vector<KeyValue>* out) {
   // worker.run (scanner.go:389-516). Snapshot-by-copy iteration like memkv
-  // (memkv/iter.go:53-85): deletes during compaction mutate the live map.
+  // (memkv/iter.go:53-85) is needed only when compact=true (deletes mutate
+  // the live map). The read path iterates live with the same early exit the
+  // reference's worker has (receiver.needMore, scanner.go:416) — snapshotting
+  // the whole range first would overstate the CPU cost of limited scans over
+  // large namespaces (the timed cpu_baseline runs through here).
   std::vector<std::pair<Bytes, Bytes>> snapshot;
-  for (auto it = store_.lower_bound(start); it != store_.end() && it->first < end; ++it)
-    snapshot.emplace_back(it->first, it->second);
+  if (compact) {
+    for (auto it = store_.lower_bound(start); it != store_.end() && it->first < end; ++it)
+      snapshot.emplace_back(it->first, it->second);
+  }
 
   int count = 0;
   Bytes prevUserKey, prevValue;
@@ -420,26 +426,25 @@ int Backend::scanRange(const Bytes& start, const Bytes& end, uint64_t revision,
   auto needMore = [&]() { return !(limit > 0 && out && (int64_t)out->size() >= limit); };
 
   bool stopped = false;
-  for (auto& kvp : snapshot) {
-    if (!needMore()) { stopped = true; break; }  // receiver.needMore (scanner.go:416)
-    const Bytes& ikey = kvp.first;
-    const Bytes& value = kvp.second;
+  // returns false to stop the scan (receiver.needMore, scanner.go:416)
+  auto process = [&](const Bytes& ikey, const Bytes& value) -> bool {
+    if (!needMore()) { stopped = true; return false; }
     Bytes curUserKey; uint64_t curRevision;
-    if (DecodeInternalKey(ikey, &curUserKey, &curRevision) != OK) continue;  // scanner.go:435-439
+    if (DecodeInternalKey(ikey, &curUserKey, &curRevision) != OK) return true;  // scanner.go:435-439
 
     // compactIfExpired (scanner.go:444-447 -> 566-591): BEFORE the rev skip
     if (compact && timeoutRevision != 0 &&
         curUserKey.find(kEvents) != Bytes::npos) {
       if (curRevision == 0) {
         uint64_t rv = BytesToU64(value.substr(0, 8));
-        if (rv <= timeoutRevision) { store_.erase(ikey); continue; }
+        if (rv <= timeoutRevision) { store_.erase(ikey); return true; }
       } else if (curRevision <= timeoutRevision) {
         store_.erase(ikey);
-        continue;
+        return true;
       }
     }
 
-    if (curRevision > revision) continue;  // scanner.go:451-453
+    if (curRevision > revision) return true;  // scanner.go:451-453
 
     if (!havePrev || curUserKey != prevUserKey) {  // scanner.go:457-462
       if (prevRevision > 0 && prevValue != kTombstone) {
@@ -456,14 +461,23 @@ int Backend::scanRange(const Bytes& start, const Bytes& end, uint64_t revision,
     // delete 9B-flagged revision rows (scanner.go:477-491)
     if (compact && curRevision == 0 && value.size() == 9) {
       uint64_t objRev = BytesToU64(value.substr(0, 8));
-      if (objRev > revision) continue;  // skip gc AND the prev update
-      store_.erase(ikey);               // DelCurrent
+      if (objRev > revision) return true;  // skip gc AND the prev update
+      store_.erase(ikey);                  // DelCurrent
     }
 
     prevRevision = curRevision;
     prevUserKey = curUserKey;
     prevValue = value;
     havePrev = true;
+    return true;
+  };
+  if (compact) {
+    for (auto& kvp : snapshot)
+      if (!process(kvp.first, kvp.second)) break;
+  } else {
+    for (auto it = store_.lower_bound(start);
+         it != store_.end() && it->first < end; ++it)
+      if (!process(it->first, it->second)) break;
   }
 
   // tail emit (scanner.go:503-509)

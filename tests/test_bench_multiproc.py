@@ -85,6 +85,9 @@ class MockStore:
     def f_flush(self, h):
         return 0
 
+    def f_sync(self, h):
+        return 0
+
     def f_perf_reset(self, h):
         return 0
 
