@@ -178,83 +178,86 @@ __device__ int64_t d_lb_fenced(const uint8_t* keys, const uint64_t* rev,
 // winner — base winners are suppressed by a delta probe, and the two ordered
 // winner lists are merged by rank (keys never collide across lists).
 
-// one run's winner scan (ordered append); returns written, *total = seen
+// all 4 bounds of every query in one massively-parallel launch (one thread
+// per bound): the 4 dependent binary-search chains per query that used to
+// serialize inside the scan block run concurrently across the whole batch,
+// taking ~one search-chain latency total instead of per query.
+__global__ void k_range_bounds(const uint8_t* __restrict__ bkeys,
+                               const uint64_t* __restrict__ brev, int64_t n,
+                               const uint8_t* __restrict__ dkeys,
+                               const uint64_t* __restrict__ drev, int64_t dn,
+                               const uint8_t* __restrict__ fkeys,
+                               const uint64_t* __restrict__ frev, int64_t nf,
+                               const DevRangeQ* __restrict__ qs, int nq,
+                               int64_t* __restrict__ qb /* [4*nq] */) {
+  int i = blockIdx.x * blockDim.x + threadIdx.x;
+  if (i >= 4 * nq) return;
+  int q = i >> 2, which = i & 3;
+  const DevRangeQ& Q = qs[q];
+  int64_t r;
+  switch (which) {
+    case 0: r = d_lb_fenced(bkeys, brev, n, fkeys, frev, nf, Q.start, Q.start_rev); break;
+    case 1: r = d_lb_fenced(bkeys, brev, n, fkeys, frev, nf, Q.end, 0); break;
+    case 2: r = d_lb_range(dkeys, drev, 0, dn, Q.start, Q.start_rev); break;
+    default: r = d_lb_range(dkeys, drev, 0, dn, Q.end, 0); break;
+  }
+  qb[i] = r;
+}
+
+constexpr int SCAN_T = 1024;  // threads per scan block (16 waves)
+
+// one run's winner scan (ordered append); returns written, *total = seen.
+// One row per thread per round (fully coalesced 8B streams), winner ranking
+// by wave ballot + popcount, cross-wave offsets via one small LDS round; the
+// running winner count is a uniform register, not LDS.
 __device__ int64_t scan_run_winners(
     const uint8_t* __restrict__ keys, const uint64_t* __restrict__ meta,
     const uint64_t* __restrict__ rev, int64_t lo, int64_t hi, uint64_t R,
     int64_t cap, uint64_t* out, int64_t out_cap, uint64_t tagbit,
     const uint8_t* __restrict__ skeys, const uint64_t* __restrict__ srev,
     int64_t slo, int64_t shi,  // suppression run (null => none)
-    int64_t* total_out, int64_t* scanned_accum, int64_t* cnt_s, int* wave_cnt) {
-  int lane = threadIdx.x & 63, w = threadIdx.x >> 6;
-  if (threadIdx.x == 0) *cnt_s = 0;
-  __syncthreads();
+    int64_t* total_out, int64_t* scanned_accum, int* wave_cnt) {
+  const int lane = threadIdx.x & 63, w = threadIdx.x >> 6;
+  const int NW = blockDim.x >> 6;
   int64_t scanned = 0;
-  // 4 consecutive rows per thread: 1024-row tiles quarter the number of
-  // latency-bound sync rounds per query vs 256-row tiles
-  const int64_t TILE = (int64_t)blockDim.x * 4;
-  for (int64_t t = lo; t < hi; t += TILE) {
-    int64_t i0 = t + (int64_t)threadIdx.x * 4;
-    uint32_t wins = 0, mycnt = 0;
-    if (i0 < hi) {
-      int nk = (int)min((int64_t)4, hi - i0);
-      uint64_t r[5], m[4];
-#pragma unroll
-      for (int k = 0; k < 4; ++k) {
-        r[k] = k < nk ? rev[i0 + k] : 0;
-        m[k] = k < nk ? meta[i0 + k] : 0;
-      }
-      // rev[i0+4] only needed when row i0+3's same_next is set (then in-bounds)
-      r[4] = (nk == 4 && (m[3] & M_SAME_NEXT)) ? rev[i0 + 4] : 0;
-#pragma unroll
-      for (int k = 0; k < 4; ++k) {
-        if (k >= nk) break;
-        bool win = false;
-        if (r[k] > 0 && r[k] <= R && !(m[k] & M_TOMB))
-          win = !(m[k] & M_SAME_NEXT) || r[k + 1] > R;
-        if (win && skeys) {
-          const uint8_t* kk = keys + (i0 + k) * KEYW;
-          int64_t lb = d_lb_range(skeys, srev, slo, shi, kk, 1);
-          if (lb < shi && srev[lb] <= R && keycmp96(skeys + lb * KEYW, kk) == 0)
-            win = false;  // a newer (delta) row of this key wins instead
-        }
-        if (win) { wins |= 1u << k; mycnt++; }
+  int64_t cnt = 0;  // uniform across the block
+  const int64_t T = blockDim.x;
+  for (int64_t t = lo; t < hi; t += T) {
+    int64_t i = t + threadIdx.x;
+    bool win = false;
+    if (i < hi) {
+      uint64_t r = rev[i], m = meta[i];
+      if (r > 0 && r <= R && !(m & M_TOMB))
+        // same_next set => row i+1 exists and shares the key
+        win = !(m & M_SAME_NEXT) || rev[i + 1] > R;
+      if (win && skeys) {
+        const uint8_t* kk = keys + i * KEYW;
+        int64_t lb = d_lb_range(skeys, srev, slo, shi, kk, 1);
+        if (lb < shi && srev[lb] <= R && keycmp96(skeys + lb * KEYW, kk) == 0)
+          win = false;  // a newer (delta) row of this key wins instead
       }
     }
-    // wave-inclusive scan of per-thread winner counts (ordered append)
-    uint32_t incl = mycnt;
-#pragma unroll
-    for (int o = 1; o < 64; o <<= 1) {
-      uint32_t v = __shfl_up(incl, o);
-      if (lane >= o) incl += v;
+    uint64_t b = __ballot(win);
+    uint32_t myrank = __popcll(b & ((1ull << lane) - 1));
+    if (lane == 0) wave_cnt[w] = (int)__popcll(b);
+    __syncthreads();
+    int64_t waveoff = 0, tile_total = 0;
+    for (int k = 0; k < NW; ++k) {
+      if (k < w) waveoff += wave_cnt[k];
+      tile_total += wave_cnt[k];
     }
-    uint32_t excl = incl - mycnt;
-    if (lane == 63) wave_cnt[w] = (int)incl;
-    __syncthreads();
-    int64_t waveoff = 0;
-    for (int k = 0; k < w; ++k) waveoff += wave_cnt[k];
-    int tile_total = wave_cnt[0] + wave_cnt[1] + wave_cnt[2] + wave_cnt[3];
-    if (out && wins) {
-      int64_t idx = *cnt_s + waveoff + excl;
-#pragma unroll
-      for (int k = 0; k < 4; ++k) {
-        if (wins & (1u << k)) {
-          if (idx < cap && idx < out_cap) out[idx] = (uint64_t)(i0 + k) | tagbit;
-          idx++;
-        }
-      }
+    if (out && win) {
+      int64_t idx = cnt + waveoff + myrank;
+      if (idx < cap && idx < out_cap) out[idx] = (uint64_t)i | tagbit;
     }
-    __syncthreads();
-    if (threadIdx.x == 0) *cnt_s += tile_total;
-    scanned += min(TILE, hi - t);
-    __syncthreads();
-    if (*cnt_s >= cap) break;
+    cnt += tile_total;
+    scanned += min(T, hi - t);
+    if (cnt >= cap) break;
+    __syncthreads();  // wave_cnt reused next round
   }
-  __syncthreads();
-  int64_t tot = *cnt_s;
-  *total_out = tot;
+  *total_out = cnt;
   if (scanned_accum) *scanned_accum += scanned;
-  int64_t written = tot < cap ? tot : cap;
+  int64_t written = cnt < cap ? cnt : cap;
   if (written > out_cap) written = out_cap;
   return out ? written : 0;
 }
@@ -279,21 +282,17 @@ __global__ void k_range_scan2(
     const uint64_t* __restrict__ brev, int64_t n,
     const uint8_t* __restrict__ dkeys, const uint64_t* __restrict__ dmeta,
     const uint64_t* __restrict__ drev, int64_t dn,
-    const uint8_t* __restrict__ fkeys, const uint64_t* __restrict__ frev,
-    int64_t nf, const DevRangeQ* __restrict__ qs, int nq, int64_t max_cap,
+    const int64_t* __restrict__ qb /* bounds from k_range_bounds */,
+    const DevRangeQ* __restrict__ qs, int nq, int64_t max_cap,
     uint64_t* __restrict__ rows_b, uint64_t* __restrict__ rows_d,
     uint64_t* __restrict__ rows_m, int64_t* __restrict__ found_out,
     int64_t* __restrict__ total_out, unsigned long long* __restrict__ scanned_out) {
   int q = blockIdx.x;
   if (q >= nq) return;
-  __shared__ int64_t lo_s, hi_s, dlo_s, dhi_s, cnt_s;
-  __shared__ int wave_cnt[4];
+  __shared__ int wave_cnt[SCAN_T / 64];
   const DevRangeQ& Q = qs[q];
-  if (threadIdx.x == 0) lo_s = d_lb_fenced(bkeys, brev, n, fkeys, frev, nf, Q.start, Q.start_rev);
-  if (threadIdx.x == 64) hi_s = d_lb_fenced(bkeys, brev, n, fkeys, frev, nf, Q.end, 0);
-  if (threadIdx.x == 128) dlo_s = d_lb_range(dkeys, drev, 0, dn, Q.start, Q.start_rev);
-  if (threadIdx.x == 192) dhi_s = d_lb_range(dkeys, drev, 0, dn, Q.end, 0);
-  __syncthreads();
+  const int64_t lo_s = qb[4 * q], hi_s = qb[4 * q + 1];
+  const int64_t dlo_s = qb[4 * q + 2], dhi_s = qb[4 * q + 3];
   const int64_t cap = Q.cap > 0 ? Q.cap : INT64_MAX;
   int64_t scanned = 0;
   int64_t dtotal = 0, btotal = 0;
@@ -301,10 +300,12 @@ __global__ void k_range_scan2(
   uint64_t* outb = Q.count_only ? nullptr : rows_b + (int64_t)q * max_cap;
   int64_t nB = scan_run_winners(dkeys, dmeta, drev, dlo_s, dhi_s, Q.read_rev,
                                 cap, outd, max_cap, ROW_TAG_DELTA, nullptr,
-                                nullptr, 0, 0, &dtotal, &scanned, &cnt_s, wave_cnt);
+                                nullptr, 0, 0, &dtotal, &scanned, wave_cnt);
+  __syncthreads();  // wave_cnt handoff between the two runs
   int64_t nA = scan_run_winners(bkeys, bmeta, brev, lo_s, hi_s, Q.read_rev, cap,
                                 outb, max_cap, 0, dn ? dkeys : nullptr, drev,
-                                dlo_s, dhi_s, &btotal, &scanned, &cnt_s, wave_cnt);
+                                dlo_s, dhi_s, &btotal, &scanned, wave_cnt);
+  __syncthreads();  // winner lists complete before the merge reads them
   // merge by rank into rows_m (keys are disjoint across the two lists)
   int64_t cap_m = nA + nB;
   if (cap_m > cap) cap_m = cap;
@@ -890,6 +891,7 @@ struct Slab::Impl {
   uint8_t* d_pack = nullptr;    // arena
   int64_t* d_goffs = nullptr;   // max_q+1
   int64_t* d_bounds = nullptr;  // compact bounds
+  int64_t* d_qb = nullptr;      // range query bounds [4*max_q]
   uint8_t* d_bkeys = nullptr;
   uint64_t* d_brevs = nullptr;
 
@@ -987,7 +989,7 @@ struct Slab::Impl {
                     (void*)d_total, (void*)d_gbytes, (void*)d_ovf,
                     (void*)d_found32, (void*)d_orev, (void*)d_ometa,
                     (void*)d_scanned, (void*)d_bytes, (void*)d_gbuf,
-                    (void*)d_pack, (void*)d_goffs, (void*)d_bounds,
+                    (void*)d_pack, (void*)d_goffs, (void*)d_bounds, (void*)d_qb,
                     (void*)d_bkeys, (void*)d_brevs, (void*)d_dkeys,
                     (void*)d_dmeta, (void*)d_drev, (void*)d_dvo, (void*)d_wpfx,
                     (void*)d_wplen, (void*)d_wlive, (void*)d_wfrom,
@@ -1205,6 +1207,7 @@ Slab* Slab::Create(int64_t max_rows, int64_t heap_cap, int device,
   HIP_CHECK_NULL(hipMalloc(&I->d_pack, I->arena_bytes));
   HIP_CHECK_NULL(hipMalloc(&I->d_goffs, (I->max_q + 1) * 8));
   HIP_CHECK_NULL(hipMalloc(&I->d_bounds, 256 * 8));
+  HIP_CHECK_NULL(hipMalloc(&I->d_qb, (int64_t)I->max_q * 4 * 8));
   HIP_CHECK_NULL(hipMalloc(&I->d_bkeys, 256 * KEYW));
   HIP_CHECK_NULL(hipMalloc(&I->d_brevs, 256 * 8));
   (void)fail;
@@ -1368,12 +1371,15 @@ bool Slab::RangeBatchStart(const std::vector<DevRangeQ>& qs, std::string* err) {
   HIP_CHECK(hipMemsetAsync(I->d_scanned, 0, 8, I->stream));
   HIP_CHECK(hipMemsetAsync(I->d_bytes, 0, 8, I->stream));
   HIP_CHECK(hipEventRecord(I->ev0, I->stream));
-  hipLaunchKernelGGL(k_range_scan2, dim3(nq), dim3(256), 0, I->stream,
+  hipLaunchKernelGGL(k_range_bounds, dim3((uint32_t)ceil_div(4 * nq, 256)),
+                     dim3(256), 0, I->stream, I->A.keys, I->A.rev, I->n,
+                     I->DA.keys, I->DA.rev, I->dn, I->d_fkeys, I->d_frev,
+                     I->nf, I->d_qs, nq, I->d_qb);
+  hipLaunchKernelGGL(k_range_scan2, dim3(nq), dim3(SCAN_T), 0, I->stream,
                      I->A.keys, I->A.meta, I->A.rev, I->n, I->DA.keys,
-                     I->DA.meta, I->DA.rev, I->dn, I->d_fkeys, I->d_frev,
-                     I->nf, I->d_qs, nq, I->max_cap,
-                     I->d_rows, I->d_rows2, I->d_rowsm, I->d_found, I->d_total,
-                     I->d_scanned);
+                     I->DA.meta, I->DA.rev, I->dn, I->d_qb, I->d_qs, nq,
+                     I->max_cap, I->d_rows, I->d_rows2, I->d_rowsm, I->d_found,
+                     I->d_total, I->d_scanned);
   HIP_CHECK(hipEventRecord(I->ev1, I->stream));
   hipLaunchKernelGGL(k_gather, dim3(nq), dim3(256), 0, I->stream, I->A.keys,
                      I->A.meta, I->A.rev, I->A.vo, I->DA.keys, I->DA.meta,
