@@ -200,8 +200,76 @@ def run_txns(store, live_keys, revs, rng, n):
 
 
 def cross_shard_leg_wrapper(store, dist, pg, rank, world, all_namespaces):
-    out = cross_shard_leg(store, dist, pg, rank, world, all_namespaces)
-    return out
+    # the product path: kb_comm_init + kb_range_global (RCCL over xGMI inside
+    # the C-ABI, comm.cc). Falls back to the torch.distributed emulation only
+    # if the C-ABI path cannot run (e.g. the CPU gloo rehearsal's mock store).
+    try:
+        return cross_shard_cabi(store, dist, rank, world, all_namespaces)
+    except Exception as e:  # noqa: BLE001
+        out = cross_shard_leg(store, dist, pg, rank, world, all_namespaces)
+        out["cabi_fallback_reason"] = repr(e)[:200]
+        return out
+
+
+def cross_shard_cabi(store, dist, rank, world, all_namespaces, n_queries=64,
+                     limit=LIMIT):
+    """configs[3] through the product C-ABI: every rank scans its shard and
+    kb_range_global runs the RCCL allgather(counts)+allgather(payload) over
+    xGMI plus the k-way merge with the global limit cut (comm.cc; semantics
+    scanner.go:269-300). The ncclUniqueId bootstraps out-of-band over the
+    gloo group."""
+    import ctypes as C
+
+    import torch
+
+    from kubebrain_amd.client import _parse_kvs
+
+    lib = store.lib
+    idb = C.create_string_buffer(256)
+    idlen = C.c_size_t()
+    if rank == 0:
+        rc = lib.kb_comm_id(idb, C.c_size_t(256), C.byref(idlen))
+        assert rc == 0, f"kb_comm_id rc={rc}"
+        idt = torch.frombuffer(bytearray(idb.raw[:128]), dtype=torch.uint8).clone()
+    else:
+        idt = torch.zeros(128, dtype=torch.uint8)
+    dist.broadcast(idt, src=0)  # gloo (CPU) bootstrap channel
+    id_bytes = bytes(idt.numpy().tobytes())
+    rc = lib.kb_comm_init(C.c_void_p(store.h), id_bytes, C.c_size_t(128),
+                          C.c_int(rank), C.c_int(world))
+    assert rc == 0, f"kb_comm_init rc={rc}"
+    try:
+        BUF = 8 << 20
+        out = C.create_string_buffer(BUF)
+        merged_counts = []
+        t0 = time.time()
+        for qi in range(n_queries):
+            lo = b"/registry/pods/ns-%04d" % (qi % max(len(all_namespaces) // 2, 1))
+            hi = b"/registry/pods0"
+            out_len = C.c_size_t()
+            hr = C.c_uint64()
+            more = C.c_int()
+            rc = lib.kb_range_global(C.c_void_p(store.h), lo, C.c_size_t(len(lo)),
+                                     hi, C.c_size_t(len(hi)), C.c_uint64(0),
+                                     C.c_longlong(limit), out, C.c_size_t(BUF),
+                                     C.byref(out_len), C.byref(hr), C.byref(more))
+            assert rc == 0, f"kb_range_global rc={rc}"
+            if rank == 0:
+                kvs = _parse_kvs(out.raw[:out_len.value])
+                assert all(kvs[i].key < kvs[i + 1].key
+                           for i in range(len(kvs) - 1)), "merge order"
+                merged_counts.append(len(kvs))
+        dt = time.time() - t0
+    finally:
+        lib.kb_comm_free(C.c_void_p(store.h))
+    return {
+        "queries": n_queries,
+        "ops_per_sec": round(n_queries / dt, 1),
+        "limit": limit,
+        "transport": "rccl-cabi (kb_range_global over xGMI)",
+        "merged_counts_min_max": [min(merged_counts), max(merged_counts)]
+        if merged_counts else None,
+    }
 
 
 def cross_shard_leg(store, dist, pg, rank, world, all_namespaces, n_queries=64,

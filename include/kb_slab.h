@@ -118,6 +118,35 @@ int kb_partitions(kb_store*, const uint8_t* start, size_t slen,
                   const uint8_t* end, size_t elen, uint8_t* out, size_t cap,
                   size_t* out_len, uint64_t* header_rev);
 
+/* ---- cross-shard exchange (RCCL over xGMI) ----
+ * The key slab shards by namespace hash across the GPUs of one node, one
+ * store per GPU. A Range that spans shards mirrors the reference's
+ * multi-partition fork/merge (pkg/backend/scanner/scanner.go:269-300): each
+ * shard scans locally, ONE collective exchanges the sorted winner runs
+ * (allgather of counts, then payload padded to the max — NCCL/RCCL has no
+ * allgatherv), and every rank k-way-merges with the global limit+1 cut.
+ * The ncclUniqueId travels out-of-band on the caller's bootstrap channel
+ * (cgo host RPC, torch.distributed gloo, ...). */
+/* rank 0 generates the 128-byte ncclUniqueId */
+int kb_comm_id(uint8_t* out, size_t cap, size_t* len);
+/* collective: all ranks call with the same id; one store per GPU */
+int kb_comm_init(kb_store*, const uint8_t* id, size_t id_len, int rank, int world);
+int kb_comm_rank(kb_store*, int* rank, int* world);
+void kb_comm_free(kb_store*);
+/* cross-shard Range: COLLECTIVE (all ranks, same arguments); out = the
+ * kb_list wire format with the globally merged result; degrades to the
+ * local List when no communicator is initialized */
+int kb_range_global(kb_store*, const uint8_t* start, size_t slen,
+                    const uint8_t* end, size_t elen, uint64_t rev,
+                    int64_t limit, uint8_t* out, size_t cap, size_t* out_len,
+                    uint64_t* header_rev, int* more);
+/* TEST-ONLY: the k-way merge + global limit cut on caller-supplied packed
+ * runs (no GPU, no communicator) — lets CPU tests pin the exchange's merge
+ * semantics; runs_cat = world runs back to back, lens their byte lengths */
+int kb_test_merge_runs(const uint8_t* runs_cat, const unsigned long long* lens,
+                       int world, long long limit, uint8_t* out, size_t cap,
+                       size_t* out_len, int* more);
+
 /* ---- revision (tso/tso.go:41-76) ---- */
 unsigned long long kb_current_rev(kb_store*);
 void kb_set_current_rev(kb_store*, unsigned long long rev); /* leader TSO init */

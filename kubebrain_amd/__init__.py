@@ -20,7 +20,7 @@ def build(force: bool = False) -> str:
 
     src_dir = os.path.join(_HERE, "csrc")
     srcs = [os.path.join(src_dir, f) for f in ("slab.hip", "store.cc", "cabi.cc",
-                                               "slab_dev.h", "store.h")]
+                                               "comm.cc", "slab_dev.h", "store.h")]
     srcs.append(os.path.join(_HERE, "..", "include", "kb_slab.h"))
     stale = force or not os.path.exists(LIB_PATH) or any(
         os.path.getmtime(s) > os.path.getmtime(LIB_PATH) for s in srcs)

@@ -207,9 +207,12 @@ __global__ void k_range_bounds(const uint8_t* __restrict__ bkeys,
 constexpr int SCAN_T = 1024;  // threads per scan block (16 waves)
 
 // one run's winner scan (ordered append); returns written, *total = seen.
-// One row per thread per round (fully coalesced 8B streams), winner ranking
-// by wave ballot + popcount, cross-wave offsets via one small LDS round; the
-// running winner count is a uniform register, not LDS.
+// Two-pass per tile: (1) each wave reads a CONTIGUOUS chunk of rows with
+// fully-coalesced 8B meta/rev streams, recording winner flags in a register
+// bitmask -- no LDS, no syncs inside the pass; (2) after one barrier
+// publishes the per-wave totals, a register-only ballot replay scatters the
+// winner row indices at their ordered offsets. Two __syncthreads per tile
+// of up to NW*4096 rows (the old design paid two per 1024 rows).
 __device__ int64_t scan_run_winners(
     const uint8_t* __restrict__ keys, const uint64_t* __restrict__ meta,
     const uint64_t* __restrict__ rev, int64_t lo, int64_t hi, uint64_t R,
@@ -219,41 +222,70 @@ __device__ int64_t scan_run_winners(
     int64_t* total_out, int64_t* scanned_accum, int* wave_cnt) {
   const int lane = threadIdx.x & 63, w = threadIdx.x >> 6;
   const int NW = blockDim.x >> 6;
+  const int64_t span = hi - lo;
+  if (span <= 0) { *total_out = 0; return 0; }
+  // per-wave chunk: cover the span in one tile when it fits (<= NW*4096
+  // rows); for capped queries over much larger spans, tile by a winner-
+  // density heuristic so the cap early-exit still bounds the read volume
+  int64_t target = span;
+  if (cap != INT64_MAX && 6 * cap < span) target = 6 * cap;
+  int64_t C = (target + NW - 1) / NW;
+  C = (C + 63) & ~63ll;
+  if (C > 4096) C = 4096;
+  const int64_t TILE = (int64_t)NW * C;
   int64_t scanned = 0;
-  int64_t cnt = 0;  // uniform across the block
-  const int64_t T = blockDim.x;
-  for (int64_t t = lo; t < hi; t += T) {
-    int64_t i = t + threadIdx.x;
-    bool win = false;
-    if (i < hi) {
-      uint64_t r = rev[i], m = meta[i];
-      if (r > 0 && r <= R && !(m & M_TOMB))
-        // same_next set => row i+1 exists and shares the key
-        win = !(m & M_SAME_NEXT) || rev[i + 1] > R;
-      if (win && skeys) {
-        const uint8_t* kk = keys + i * KEYW;
-        int64_t lb = d_lb_range(skeys, srev, slo, shi, kk, 1);
-        if (lb < shi && srev[lb] <= R && keycmp96(skeys + lb * KEYW, kk) == 0)
-          win = false;  // a newer (delta) row of this key wins instead
+  int64_t cnt = 0;  // winners before this tile (uniform across the block)
+  for (int64_t t = lo; t < hi; t += TILE) {
+    const int64_t wbase = t + (int64_t)w * C;
+    int rounds = 0;
+    if (wbase < hi) rounds = (int)((min(C, hi - wbase) + 63) >> 6);
+    // pass 1: predicate + winner flags (register-only, no syncs)
+    uint64_t flags = 0;
+    uint32_t wcnt = 0;
+    for (int r = 0; r < rounds; ++r) {
+      int64_t i = wbase + ((int64_t)r << 6) + lane;
+      bool win = false;
+      if (i < hi) {
+        uint64_t rv = rev[i], m = meta[i];
+        if (rv > 0 && rv <= R && !(m & M_TOMB))
+          // same_next set => row i+1 exists and shares the key
+          win = !(m & M_SAME_NEXT) || rev[i + 1] > R;
+        if (win && skeys) {
+          const uint8_t* kk = keys + i * KEYW;
+          int64_t lb = d_lb_range(skeys, srev, slo, shi, kk, 1);
+          if (lb < shi && srev[lb] <= R && keycmp96(skeys + lb * KEYW, kk) == 0)
+            win = false;  // a newer (delta) row of this key wins instead
+        }
       }
+      uint64_t b = __ballot(win);
+      if (win) flags |= 1ull << r;
+      wcnt += (uint32_t)__popcll(b);
     }
-    uint64_t b = __ballot(win);
-    uint32_t myrank = __popcll(b & ((1ull << lane) - 1));
-    if (lane == 0) wave_cnt[w] = (int)__popcll(b);
+    if (lane == 0) wave_cnt[w] = (int)wcnt;
     __syncthreads();
     int64_t waveoff = 0, tile_total = 0;
     for (int k = 0; k < NW; ++k) {
       if (k < w) waveoff += wave_cnt[k];
       tile_total += wave_cnt[k];
     }
-    if (out && win) {
-      int64_t idx = cnt + waveoff + myrank;
-      if (idx < cap && idx < out_cap) out[idx] = (uint64_t)i | tagbit;
+    // pass 2: ordered scatter by ballot replay (no memory re-reads)
+    if (out && wcnt) {
+      uint32_t done = 0;
+      for (int r = 0; r < rounds; ++r) {
+        uint64_t b = __ballot((flags >> r) & 1);
+        if ((flags >> r) & 1) {
+          int64_t idx = cnt + waveoff + done +
+                        (int64_t)__popcll(b & ((1ull << lane) - 1));
+          if (idx < cap && idx < out_cap)
+            out[idx] = (uint64_t)(wbase + ((int64_t)r << 6) + lane) | tagbit;
+        }
+        done += (uint32_t)__popcll(b);
+      }
     }
     cnt += tile_total;
-    scanned += min(T, hi - t);
+    scanned += min(TILE, hi - t);
     if (cnt >= cap) break;
-    __syncthreads();  // wave_cnt reused next round
+    __syncthreads();  // wave_cnt reused next tile
   }
   *total_out = cnt;
   if (scanned_accum) *scanned_accum += scanned;
