@@ -204,7 +204,9 @@ __global__ void k_range_bounds(const uint8_t* __restrict__ bkeys,
   qb[i] = r;
 }
 
-constexpr int SCAN_T = 1024;  // threads per scan block (16 waves)
+// scan block width: runtime-tunable (KB_SCAN_T in {256,512,1024}); the
+// kernel reads blockDim.x, only the LDS wave-count array is sized for the max
+constexpr int SCAN_T_MAX = 1024;
 
 // one run's winner scan (ordered append); returns written, *total = seen.
 // Two-pass per tile: (1) each wave reads a CONTIGUOUS chunk of rows with
@@ -321,7 +323,7 @@ __global__ void k_range_scan2(
     int64_t* __restrict__ total_out, unsigned long long* __restrict__ scanned_out) {
   int q = blockIdx.x;
   if (q >= nq) return;
-  __shared__ int wave_cnt[SCAN_T / 64];
+  __shared__ int wave_cnt[SCAN_T_MAX / 64];
   const DevRangeQ& Q = qs[q];
   const int64_t lo_s = qb[4 * q], hi_s = qb[4 * q + 1];
   const int64_t dlo_s = qb[4 * q + 2], dhi_s = qb[4 * q + 3];
@@ -378,7 +380,7 @@ __global__ void k_get2(const uint8_t* __restrict__ bkeys,
                        const uint8_t* __restrict__ fkeys,
                        const uint64_t* __restrict__ frev, int64_t nf,
                        const uint8_t* __restrict__ heap,
-                       const DevGetQ* __restrict__ qs, int nq,
+                       const DevGetQ* __restrict__ qs, int nq, int copy_vals,
                        uint8_t* __restrict__ out, int64_t slot,
                        uint64_t* __restrict__ orev, uint64_t* __restrict__ ometa,
                        int32_t* __restrict__ ofound, int32_t* __restrict__ oovf) {
@@ -426,9 +428,9 @@ __global__ void k_get2(const uint8_t* __restrict__ bkeys,
     ofound[q] = 1;
     orev[q] = rev[row];
     ometa[q] = m;
-    oovf[q] = vlen > slot ? 1 : 0;
+    oovf[q] = (copy_vals && vlen > slot) ? 1 : 0;
   }
-  if (vlen > slot) return;
+  if (!copy_vals || vlen > slot) return;
   const uint8_t* vs = heap + vo[row];
   uint8_t* vd = out + (int64_t)q * slot;
   uint32_t w16 = vlen >> 4;
@@ -904,6 +906,7 @@ struct Slab::Impl {
 
   // range/get scratch
   int max_q = 1024;
+  int scan_t = 256;             // KB_SCAN_T: threads per scan block
   int64_t max_cap = 4352;       // winners per query cap (>= limit+1 for etcd's 500)
   int64_t arena_bytes = 384ll << 20;
   DevRangeQ* d_qs = nullptr;
@@ -945,6 +948,9 @@ struct Slab::Impl {
   int64_t bitmap_cap = 0;
 
   hipEvent_t ev0 = nullptr, ev1 = nullptr, ev2 = nullptr, ev3 = nullptr;
+  hipEvent_t ev_g0 = nullptr, ev_g1 = nullptr;  // async get lookup
+  uint64_t* h_gmeta = nullptr;   // pinned [2*max_q]: orev | ometa
+  int32_t* h_gfound = nullptr;   // pinned [max_q]
 
   // host staging (pinned for fast D2H)
   uint8_t* h_pack = nullptr;
@@ -1029,6 +1035,10 @@ struct Slab::Impl {
       if (p) (void)hipFree(p);
     }
     if (h_pack) (void)hipHostFree(h_pack);
+    if (h_gmeta) (void)hipHostFree(h_gmeta);
+    if (h_gfound) (void)hipHostFree(h_gfound);
+    if (ev_g0) (void)hipEventDestroy(ev_g0);
+    if (ev_g1) (void)hipEventDestroy(ev_g1);
     if (ev0) (void)hipEventDestroy(ev0);
     if (ev1) (void)hipEventDestroy(ev1);
     if (ev2) (void)hipEventDestroy(ev2);
@@ -1180,6 +1190,10 @@ Slab* Slab::Create(int64_t max_rows, int64_t heap_cap, int device,
   I->max_rows = max_rows;
   I->heap_cap = heap_cap;
   I->max_q = (int)env_i64("KB_MAX_Q", 1024);
+  I->scan_t = (int)env_i64("KB_SCAN_T", 256);
+  if (I->scan_t < 64) I->scan_t = 64;
+  if (I->scan_t > 1024) I->scan_t = 1024;
+  I->scan_t &= ~63;
   I->max_cap = env_i64("KB_MAX_CAP", 4352);
   I->arena_bytes = env_i64("KB_ARENA_BYTES", 384ll << 20);
   std::string lerr;
@@ -1190,6 +1204,10 @@ Slab* Slab::Create(int64_t max_rows, int64_t heap_cap, int device,
   HIP_CHECK_NULL(hipEventCreate(&I->ev1));
   HIP_CHECK_NULL(hipEventCreate(&I->ev2));
   HIP_CHECK_NULL(hipEventCreate(&I->ev3));
+  HIP_CHECK_NULL(hipEventCreate(&I->ev_g0));
+  HIP_CHECK_NULL(hipEventCreate(&I->ev_g1));
+  HIP_CHECK_NULL(hipHostMalloc(&I->h_gmeta, (int64_t)I->max_q * 16));
+  HIP_CHECK_NULL(hipHostMalloc(&I->h_gfound, (int64_t)I->max_q * 4));
   for (Impl::Col* c : {&I->A, &I->B}) {
     HIP_CHECK_NULL(hipMalloc(&c->keys, max_rows * KEYW));
     HIP_CHECK_NULL(hipMalloc(&c->meta, max_rows * 8));
@@ -1407,7 +1425,7 @@ bool Slab::RangeBatchStart(const std::vector<DevRangeQ>& qs, std::string* err) {
                      dim3(256), 0, I->stream, I->A.keys, I->A.rev, I->n,
                      I->DA.keys, I->DA.rev, I->dn, I->d_fkeys, I->d_frev,
                      I->nf, I->d_qs, nq, I->d_qb);
-  hipLaunchKernelGGL(k_range_scan2, dim3(nq), dim3(SCAN_T), 0, I->stream,
+  hipLaunchKernelGGL(k_range_scan2, dim3(nq), dim3((uint32_t)I->scan_t), 0, I->stream,
                      I->A.keys, I->A.meta, I->A.rev, I->n, I->DA.keys,
                      I->DA.meta, I->DA.rev, I->dn, I->d_qb, I->d_qs, nq,
                      I->max_cap, I->d_rows, I->d_rows2, I->d_rowsm, I->d_found,
@@ -1553,6 +1571,54 @@ bool Slab::DrainD2H(std::string* err) {
 
 bool Slab::GetBatch(const std::vector<DevGetQ>& qs, std::vector<GetResult>* outs,
                     std::string* err) {
+  return GetBatchEx(qs, true, outs, err);
+}
+
+bool Slab::GetBatchStart(const std::vector<DevGetQ>& qs, std::string* err) {
+  Impl* I = p;
+  int nq = (int)qs.size();
+  if (nq == 0) return true;
+  if (nq > I->max_q) { if (err) *err = "too many gets per batch"; return false; }
+  HIP_CHECK(hipMemcpyAsync(I->d_gq, qs.data(), sizeof(DevGetQ) * nq,
+                           hipMemcpyHostToDevice, I->stream));
+  HIP_CHECK(hipEventRecord(I->ev_g0, I->stream));
+  int blocks = (int)ceil_div(nq, 4);
+  hipLaunchKernelGGL(k_get2, dim3(blocks), dim3(256), 0, I->stream, I->A.keys,
+                     I->A.meta, I->A.rev, I->A.vo, I->DA.keys, I->DA.meta,
+                     I->DA.rev, I->DA.vo, I->n, I->dn, I->d_fkeys, I->d_frev,
+                     I->nf, I->heapA, I->d_gq, nq, 0,
+                     I->d_gbuf, 0, I->d_orev, I->d_ometa, I->d_found32,
+                     I->d_ovf);
+  HIP_CHECK(hipMemcpyAsync(I->h_gmeta, I->d_orev, nq * 8, hipMemcpyDeviceToHost, I->stream));
+  HIP_CHECK(hipMemcpyAsync(I->h_gmeta + I->max_q, I->d_ometa, nq * 8, hipMemcpyDeviceToHost, I->stream));
+  HIP_CHECK(hipMemcpyAsync(I->h_gfound, I->d_found32, nq * 4, hipMemcpyDeviceToHost, I->stream));
+  HIP_CHECK(hipEventRecord(I->ev_g1, I->stream));
+  return true;
+}
+
+bool Slab::GetBatchFinish(int nq, std::vector<GetResult>* outs, std::string* err) {
+  Impl* I = p;
+  outs->assign(nq, GetResult());
+  if (nq == 0) return true;
+  HIP_CHECK(hipEventSynchronize(I->ev_g1));
+  float ms = 0;
+  (void)hipEventElapsedTime(&ms, I->ev_g0, I->ev_g1);
+  perf.get_ms += ms;
+  perf.get_launches++;
+  for (int q = 0; q < nq; ++q) {
+    GetResult& g = (*outs)[q];
+    g.found = I->h_gfound[q] != 0;
+    if (!g.found) continue;
+    uint64_t m = I->h_gmeta[I->max_q + q];
+    g.rev = I->h_gmeta[q];
+    g.tomb = (m & M_TOMB) != 0;
+    g.vlen = meta_vlen(m);
+  }
+  return true;
+}
+
+bool Slab::GetBatchEx(const std::vector<DevGetQ>& qs, bool values,
+                      std::vector<GetResult>* outs, std::string* err) {
   Impl* I = p;
   auto tt0 = std::chrono::steady_clock::now();
   auto lap = [&](double* acc) {
@@ -1575,7 +1641,7 @@ bool Slab::GetBatch(const std::vector<DevGetQ>& qs, std::vector<GetResult>* outs
   hipLaunchKernelGGL(k_get2, dim3(blocks), dim3(256), 0, I->stream, I->A.keys,
                      I->A.meta, I->A.rev, I->A.vo, I->DA.keys, I->DA.meta,
                      I->DA.rev, I->DA.vo, I->n, I->dn, I->d_fkeys, I->d_frev,
-                     I->nf, I->heapA, I->d_gq, nq,
+                     I->nf, I->heapA, I->d_gq, nq, values ? 1 : 0,
                      I->d_gbuf, slot, I->d_orev, I->d_ometa, I->d_found32,
                      I->d_ovf);
   HIP_CHECK(hipEventRecord(I->ev1, I->stream));
@@ -1600,6 +1666,8 @@ bool Slab::GetBatch(const std::vector<DevGetQ>& qs, std::vector<GetResult>* outs
     g.rev = orev[q];
     g.tomb = (ometa[q] & M_TOMB) != 0;
     uint32_t vlen = meta_vlen(ometa[q]);
+    if (!values) { g.vlen = vlen; continue; }
+    g.vlen = vlen;
     g.val.resize(vlen);
     if (vlen)
       HIP_CHECK(hipMemcpy(g.val.data(), I->d_gbuf + (int64_t)q * slot, vlen,

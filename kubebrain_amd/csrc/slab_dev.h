@@ -73,6 +73,7 @@ struct GetResult {
   bool found = false;
   bool tomb = false;
   uint64_t rev = 0;
+  uint32_t vlen = 0;  // value length (filled even in meta-only mode)
   std::string val;
 };
 
@@ -162,9 +163,19 @@ class Slab {
                         std::vector<RangeResult>* outs, std::string* err);
   bool DrainD2H(std::string* err);
 
-  // batched MVCC point read (range.go:91-121 reverse-iter semantics)
+  // batched MVCC point read (range.go:91-121 reverse-iter semantics).
+  // GetBatchEx(values=false) skips the value copy (meta-only: found/rev/
+  // tomb) — the device-side revIndex lookup of the batched txn path.
   bool GetBatch(const std::vector<DevGetQ>& qs, std::vector<GetResult>* outs,
                 std::string* err);
+  bool GetBatchEx(const std::vector<DevGetQ>& qs, bool values,
+                  std::vector<GetResult>* outs, std::string* err);
+  // async split: Start launches the lookup; Finish waits ONLY on the
+  // lookup's completion event, so kernels launched on the stream AFTER
+  // Start (e.g. the range batch) keep running while the host consumes the
+  // results. values=false only (meta-only lookups).
+  bool GetBatchStart(const std::vector<DevGetQ>& qs, std::string* err);
+  bool GetBatchFinish(int nq, std::vector<GetResult>* outs, std::string* err);
 
   // compaction mark+sweep over encoded borders (compact.go:55-68 +
   // scanner.go:444-491, 566-591). Bounds are (key96, rev) pairs.

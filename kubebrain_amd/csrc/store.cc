@@ -1048,13 +1048,57 @@ bool Store::BenchStep(const uint8_t* qbuf, size_t nq, const uint8_t* tbuf,
   std::vector<int64_t> limits;
   parseBenchQueries(qbuf, nq, committed_, mode, &qall, &limits);
   const bool d2h = (mode & 1) != 0;
-  auto t0 = std::chrono::steady_clock::now();
-  if (!slab_->RangeBatchStart(qall, err)) return false;
-  // overlap: the kernels read device state snapshotted at launch; the txn
-  // batch below touches only host staging until the next syncReads, so the
-  // in-flight scan's results are unchanged (MVCC readRev pinning)
+  // txn ops parsed up front so the batched CAS lookup (f1) can launch BEFORE
+  // the range batch: stream order runs the small lookup first, the host
+  // applies the conditional-update protocol against the device results while
+  // the range kernels are still in flight (both read the same pre-step
+  // snapshot; writes stage host-side until the next syncReads)
+  std::vector<BatchOp> tops(ntx);
+  bool tuniq = ntx > 0 && ntx <= 1024;
   if (ntx > 0) {
-    if (!BenchTxn(tbuf, ntx, out_revs, err)) return false;
+    std::unordered_set<Bytes> seen;
+    const uint8_t* p = tbuf;
+    for (size_t i = 0; i < ntx; ++i) {
+      uint32_t klen, vlen;
+      uint64_t prev;
+      memcpy(&klen, p, 4); p += 4;
+      memcpy(&prev, p, 8); p += 8;
+      memcpy(&vlen, p, 4); p += 4;
+      tops[i].key.assign((const char*)p, klen); p += klen;
+      tops[i].val.assign((const char*)p, vlen); p += vlen;
+      tops[i].prev = prev;
+      if (tuniq && (!seen.insert(tops[i].key).second ||
+                    validateKey(tops[i].key) != OK))
+        tuniq = false;
+    }
+  }
+  auto t0 = std::chrono::steady_clock::now();
+  if (tuniq) {
+    std::vector<DevGetQ> gq(ntx);
+    for (size_t i = 0; i < ntx; ++i) {
+      pad96(tops[i].key, gq[i].key);
+      gq[i].read_rev = UINT64_MAX;
+    }
+    if (!slab_->GetBatchStart(gq, err)) return false;
+  }
+  if (!slab_->RangeBatchStart(qall, err)) return false;
+  if (ntx > 0) {
+    if (tuniq) {
+      std::vector<kbslab::GetResult> cur;
+      if (!slab_->GetBatchFinish((int)ntx, &cur, err)) return false;
+      if (!applyTxnOps(tops.data(), ntx, cur, out_revs, err)) return false;
+    } else {
+      // duplicate keys: serial reference protocol (overlapped, host-only)
+      for (size_t i = 0; i < ntx; ++i) {
+        Status st;
+        auto r = Update(tops[i].key, tops[i].val, tops[i].prev, &st);
+        if (st != OK) {
+          if (err) *err = "bench txn failed st=" + std::to_string(st);
+          return false;
+        }
+        out_revs[i] = r.succeeded ? r.header_revision : 0;
+      }
+    }
   }
   std::vector<kbslab::RangeResult> outs;
   if (!slab_->RangeBatchFinish((int)nq, d2h, false, &outs, err)) return false;
@@ -1072,41 +1116,170 @@ bool Store::BenchStep(const uint8_t* qbuf, size_t nq, const uint8_t* tbuf,
 
 bool Store::BenchTxn(const uint8_t* tbuf, size_t n, uint64_t* out_revs,
                      std::string* err) {
-  // packed records: {u32 klen; u64 prev_rev; u32 vlen; key; val} x n
-  const uint8_t* p = tbuf;
-  for (size_t i = 0; i < n; ++i) {
-    uint32_t klen, vlen;
-    uint64_t prev;
-    memcpy(&klen, p, 4); p += 4;
-    memcpy(&prev, p, 8); p += 8;
-    memcpy(&vlen, p, 4); p += 4;
-    Bytes key((const char*)p, klen); p += klen;
-    Bytes val((const char*)p, vlen); p += vlen;
-    Status st;
-    auto r = Update(key, val, prev, &st);
-    if (st != OK) {
-      if (err) *err = "bench txn failed st=" + std::to_string(st);
-      return false;
+  // f1, GPU-batched conditional updates: the CAS reads of the
+  // reference's txn protocol (txn.go:249-265, creator/naive.go:48-105) run
+  // as ONE batched device lookup against the slab's revision state
+  // (base+delta runs, k_get2 meta-only at R=+inf ≡ the revision-row content
+  // — rows and revision-rows are always staged together, so the latest
+  // object row's (rev, tomb) IS the revIndex entry). The batched path never
+  // reads the host revIndex_ map. Batches with duplicate keys fall back to
+  // the serial protocol (within-batch CAS chains need sequencing).
+  // Packed records: {u32 klen; u64 prev_rev; u32 vlen; key; val} x n.
+  std::vector<BatchOp> ops(n);
+  std::unordered_set<Bytes> seen;
+  bool uniq = true;
+  {
+    const uint8_t* p = tbuf;
+    for (size_t i = 0; i < n; ++i) {
+      uint32_t klen, vlen;
+      uint64_t prev;
+      memcpy(&klen, p, 4); p += 4;
+      memcpy(&prev, p, 8); p += 8;
+      memcpy(&vlen, p, 4); p += 4;
+      ops[i].key.assign((const char*)p, klen); p += klen;
+      ops[i].val.assign((const char*)p, vlen); p += vlen;
+      ops[i].prev = prev;
+      if (uniq && !seen.insert(ops[i].key).second) uniq = false;
     }
-    out_revs[i] = r.succeeded ? r.header_revision : 0;
+  }
+  std::lock_guard<std::recursive_mutex> lk(mu_);
+  if (!uniq) {
+    for (size_t i = 0; i < n; ++i) {
+      Status st;
+      auto r = Update(ops[i].key, ops[i].val, ops[i].prev, &st);
+      if (st != OK) {
+        if (err) *err = "bench txn failed st=" + std::to_string(st);
+        return false;
+      }
+      out_revs[i] = r.succeeded ? r.header_revision : 0;
+    }
+    return true;
+  }
+  const size_t kChunk = 1024;  // device query batch bound (KB_MAX_Q)
+  for (size_t c0 = 0; c0 < n; c0 += kChunk) {
+    size_t cn = std::min(kChunk, n - c0);
+    if (!syncReads(err)) return false;  // device revision state current
+    std::vector<DevGetQ> qs(cn);
+    for (size_t i = 0; i < cn; ++i) {
+      Status v = validateKey(ops[c0 + i].key);
+      if (v != OK) { if (err) *err = "bench txn bad key"; return false; }
+      pad96(ops[c0 + i].key, qs[i].key);
+      qs[i].read_rev = UINT64_MAX;
+    }
+    std::vector<kbslab::GetResult> cur;
+    if (!slab_->GetBatchEx(qs, /*values=*/false, &cur, err)) return false;
+    if (!applyTxnOps(&ops[c0], cn, cur, out_revs + c0, err)) return false;
+  }
+  return true;
+}
+
+// the host half of the batched conditional-update protocol: consumes the
+// device CAS-lookup results in op order (revisions are consumed per op,
+// success or not — txn.go:139-142)
+bool Store::applyTxnOps(const BatchOp* ops, size_t cn,
+                        const std::vector<kbslab::GetResult>& cur,
+                        uint64_t* out_revs, std::string* err) {
+  for (size_t i = 0; i < cn; ++i) {
+    const BatchOp& op = ops[i];
+    ops_update_++;
+    Status dst;
+    uint64_t newRev = deal(op.prev, &dst);
+    if (dst != OK) {
+      if (err) *err = "bench txn rev drift";
+      return false;  // Update surfaces dst as *st (txn.go:139-142)
+    }
+    bool ok;
+    if (op.prev == 0) {
+      // create path (creator/naive.go:48-105) against device state
+      ok = !cur[i].found || (cur[i].tomb && cur[i].rev < newRev);
+    } else {
+      // CAS(revKey, new8, old8): the flagged 9B value never matches
+      ok = cur[i].found && !cur[i].tomb && cur[i].rev == op.prev;
+    }
+    if (ok) {
+      putRevRow(op.key, newRev, false);
+      putRow(op.key, newRev, op.val);
+    }
+    notify(op.key, op.val, newRev, op.prev, ok,
+           op.prev == 0 ? Event::CREATE : Event::PUT);
+    out_revs[i] = ok ? newRev : 0;
   }
   return true;
 }
 
 bool Store::BenchDel(const uint8_t* dbuf, size_t n, uint64_t* out_revs,
                      std::string* err) {
-  // packed: {u32 klen; u64 prev_rev; key} x n ; out_revs[i]=rev or 0
-  const uint8_t* p = dbuf;
-  for (size_t i = 0; i < n; ++i) {
-    uint32_t klen;
-    uint64_t prev;
-    memcpy(&klen, p, 4); p += 4;
-    memcpy(&prev, p, 8); p += 8;
-    Bytes key((const char*)p, klen); p += klen;
-    Status st;
-    auto r = Delete(key, prev, &st);
-    if (st != OK) { if (err) *err = "bench del st=" + std::to_string(st); return false; }
-    out_revs[i] = r.succeeded ? r.header_revision : 0;
+  // Batched deletes: ONE device lookup prefetches every key's current
+  // (value, modRevision), then the reference's delete protocol
+  // (txn.go:79-190) runs against the prefetched state. Unique keys per
+  // batch; duplicates fall back to the serial path.
+  // Packed: {u32 klen; u64 prev_rev; key} x n ; out_revs[i]=rev or 0.
+  struct Op { Bytes key; uint64_t prev; };
+  std::vector<Op> ops(n);
+  std::unordered_set<Bytes> seen;
+  bool uniq = true;
+  {
+    const uint8_t* p = dbuf;
+    for (size_t i = 0; i < n; ++i) {
+      uint32_t klen;
+      uint64_t prev;
+      memcpy(&klen, p, 4); p += 4;
+      memcpy(&prev, p, 8); p += 8;
+      ops[i].key.assign((const char*)p, klen); p += klen;
+      ops[i].prev = prev;
+      if (uniq && !seen.insert(ops[i].key).second) uniq = false;
+    }
+  }
+  std::lock_guard<std::recursive_mutex> lk(mu_);
+  if (!uniq) {
+    for (size_t i = 0; i < n; ++i) {
+      Status st;
+      auto r = Delete(ops[i].key, ops[i].prev, &st);
+      if (st != OK) { if (err) *err = "bench del st=" + std::to_string(st); return false; }
+      out_revs[i] = r.succeeded ? r.header_revision : 0;
+    }
+    return true;
+  }
+  const size_t kChunk = 1024;
+  for (size_t c0 = 0; c0 < n; c0 += kChunk) {
+    size_t cn = std::min(kChunk, n - c0);
+    if (!syncReads(err)) return false;
+    std::vector<DevGetQ> qs(cn);
+    for (size_t i = 0; i < cn; ++i) {
+      Status v = validateKey(ops[c0 + i].key);
+      if (v != OK) { if (err) *err = "bench del bad key"; return false; }
+      pad96(ops[c0 + i].key, qs[i].key);
+      qs[i].read_rev = UINT64_MAX;
+    }
+    std::vector<kbslab::GetResult> cur;
+    if (!slab_->GetBatchEx(qs, /*values=*/true, &cur, err)) return false;
+    for (size_t i = 0; i < cn; ++i) {
+      const Op& op = ops[c0 + i];
+      ops_delete_++;
+      if (!cur[i].found || cur[i].tomb) {  // NOTFOUND leg (txn.go:148-151)
+        uint64_t rev = mustDeal(op.prev);
+        notify(op.key, Bytes(), rev, 0, false, Event::DELETE);
+        out_revs[c0 + i] = 0;
+        continue;
+      }
+      uint64_t modRevision = cur[i].rev;
+      Status dst;
+      uint64_t newRev = deal(op.prev, &dst);
+      if (dst != OK) { if (err) *err = "bench del rev drift"; return false; }
+      bool ok;
+      if (op.prev > 0 && op.prev != modRevision) {
+        ok = false;  // txn.go:162-166
+      } else {
+        // CAS(revKey, rev8+0x00, expected8) with expected == modRevision;
+        // device state is consistent by construction (rows + rev-rows are
+        // staged together), so the CAS succeeds (txn.go:177-186)
+        ok = true;
+        putRevRow(op.key, newRev, true);
+        putRow(op.key, newRev, kTombstone);
+      }
+      notify(op.key, cur[i].val, newRev, modRevision, ok, Event::DELETE);
+      out_revs[c0 + i] = ok ? newRev : 0;
+    }
   }
   return true;
 }

@@ -141,6 +141,10 @@ class Store {
   void putRow(const Bytes& key, uint64_t rev, const Bytes& val);
   void putRevRow(const Bytes& key, uint64_t objrev, bool flag9);
   void pumpEvents();  // fan-out pending events via the GPU filter
+  struct BatchOp { Bytes key, val; uint64_t prev; };
+  bool applyTxnOps(const BatchOp* ops, size_t cn,
+                   const std::vector<kbslab::GetResult>& cur,
+                   uint64_t* out_revs, std::string* err);
   struct Watcher;
   void releaseSlot(Watcher& w);  // recycle a device watcher slot (idempotent)
   // push pending values + new rows to the device (delta-run merge); folds the

@@ -516,3 +516,91 @@ def test_range_global_cabi(dual):
     check_against_list(0)
     lib.kb_comm_free(C.c_void_p(dual.p.h))
     check_against_list(5)    # degrade path again after free
+
+
+def test_bench_txn_batched_parity():
+    """kb_bench_txn / kb_bench_del (f1: the batched conditional-update CAS
+    runs as ONE device lookup against the slab's revision state, comm-free
+    host apply) must produce byte-identical store state, responses and
+    event streams to the oracle's serial Update/Delete protocol."""
+    import ctypes as C
+    import struct as S
+
+    import numpy as np
+    d = parity.Dual()
+    try:
+        ns = b"/registry/pods/ns-88"
+        revs = {}
+        for i in range(40):
+            r = d.create(ns + b"/o-%03d" % i, b"v0")
+            revs[i] = r.header_revision
+        w = d.watch(ns + b"/", 0)
+        # batch 1: mixed success/CAS-fail/create-path updates (unique keys)
+        ops = []
+        for i in range(30):
+            if i % 3 == 0:
+                ops.append((ns + b"/o-%03d" % i, revs[i], b"u1"))       # ok
+            elif i % 3 == 1:
+                ops.append((ns + b"/o-%03d" % i, revs[i] + 7, b"u1"))   # CAS fail
+            else:
+                ops.append((ns + b"/new-%03d" % i, 0, b"c1"))           # create path
+        parts = []
+        for k, pr, v in ops:
+            parts.append(S.pack("<IQI", len(k), pr, len(v)))
+            parts.append(k)
+            parts.append(v)
+        out = np.empty(len(ops), dtype=np.uint64)
+        rc = d.p._f("bench_txn")(C.c_void_p(d.p.h), b"".join(parts),
+                                 C.c_size_t(len(ops)),
+                                 out.ctypes.data_as(C.POINTER(C.c_uint64)))
+        assert rc == 0
+        # oracle: the same ops through the serial protocol
+        for (k, pr, v), nr in zip(ops, out):
+            ro = d.o.update(k, v, pr)
+            assert (ro.succeeded and ro.header_revision or 0) == int(nr), (k, ro, nr)
+        # batch 2: deletes — hit, miss, wrong prev, tombstone (unique keys)
+        dels = [(ns + b"/o-000", 0), (ns + b"/o-003", int(out[3]) or revs[3]),
+                (ns + b"/none", 0), (ns + b"/o-006", 12345)]
+        parts = []
+        for k, pr in dels:
+            parts.append(S.pack("<IQ", len(k), pr))
+            parts.append(k)
+        outd = np.empty(len(dels), dtype=np.uint64)
+        rc = d.p._f("bench_del")(C.c_void_p(d.p.h), b"".join(parts),
+                                 C.c_size_t(len(dels)),
+                                 outd.ctypes.data_as(C.POINTER(C.c_uint64)))
+        assert rc == 0
+        for (k, pr), nr in zip(dels, outd):
+            ro = d.o.delete(k, pr)
+            assert (ro.succeeded and ro.header_revision or 0) == int(nr), (k, ro, nr)
+        # delete an already-deleted key in a fresh batch (tombstone leg)
+        k0 = ns + b"/o-000"
+        parts = [S.pack("<IQ", len(k0), 0), k0]
+        out1 = np.empty(1, dtype=np.uint64)
+        rc = d.p._f("bench_del")(C.c_void_p(d.p.h), b"".join(parts),
+                                 C.c_size_t(1),
+                                 out1.ctypes.data_as(C.POINTER(C.c_uint64)))
+        assert rc == 0
+        ro = d.o.delete(k0, 0)
+        assert (ro.succeeded and ro.header_revision or 0) == int(out1[0])
+        # duplicate-key batch falls back to the serial path — same results
+        k1 = ns + b"/o-012"
+        pr1 = revs[12]
+        parts = []
+        for pr in (pr1, pr1):  # second op must CAS-fail after the first
+            parts.append(S.pack("<IQI", len(k1), pr, 2))
+            parts.append(k1)
+            parts.append(b"dd")
+        out2 = np.empty(2, dtype=np.uint64)
+        rc = d.p._f("bench_txn")(C.c_void_p(d.p.h), b"".join(parts),
+                                 C.c_size_t(2),
+                                 out2.ctypes.data_as(C.POINTER(C.c_uint64)))
+        assert rc == 0
+        for pr, nr in zip((pr1, pr1), out2):
+            ro = d.o.update(k1, b"dd", pr)
+            assert (ro.succeeded and ro.header_revision or 0) == int(nr)
+        d.poll(w)
+        d.diff_dump()
+        d.diff_event_log()
+    finally:
+        d.close()
