@@ -587,29 +587,57 @@ def main():
         elapsed = float(e.item())
     p = perf(store)
 
+    # one-off PCIe reference: pinned D2H bandwidth of this box (torch copy)
+    pcie_gbps = None
+    if torch.cuda.is_available():
+        try:
+            src = torch.empty(256 << 20, dtype=torch.uint8, device="cuda")
+            dst = torch.empty(256 << 20, dtype=torch.uint8, pin_memory=True)
+            dst.copy_(src, non_blocking=True)
+            torch.cuda.synchronize()
+            t0 = time.time()
+            for _ in range(4):
+                dst.copy_(src, non_blocking=True)
+            torch.cuda.synchronize()
+            pcie_gbps = round(4 * (256 << 20) / (time.time() - t0) / 1e9, 1)
+            del src, dst
+        except Exception:
+            pass
+
     # PCIe-inclusive measurement (separate, untimed-region): full records,
     # payload copy pipelined across steps (kb_sync drains the tail before the
     # clock stops). Two untimed warmup steps first: the first pipelined call
     # allocates the ping-pong pack arenas + pinned staging (~1s one-off).
+    def d2h_leg(mode, n_steps):
+        for _ in range(2):
+            one_step(mode=mode)
+        drain()
+        p0 = perf(store)
+        t0 = time.time()
+        for _ in range(n_steps):
+            one_step(mode=mode)
+        drain()
+        dt = time.time() - t0
+        p1 = perf(store)
+        detail = {
+            "wall_ms_per_step": round(dt / n_steps * 1e3, 3),
+            "pack_d2h_ms_per_step": round(
+                (p1["pack_d2h_ms"] - p0["pack_d2h_ms"]) / n_steps, 3),
+            "kernel_ms_per_step": round(
+                (p1["scan_ms"] + p1["gather_ms"] - p0["scan_ms"] - p0["gather_ms"])
+                / n_steps, 3),
+            "get_ms_per_step": round((p1["get_ms"] - p0["get_ms"]) / n_steps, 3),
+            "MB_per_step": round((p1["bytes_gathered"] - p0["bytes_gathered"])
+                                 / n_steps / 1e6, 1),
+            "sync_s": round(p1["sync_s"] - p0["sync_s"], 3),
+        }
+        return dt, detail
+
     d2h_steps = max(4, args.steps // 4)
-    for _ in range(2):
-        one_step(mode=1)
-    drain()
-    t0 = time.time()
-    for _ in range(d2h_steps):
-        one_step(mode=1)
-    drain()
-    d2h_elapsed = time.time() - t0
+    d2h_elapsed, d2h_detail = d2h_leg(1, d2h_steps)
     # keys-only variant (etcd3 KeysOnly semantics; response payload is
     # key + mod-revision per winner — an extension, see module docstring)
-    for _ in range(2):
-        one_step(mode=3)
-    drain()
-    t0 = time.time()
-    for _ in range(d2h_steps):
-        one_step(mode=3)
-    drain()
-    d2h_ko_elapsed = time.time() - t0
+    d2h_ko_elapsed, d2h_ko_detail = d2h_leg(3, d2h_steps)
 
     # cross-shard Range leg (configs[3]) at N>1: RCCL/xGMI exchange + merge.
     # Exception-guarded: a failure degrades to a JSON note, never the run.
@@ -737,6 +765,9 @@ def main():
             "ops_per_sec_with_d2h": round(OPS_PER_STEP * d2h_steps * world / d2h_elapsed, 1),
             "ops_per_sec_with_d2h_keys_only": round(
                 OPS_PER_STEP * d2h_steps * world / d2h_ko_elapsed, 1),
+            "d2h_detail": d2h_detail,
+            "d2h_keys_only_detail": d2h_ko_detail,
+            "pcie_d2h_GBps": pcie_gbps,
             "step_split_ms": {"range": round(split_snapshot["range_s"] / args.steps * 1e3, 3),
                                "txn": round(split_snapshot["txn_s"] / args.steps * 1e3, 3)},
             "roofline": roofline,
