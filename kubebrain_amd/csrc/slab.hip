@@ -54,122 +54,133 @@ __device__ __forceinline__ int keycmp96(const uint8_t* a, const uint8_t* b) {
   return 0;
 }
 
-// first row with (key,rev) >= (qkey,qrev)
-__device__ int64_t d_lower_bound(const uint8_t* keys, const uint64_t* rev,
-                                 int64_t n, const uint8_t* qkey, uint64_t qrev) {
-  int64_t lo = 0, hi = n;
+// Full-key compare with spill tails (keys > KEYW): the 96B
+// zero-padded prefix decides every compare except exact 96-byte prefix ties
+// — only possible when BOTH keys are >= 96B (key bytes are > 0x24 > 0x00) —
+// which compare their tails from the spill heaps. The tail loop runs on the
+// rare tie path only; the hot predicate never touches it.
+__device__ __forceinline__ int keycmp_ext(
+    const uint8_t* a96, uint32_t alen, uint64_t ako, const uint8_t* aspill,
+    const uint8_t* b96, uint32_t blen, uint64_t bko, const uint8_t* bspill) {
+  int c = keycmp96(a96, b96);
+  if (c || (alen <= (uint32_t)KEYW && blen <= (uint32_t)KEYW)) return c;
+  uint32_t at = alen > (uint32_t)KEYW ? alen - KEYW : 0;
+  uint32_t bt = blen > (uint32_t)KEYW ? blen - KEYW : 0;
+  const uint8_t* ap = aspill + ako;
+  const uint8_t* bp = bspill + bko;
+  uint32_t m = at < bt ? at : bt;
+  for (uint32_t i = 0; i < m; ++i)
+    if (ap[i] != bp[i]) return ap[i] < bp[i] ? -1 : 1;
+  return at == bt ? 0 : (at < bt ? -1 : 1);
+}
+
+// one sorted run's columns (base or delta); spill is shared store-wide
+struct Run {
+  const uint8_t* keys;
+  const uint64_t* meta;
+  const uint64_t* rev;
+  const uint64_t* vo;
+  const uint64_t* ko;
+};
+
+// a query-side key: 96B padded prefix + optional tail in `tails`
+struct QKey {
+  const uint8_t* k96;
+  uint32_t len;
+  uint64_t ko;
+  const uint8_t* tails;
+};
+
+__device__ __forceinline__ QKey row_qk(const Run& r, const uint8_t* spill,
+                                       int64_t i) {
+  return QKey{r.keys + i * KEYW, meta_klen(r.meta[i]), r.ko[i], spill};
+}
+
+__device__ __forceinline__ int rowcmp_q(const Run& r, const uint8_t* spill,
+                                        int64_t i, const QKey& q) {
+  return keycmp_ext(r.keys + i * KEYW, meta_klen(r.meta[i]), r.ko[i], spill,
+                    q.k96, q.len, q.ko, q.tails);
+}
+
+// first row in [lo,hi) with (key,rev) >= (q,qrev)
+__device__ int64_t d_lb_range(const Run& r, const uint8_t* spill, int64_t lo,
+                              int64_t hi, const QKey& q, uint64_t qrev) {
   while (lo < hi) {
     int64_t mid = (lo + hi) >> 1;
-    int c = keycmp96(keys + mid * KEYW, qkey);
-    if (c < 0 || (c == 0 && rev[mid] < qrev)) lo = mid + 1; else hi = mid;
+    int c = rowcmp_q(r, spill, mid, q);
+    if (c < 0 || (c == 0 && r.rev[mid] < qrev)) lo = mid + 1; else hi = mid;
   }
   return lo;
 }
 
-// first row with (key,rev) > (qkey,qrev)
-__device__ int64_t d_upper_bound(const uint8_t* keys, const uint64_t* rev,
-                                 int64_t n, const uint8_t* qkey, uint64_t qrev) {
-  int64_t lo = 0, hi = n;
-  while (lo < hi) {
-    int64_t mid = (lo + hi) >> 1;
-    int c = keycmp96(keys + mid * KEYW, qkey);
-    if (c < 0 || (c == 0 && rev[mid] <= qrev)) lo = mid + 1; else hi = mid;
+// wave-cooperative 64-ary lower bound: every round all 64 lanes probe one
+// pivot each (64 independent loads in ONE memory round trip), the ballot of
+// "pivot < q" collapses the range 64x. Depth log64(n): a 21M-row slab costs
+// ~5 round trips where the serial binary search pays ~25 dependent ones.
+// All lanes of the wave must call; the result is wave-uniform.
+__device__ int64_t d_lb_wave(const Run& r, const uint8_t* spill, int64_t n,
+                             const QKey& q, uint64_t qrev) {
+  const int lane = threadIdx.x & 63;
+  int64_t lo = 0, hi = n;  // answer in [lo, hi]
+  while (hi - lo > 64) {
+    int64_t step = (hi - lo) >> 6;  // >= 1
+    int64_t p = lo + (int64_t)(lane + 1) * step;
+    bool lt = false;  // keys[p] < q (monotone non-increasing in p)
+    if (p < hi) {
+      int c = rowcmp_q(r, spill, p, q);
+      lt = c < 0 || (c == 0 && r.rev[p] < qrev);
+    }
+    int k = __popcll(__ballot(lt));
+    int64_t nlo = lo + (int64_t)k * step;
+    int64_t nhi = k == 64 ? hi : min(lo + (int64_t)(k + 1) * step, hi);
+    lo = nlo;
+    hi = nhi;
   }
-  return lo;
+  int64_t p = lo + lane;
+  bool ge = false;  // first in-range row with keys[p] >= q
+  if (p < hi) {
+    int c = rowcmp_q(r, spill, p, q);
+    ge = !(c < 0 || (c == 0 && r.rev[p] < qrev));
+  }
+  uint64_t b = __ballot(ge);
+  return b ? lo + (__ffsll((unsigned long long)b) - 1) : hi;
 }
 
-// ---- same_next: adjacent-key equality bits, LDS-tiled coalesced loads ----
+// adjacent-row full-key equality (same spill both sides)
+__device__ __forceinline__ bool rows_same_key(const uint8_t* keys,
+                                              const uint64_t* meta,
+                                              const uint64_t* ko,
+                                              const uint8_t* spill, int64_t i,
+                                              int64_t j) {
+  const uint64_t* a = (const uint64_t*)(keys + i * KEYW);
+  const uint64_t* b = (const uint64_t*)(keys + j * KEYW);
+#pragma unroll
+  for (int k = 0; k < KEYW / 8; ++k)
+    if (a[k] != b[k]) return false;
+  uint32_t al = meta_klen(meta[i]), bl = meta_klen(meta[j]);
+  if (al <= (uint32_t)KEYW && bl <= (uint32_t)KEYW) return true;
+  if (al != bl) return false;
+  const uint8_t* ap = spill + ko[i];
+  const uint8_t* bp = spill + ko[j];
+  for (uint32_t t = 0; t < al - (uint32_t)KEYW; ++t)
+    if (ap[t] != bp[t]) return false;
+  return true;
+}
+
+// ---- same_next: adjacent-key equality bits over one run ----
 __global__ void k_same_next(const uint8_t* __restrict__ keys,
-                            uint64_t* __restrict__ meta, int64_t n) {
-  __shared__ uint64_t tile64[(256 + 1) * KEYW / 8];  // u64 storage: aligned
-  uint8_t* tile = (uint8_t*)tile64;
-  int64_t base = (int64_t)blockIdx.x * 256;
-  if (base >= n) return;
-  int64_t count = min((int64_t)256, n - base);
-  int64_t tload = min(count + 1, n - base);  // one extra row for the seam
-  int64_t words = tload * KEYW / 4;
-  const uint32_t* src = (const uint32_t*)(keys + base * KEYW);
-  uint32_t* dst = (uint32_t*)tile;
-  for (int64_t w = threadIdx.x; w < words; w += blockDim.x) dst[w] = src[w];
-  __syncthreads();
-  int64_t i = base + threadIdx.x;
+                            uint64_t* __restrict__ meta,
+                            const uint64_t* __restrict__ ko,
+                            const uint8_t* __restrict__ spill, int64_t n) {
+  int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
   if (i >= n) return;
   uint64_t m = meta[i];
-  bool same = false;
-  if (i + 1 < n) {  // tload always covers row i+1 here
-    const uint64_t* a = (const uint64_t*)(tile + (int64_t)threadIdx.x * KEYW);
-    const uint64_t* b = (const uint64_t*)(tile + ((int64_t)threadIdx.x + 1) * KEYW);
-    same = true;
-#pragma unroll
-    for (int k = 0; k < KEYW / 8; ++k)
-      if (a[k] != b[k]) { same = false; break; }
-  }
+  bool same = i + 1 < n && rows_same_key(keys, meta, ko, spill, i, i + 1);
   meta[i] = same ? (m | M_SAME_NEXT) : (m & ~M_SAME_NEXT);
-}
-
-// first row in [lo,hi) with (key,rev) >= (qkey,qrev)
-__device__ int64_t d_lb_range(const uint8_t* keys, const uint64_t* rev,
-                              int64_t lo, int64_t hi, const uint8_t* qkey,
-                              uint64_t qrev) {
-  while (lo < hi) {
-    int64_t mid = (lo + hi) >> 1;
-    int c = keycmp96(keys + mid * KEYW, qkey);
-    if (c < 0 || (c == 0 && rev[mid] < qrev)) lo = mid + 1; else hi = mid;
-  }
-  return lo;
 }
 
 constexpr uint64_t ROW_TAG_DELTA = 1ull << 63;  // winner row lives in the delta run
 constexpr uint64_t ROW_MASK = ROW_TAG_DELTA - 1;
-
-// two-level search: fence[i] = (key,rev) of row i*256; the fence (~96B per
-// 256 rows) stays L2/L3-resident, so a cold search costs ~1 HBM window
-// instead of log2(n) dependent misses.
-constexpr int64_t FENCE_STRIDE = 256;
-
-__global__ void k_build_fence(const uint8_t* __restrict__ keys,
-                              const uint64_t* __restrict__ rev, int64_t n,
-                              uint8_t* __restrict__ fkeys,
-                              uint64_t* __restrict__ frev, int64_t nf) {
-  int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
-  if (i >= nf) return;
-  int64_t row = i * FENCE_STRIDE;
-  const uint64_t* ks = (const uint64_t*)(keys + row * KEYW);
-  uint64_t* kd = (uint64_t*)(fkeys + i * KEYW);
-#pragma unroll
-  for (int k = 0; k < KEYW / 8; ++k) kd[k] = ks[k];
-  frev[i] = rev[row];
-}
-
-__device__ int64_t d_lb_fenced(const uint8_t* keys, const uint64_t* rev,
-                               int64_t n, const uint8_t* fkeys,
-                               const uint64_t* frev, int64_t nf,
-                               const uint8_t* qkey, uint64_t qrev) {
-  if (nf <= 0) {
-    int64_t lo = 0, hi = n;
-    while (lo < hi) {
-      int64_t mid = (lo + hi) >> 1;
-      int c = keycmp96(keys + mid * KEYW, qkey);
-      if (c < 0 || (c == 0 && rev[mid] < qrev)) lo = mid + 1; else hi = mid;
-    }
-    return lo;
-  }
-  int64_t lo = 0, hi = nf;
-  while (lo < hi) {  // first fence entry >= q
-    int64_t mid = (lo + hi) >> 1;
-    int c = keycmp96(fkeys + mid * KEYW, qkey);
-    if (c < 0 || (c == 0 && frev[mid] < qrev)) lo = mid + 1; else hi = mid;
-  }
-  int64_t rlo = lo == 0 ? 0 : (lo - 1) * FENCE_STRIDE + 1;
-  int64_t rhi = lo < nf ? lo * FENCE_STRIDE + 1 : n;  // result may be fence row itself
-  if (rhi > n) rhi = n;
-  while (rlo < rhi) {
-    int64_t mid = (rlo + rhi) >> 1;
-    int c = keycmp96(keys + mid * KEYW, qkey);
-    if (c < 0 || (c == 0 && rev[mid] < qrev)) rlo = mid + 1; else rhi = mid;
-  }
-  return rlo;
-}
 
 // ---- two-run variant: base run + sorted delta run merged at scan time ---
 // The delta run holds rows strictly newer than the base run's rows of the
@@ -177,32 +188,6 @@ __device__ int64_t d_lb_fenced(const uint8_t* keys, const uint64_t* rev,
 // winner when the delta has any row of the key with rev<=R, else the base
 // winner — base winners are suppressed by a delta probe, and the two ordered
 // winner lists are merged by rank (keys never collide across lists).
-
-// all 4 bounds of every query in one massively-parallel launch (one thread
-// per bound): the 4 dependent binary-search chains per query that used to
-// serialize inside the scan block run concurrently across the whole batch,
-// taking ~one search-chain latency total instead of per query.
-__global__ void k_range_bounds(const uint8_t* __restrict__ bkeys,
-                               const uint64_t* __restrict__ brev, int64_t n,
-                               const uint8_t* __restrict__ dkeys,
-                               const uint64_t* __restrict__ drev, int64_t dn,
-                               const uint8_t* __restrict__ fkeys,
-                               const uint64_t* __restrict__ frev, int64_t nf,
-                               const DevRangeQ* __restrict__ qs, int nq,
-                               int64_t* __restrict__ qb /* [4*nq] */) {
-  int i = blockIdx.x * blockDim.x + threadIdx.x;
-  if (i >= 4 * nq) return;
-  int q = i >> 2, which = i & 3;
-  const DevRangeQ& Q = qs[q];
-  int64_t r;
-  switch (which) {
-    case 0: r = d_lb_fenced(bkeys, brev, n, fkeys, frev, nf, Q.start, Q.start_rev); break;
-    case 1: r = d_lb_fenced(bkeys, brev, n, fkeys, frev, nf, Q.end, 0); break;
-    case 2: r = d_lb_range(dkeys, drev, 0, dn, Q.start, Q.start_rev); break;
-    default: r = d_lb_range(dkeys, drev, 0, dn, Q.end, 0); break;
-  }
-  qb[i] = r;
-}
 
 // scan block width: runtime-tunable (KB_SCAN_T in {256,512,1024}); the
 // kernel reads blockDim.x, only the LDS wave-count array is sized for the max
@@ -216,12 +201,12 @@ constexpr int SCAN_T_MAX = 1024;
 // winner row indices at their ordered offsets. Two __syncthreads per tile
 // of up to NW*4096 rows (the old design paid two per 1024 rows).
 __device__ int64_t scan_run_winners(
-    const uint8_t* __restrict__ keys, const uint64_t* __restrict__ meta,
-    const uint64_t* __restrict__ rev, int64_t lo, int64_t hi, uint64_t R,
+    const Run& run, const uint8_t* spill, int64_t lo, int64_t hi, uint64_t R,
     int64_t cap, uint64_t* out, int64_t out_cap, uint64_t tagbit,
-    const uint8_t* __restrict__ skeys, const uint64_t* __restrict__ srev,
-    int64_t slo, int64_t shi,  // suppression run (null => none)
+    const Run* srun, int64_t slo, int64_t shi,  // suppression run (null => none)
     int64_t* total_out, int64_t* scanned_accum, int* wave_cnt) {
+  const uint64_t* __restrict__ rev = run.rev;
+  const uint64_t* __restrict__ meta = run.meta;
   const int lane = threadIdx.x & 63, w = threadIdx.x >> 6;
   const int NW = blockDim.x >> 6;
   const int64_t span = hi - lo;
@@ -252,10 +237,11 @@ __device__ int64_t scan_run_winners(
         if (rv > 0 && rv <= R && !(m & M_TOMB))
           // same_next set => row i+1 exists and shares the key
           win = !(m & M_SAME_NEXT) || rev[i + 1] > R;
-        if (win && skeys) {
-          const uint8_t* kk = keys + i * KEYW;
-          int64_t lb = d_lb_range(skeys, srev, slo, shi, kk, 1);
-          if (lb < shi && srev[lb] <= R && keycmp96(skeys + lb * KEYW, kk) == 0)
+        if (win && srun) {
+          QKey kk = row_qk(run, spill, i);
+          int64_t lb = d_lb_range(*srun, spill, slo, shi, kk, 1);
+          if (lb < shi && srun->rev[lb] <= R &&
+              rowcmp_q(*srun, spill, lb, kk) == 0)
             win = false;  // a newer (delta) row of this key wins instead
         }
       }
@@ -297,47 +283,59 @@ __device__ int64_t scan_run_winners(
 }
 
 // lower_bound over a winner list via key indirection (no cross-list ties)
-__device__ int64_t d_lb_winlist(const uint64_t* rows, int64_t n,
-                                const uint8_t* bkeys, const uint8_t* dkeys,
-                                const uint8_t* qkey) {
+__device__ int64_t d_lb_winlist(const uint64_t* rows, int64_t n, const Run& b,
+                                const Run& d, const uint8_t* spill,
+                                const QKey& q) {
   int64_t lo = 0, hi = n;
   while (lo < hi) {
     int64_t mid = (lo + hi) >> 1;
     uint64_t rt = rows[mid];
-    const uint8_t* mk = (rt & ROW_TAG_DELTA ? dkeys : bkeys) +
-                        (rt & ROW_MASK) * KEYW;
-    if (keycmp96(mk, qkey) < 0) lo = mid + 1; else hi = mid;
+    const Run& rr = (rt & ROW_TAG_DELTA) ? d : b;
+    if (rowcmp_q(rr, spill, (int64_t)(rt & ROW_MASK), q) < 0) lo = mid + 1;
+    else hi = mid;
   }
   return lo;
 }
 
 __global__ void k_range_scan2(
-    const uint8_t* __restrict__ bkeys, const uint64_t* __restrict__ bmeta,
-    const uint64_t* __restrict__ brev, int64_t n,
-    const uint8_t* __restrict__ dkeys, const uint64_t* __restrict__ dmeta,
-    const uint64_t* __restrict__ drev, int64_t dn,
-    const int64_t* __restrict__ qb /* bounds from k_range_bounds */,
-    const DevRangeQ* __restrict__ qs, int nq, int64_t max_cap,
+    Run b, int64_t n, Run d, int64_t dn, const uint8_t* __restrict__ spill,
+    const uint8_t* __restrict__ qtails, const DevRangeQ* __restrict__ qs,
+    int nq, int64_t max_cap,
     uint64_t* __restrict__ rows_b, uint64_t* __restrict__ rows_d,
     uint64_t* __restrict__ rows_m, int64_t* __restrict__ found_out,
     int64_t* __restrict__ total_out, unsigned long long* __restrict__ scanned_out) {
   int q = blockIdx.x;
   if (q >= nq) return;
   __shared__ int wave_cnt[SCAN_T_MAX / 64];
+  __shared__ int64_t bounds_s[4];
   const DevRangeQ& Q = qs[q];
-  const int64_t lo_s = qb[4 * q], hi_s = qb[4 * q + 1];
-  const int64_t dlo_s = qb[4 * q + 2], dhi_s = qb[4 * q + 3];
+  const QKey qstart{Q.start, Q.start_klen, Q.start_ko, qtails};
+  const QKey qend{Q.end, Q.end_klen, Q.end_ko, qtails};
+  // prologue: waves 0-3 resolve the 4 bounds concurrently, each via the
+  // wave-cooperative 64-ary search (blockDim >= 256 always)
+  {
+    int w = threadIdx.x >> 6;
+    int64_t r = -1;
+    if (w == 0) r = d_lb_wave(b, spill, n, qstart, Q.start_rev);
+    else if (w == 1) r = d_lb_wave(b, spill, n, qend, 0);
+    else if (w == 2) r = d_lb_wave(d, spill, dn, qstart, Q.start_rev);
+    else if (w == 3) r = d_lb_wave(d, spill, dn, qend, 0);
+    if (w < 4 && (threadIdx.x & 63) == 0) bounds_s[w] = r;
+  }
+  __syncthreads();
+  const int64_t lo_s = bounds_s[0], hi_s = bounds_s[1];
+  const int64_t dlo_s = bounds_s[2], dhi_s = bounds_s[3];
   const int64_t cap = Q.cap > 0 ? Q.cap : INT64_MAX;
   int64_t scanned = 0;
   int64_t dtotal = 0, btotal = 0;
   uint64_t* outd = Q.count_only ? nullptr : rows_d + (int64_t)q * max_cap;
   uint64_t* outb = Q.count_only ? nullptr : rows_b + (int64_t)q * max_cap;
-  int64_t nB = scan_run_winners(dkeys, dmeta, drev, dlo_s, dhi_s, Q.read_rev,
+  int64_t nB = scan_run_winners(d, spill, dlo_s, dhi_s, Q.read_rev,
                                 cap, outd, max_cap, ROW_TAG_DELTA, nullptr,
-                                nullptr, 0, 0, &dtotal, &scanned, wave_cnt);
+                                0, 0, &dtotal, &scanned, wave_cnt);
   __syncthreads();  // wave_cnt handoff between the two runs
-  int64_t nA = scan_run_winners(bkeys, bmeta, brev, lo_s, hi_s, Q.read_rev, cap,
-                                outb, max_cap, 0, dn ? dkeys : nullptr, drev,
+  int64_t nA = scan_run_winners(b, spill, lo_s, hi_s, Q.read_rev, cap,
+                                outb, max_cap, 0, dn ? &d : nullptr,
                                 dlo_s, dhi_s, &btotal, &scanned, wave_cnt);
   __syncthreads();  // winner lists complete before the merge reads them
   // merge by rank into rows_m (keys are disjoint across the two lists)
@@ -351,12 +349,12 @@ __global__ void k_range_scan2(
       int64_t pos;
       if (j < nA) {
         rt = outb[j];
-        const uint8_t* k = bkeys + (rt & ROW_MASK) * KEYW;
-        pos = j + d_lb_winlist(outd, nB, bkeys, dkeys, k);
+        QKey k = row_qk(b, spill, (int64_t)(rt & ROW_MASK));
+        pos = j + d_lb_winlist(outd, nB, b, d, spill, k);
       } else {
         rt = outd[j - nA];
-        const uint8_t* k = dkeys + (rt & ROW_MASK) * KEYW;
-        pos = (j - nA) + d_lb_winlist(outb, nA, bkeys, dkeys, k);
+        QKey k = row_qk(d, spill, (int64_t)(rt & ROW_MASK));
+        pos = (j - nA) + d_lb_winlist(outb, nA, b, d, spill, k);
       }
       if (pos < cap_m) outm[pos] = rt;
     }
@@ -369,16 +367,9 @@ __global__ void k_range_scan2(
 }
 
 // two-run point read: the delta run wins when it has any row of the key <= R
-__global__ void k_get2(const uint8_t* __restrict__ bkeys,
-                       const uint64_t* __restrict__ bmeta,
-                       const uint64_t* __restrict__ brev,
-                       const uint64_t* __restrict__ bvo,
-                       const uint8_t* __restrict__ dkeys,
-                       const uint64_t* __restrict__ dmeta,
-                       const uint64_t* __restrict__ drev,
-                       const uint64_t* __restrict__ dvo, int64_t n, int64_t dn,
-                       const uint8_t* __restrict__ fkeys,
-                       const uint64_t* __restrict__ frev, int64_t nf,
+__global__ void k_get2(Run b, Run d, int64_t n, int64_t dn,
+                       const uint8_t* __restrict__ spill,
+                       const uint8_t* __restrict__ qtails,
                        const uint8_t* __restrict__ heap,
                        const DevGetQ* __restrict__ qs, int nq, int copy_vals,
                        uint8_t* __restrict__ out, int64_t slot,
@@ -388,40 +379,37 @@ __global__ void k_get2(const uint8_t* __restrict__ bkeys,
   int lane = threadIdx.x & 63;
   if (q >= nq) return;
   const DevGetQ& Q = qs[q];
+  const QKey qk{Q.key, Q.klen, Q.ko, qtails};
   int64_t row = -1;
   int isdelta = 0;
-  if (lane == 0) {
-    if (dn > 0) {
-      int64_t ub = d_upper_bound(dkeys, drev, dn, Q.key, Q.read_rev);
-      if (ub > 0) {
-        int64_t c = ub - 1;
-        if (drev[c] >= 1 && keycmp96(dkeys + c * KEYW, Q.key) == 0) {
-          row = c;
-          isdelta = 1;
-        }
-      }
-    }
-    if (row < 0) {
-      // upper_bound(key,R) == lower_bound(key,R+1) (R < UINT64_MAX-1 here or
-      // saturates safely: rev values are far below UINT64_MAX)
-      uint64_t nr = Q.read_rev == UINT64_MAX ? UINT64_MAX : Q.read_rev + 1;
-      int64_t ub = nr == UINT64_MAX ? d_upper_bound(bkeys, brev, n, Q.key, Q.read_rev)
-                                    : d_lb_fenced(bkeys, brev, n, fkeys, frev, nf, Q.key, nr);
-      if (ub > 0) {
-        int64_t c = ub - 1;
-        if (brev[c] >= 1 && keycmp96(bkeys + c * KEYW, Q.key) == 0) row = c;
+  // upper_bound(key,R) == lower_bound(key,R+1); R+1 saturates safely (real
+  // revisions are far below UINT64_MAX). Wave-cooperative searches: the
+  // whole wave runs each 64-ary probe round in one memory round trip.
+  uint64_t nr = Q.read_rev == UINT64_MAX ? UINT64_MAX : Q.read_rev + 1;
+  if (dn > 0) {
+    int64_t ub = d_lb_wave(d, spill, dn, qk, nr);
+    if (ub > 0) {
+      int64_t c = ub - 1;
+      if (d.rev[c] >= 1 && rowcmp_q(d, spill, c, qk) == 0) {
+        row = c;
+        isdelta = 1;
       }
     }
   }
-  row = __shfl(row, 0);
-  isdelta = __shfl(isdelta, 0);
+  if (row < 0) {
+    int64_t ub = d_lb_wave(b, spill, n, qk, nr);
+    if (ub > 0) {
+      int64_t c = ub - 1;
+      if (b.rev[c] >= 1 && rowcmp_q(b, spill, c, qk) == 0) row = c;
+    }
+  }
   if (row < 0) {
     if (lane == 0) { ofound[q] = 0; oovf[q] = 0; }
     return;
   }
-  const uint64_t* meta = isdelta ? dmeta : bmeta;
-  const uint64_t* rev = isdelta ? drev : brev;
-  const uint64_t* vo = isdelta ? dvo : bvo;
+  const uint64_t* meta = isdelta ? d.meta : b.meta;
+  const uint64_t* rev = isdelta ? d.rev : b.rev;
+  const uint64_t* vo = isdelta ? d.vo : b.vo;
   uint64_t m = meta[row];
   uint32_t vlen = meta_vlen(m);
   if (lane == 0) {
@@ -447,14 +435,7 @@ __device__ __forceinline__ int64_t rec_bytes(uint32_t klen, uint32_t vlen) {
   return 16 + ((klen + 15) & ~15u) + ((vlen + 15) & ~15u);
 }
 
-__global__ void k_gather(const uint8_t* __restrict__ bkeys,
-                         const uint64_t* __restrict__ bmeta,
-                         const uint64_t* __restrict__ brev,
-                         const uint64_t* __restrict__ bvo,
-                         const uint8_t* __restrict__ dkeys,
-                         const uint64_t* __restrict__ dmeta,
-                         const uint64_t* __restrict__ drev,
-                         const uint64_t* __restrict__ dvo,
+__global__ void k_gather(Run b, Run d,
                          const uint8_t* __restrict__ heap,
                          const uint64_t* __restrict__ rows_out, int64_t max_cap,
                          const int64_t* __restrict__ found_out,
@@ -480,7 +461,7 @@ __global__ void k_gather(const uint8_t* __restrict__ bkeys,
     int64_t sz = 0;
     if (j < nwin) {
       uint64_t rt = rows[j];
-      uint64_t m = (rt & ROW_TAG_DELTA ? dmeta : bmeta)[rt & ROW_MASK];
+      uint64_t m = (rt & ROW_TAG_DELTA ? d.meta : b.meta)[rt & ROW_MASK];
       sz = rec_bytes(meta_klen(m), konly ? 0 : meta_vlen(m));
     }
     lds[threadIdx.x] = sz;
@@ -506,14 +487,8 @@ __global__ void k_gather(const uint8_t* __restrict__ bkeys,
 }
 
 // copy winners' records into the arena; launched with 512 threads per query
-__global__ void k_gather_copy(const uint8_t* __restrict__ bkeys,
-                              const uint64_t* __restrict__ bmeta,
-                              const uint64_t* __restrict__ brev,
-                              const uint64_t* __restrict__ bvo,
-                              const uint8_t* __restrict__ dkeys,
-                              const uint64_t* __restrict__ dmeta,
-                              const uint64_t* __restrict__ drev,
-                              const uint64_t* __restrict__ dvo,
+__global__ void k_gather_copy(Run rb, Run rd,
+                              const uint8_t* __restrict__ spill,
                               const uint8_t* __restrict__ heap,
                               const uint64_t* __restrict__ rows_out,
                               int64_t max_cap,
@@ -537,21 +512,28 @@ __global__ void k_gather_copy(const uint8_t* __restrict__ bkeys,
     uint64_t rt = rows[j];
     bool isd = (rt & ROW_TAG_DELTA) != 0;
     int64_t row = (int64_t)(rt & ROW_MASK);
-    uint64_t m = (isd ? dmeta : bmeta)[row];
+    const Run& r = isd ? rd : rb;
+    uint64_t m = r.meta[row];
     uint32_t klen = meta_klen(m), vlen = konly ? 0 : meta_vlen(m);
     uint8_t* dst = qb + qoffs[j];
     if (gl == 0) {
-      *(uint64_t*)dst = (isd ? drev : brev)[row];
+      *(uint64_t*)dst = r.rev[row];
       ((uint32_t*)dst)[2] = klen;
       ((uint32_t*)dst)[3] = vlen;
     }
-    const uint8_t* ks = (isd ? dkeys : bkeys) + row * KEYW;
+    const uint8_t* ks = r.keys + row * KEYW;
     uint8_t* kd = dst + 16;
-    uint32_t kw = (klen + 15) & ~15u;  // key rows are 96B: in-bounds, 4-aligned
+    uint32_t kin = klen > (uint32_t)KEYW ? (uint32_t)KEYW : klen;
+    uint32_t kw = (kin + 3) & ~3u;  // the 96B column part (4-aligned reads)
     for (uint32_t b = gl; b < kw / 4; b += 16)
       ((uint32_t*)kd)[b] = ((const uint32_t*)ks)[b];
+    if (klen > (uint32_t)KEYW) {  // spill tail (keys > 96B)
+      const uint8_t* ts = spill + r.ko[row];
+      for (uint32_t t = gl; t < klen - (uint32_t)KEYW; t += 16)
+        kd[KEYW + t] = ts[t];
+    }
     if (konly) continue;
-    const uint8_t* vs = heap + (isd ? dvo : bvo)[row];
+    const uint8_t* vs = heap + r.vo[row];
     uint8_t* vd = dst + 16 + ((klen + 15) & ~15u);
     uint32_t w16 = vlen >> 4;
     for (uint32_t b = gl; b < w16; b += 16)
@@ -614,11 +596,17 @@ __global__ void k_fill_u64(uint64_t* a, uint64_t v, int64_t n) {
   if (i < n) a[i] = v;
 }
 
-__global__ void k_find_bounds(const uint8_t* keys, const uint64_t* rev, int64_t n,
+__global__ void k_find_bounds(Run a, int64_t n, const uint8_t* spill,
                               const uint8_t* bkeys, const uint64_t* brevs, int nb,
                               int64_t* out) {
   int i = blockIdx.x * blockDim.x + threadIdx.x;
-  if (i < nb) out[i] = d_lower_bound(keys, rev, n, bkeys + (int64_t)i * KEYW, brevs[i]);
+  // borders are <= KEYW (validated host-side); len=KEYW orders a border
+  // before any longer key sharing its full 96B prefix, which is exactly
+  // lower-bound semantics
+  if (i < nb)
+    out[i] = d_lb_range(a, spill, 0, n,
+                        QKey{bkeys + (int64_t)i * KEYW, (uint32_t)KEYW, 0, nullptr},
+                        brevs[i]);
 }
 
 __global__ void k_compact_mark(const uint64_t* __restrict__ meta,
@@ -659,23 +647,55 @@ __global__ void k_compact_scatter(const uint8_t* __restrict__ keysA,
                                   const uint64_t* __restrict__ metaA,
                                   const uint64_t* __restrict__ revA,
                                   const uint64_t* __restrict__ voA,
+                                  const uint64_t* __restrict__ koA,
                                   const uint64_t* __restrict__ keep,
                                   const uint64_t* __restrict__ nidx,
                                   const uint64_t* __restrict__ heap_off,
+                                  const uint64_t* __restrict__ spill_off,
                                   uint8_t* __restrict__ keysB,
                                   uint64_t* __restrict__ metaB,
                                   uint64_t* __restrict__ revB,
-                                  uint64_t* __restrict__ voB, int64_t n) {
+                                  uint64_t* __restrict__ voB,
+                                  uint64_t* __restrict__ koB, int64_t n) {
   int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
   if (i >= n || !keep[i]) return;
   int64_t j = (int64_t)nidx[i];
   revB[j] = revA[i];
   metaB[j] = metaA[i];
   voB[j] = revA[i] == 0 ? voA[i] : heap_off[i];
+  koB[j] = meta_klen(metaA[i]) > (uint32_t)KEYW ? spill_off[i] : 0;
   const uint64_t* ks = (const uint64_t*)(keysA + i * KEYW);
   uint64_t* kd = (uint64_t*)(keysB + j * KEYW);
 #pragma unroll
   for (int k = 0; k < KEYW / 8; ++k) kd[k] = ks[k];
+}
+
+// key-spill compaction (keys > KEYW): surviving tails are packed into the
+// fresh spill heap, mirroring the value-heap sweep
+__global__ void k_spill_sizes(const uint64_t* __restrict__ keep,
+                              const uint64_t* __restrict__ meta,
+                              uint64_t* __restrict__ sz, int64_t n) {
+  int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  if (i >= n) return;
+  uint32_t kl = meta_klen(meta[i]);
+  sz[i] = (keep[i] && kl > (uint32_t)KEYW) ? (((uint64_t)kl - KEYW + 15) & ~15ull) : 0;
+}
+
+__global__ void k_spill_scatter(const uint64_t* __restrict__ keep,
+                                const uint64_t* __restrict__ meta,
+                                const uint64_t* __restrict__ koA,
+                                const uint64_t* __restrict__ spill_off,
+                                const uint8_t* __restrict__ spillA,
+                                uint8_t* __restrict__ spillB, int64_t n) {
+  int64_t i = (int64_t)blockIdx.x * (blockDim.x / 64) + (threadIdx.x >> 6);
+  int lane = threadIdx.x & 63;
+  if (i >= n || !keep[i]) return;
+  uint32_t kl = meta_klen(meta[i]);
+  if (kl <= (uint32_t)KEYW) return;
+  uint32_t t = kl - (uint32_t)KEYW;
+  const uint8_t* src = spillA + koA[i];
+  uint8_t* dst = spillB + spill_off[i];
+  for (uint32_t x = lane; x < t; x += 64) dst[x] = src[x];
 }
 
 __global__ void k_heap_scatter(const uint64_t* __restrict__ keep,
@@ -695,55 +715,55 @@ __global__ void k_heap_scatter(const uint64_t* __restrict__ keep,
 }
 
 // ---- merge (memtable flush): delta ranks + scatter ----------------------
-__global__ void k_merge_rank(const uint8_t* __restrict__ keysA,
-                             const uint64_t* __restrict__ revA, int64_t n,
-                             const uint8_t* __restrict__ keysD,
-                             const uint64_t* __restrict__ revD, int64_t m,
+__global__ void k_merge_rank(Run a, int64_t n, Run dnew, int64_t m,
+                             const uint8_t* __restrict__ spill,
                              uint64_t* __restrict__ rank,
                              uint64_t* __restrict__ drop) {
   int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
   if (i >= n) return;
-  const uint8_t* k = keysA + i * KEYW;
-  uint64_t r = revA[i];
-  int64_t lb = d_lower_bound(keysD, revD, m, k, r);
+  QKey k = row_qk(a, spill, i);
+  uint64_t r = a.rev[i];
+  int64_t lb = d_lb_range(dnew, spill, 0, m, k, r);
   rank[i] = (uint64_t)lb;
   // delta rev-rows REPLACE base rev-rows of the same key (memtable holds the
   // updated revision-row value; equal internal key only happens at rev==0)
-  bool eq = lb < m && revD[lb] == r && keycmp96(keysD + lb * KEYW, k) == 0;
+  bool eq = lb < m && dnew.rev[lb] == r && rowcmp_q(dnew, spill, lb, k) == 0;
   drop[i] = eq ? 1 : 0;
 }
 
 __global__ void k_merge_scatter_base(
-    const uint8_t* keysA, const uint64_t* metaA, const uint64_t* revA,
-    const uint64_t* voA, const uint64_t* __restrict__ rank,
+    Run a, const uint64_t* __restrict__ rank,
     const uint64_t* __restrict__ drop, const uint64_t* __restrict__ dropx,
-    uint8_t* keysB, uint64_t* metaB, uint64_t* revB, uint64_t* voB, int64_t n) {
+    uint8_t* keysB, uint64_t* metaB, uint64_t* revB, uint64_t* voB,
+    uint64_t* koB, int64_t n) {
   int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
   if (i >= n || drop[i]) return;
   int64_t j = i - (int64_t)dropx[i] + (int64_t)rank[i];
-  revB[j] = revA[i];
-  metaB[j] = metaA[i];
-  voB[j] = voA[i];
-  const uint64_t* ks = (const uint64_t*)(keysA + i * KEYW);
+  revB[j] = a.rev[i];
+  metaB[j] = a.meta[i];
+  voB[j] = a.vo[i];
+  koB[j] = a.ko[i];
+  const uint64_t* ks = (const uint64_t*)(a.keys + i * KEYW);
   uint64_t* kd = (uint64_t*)(keysB + j * KEYW);
 #pragma unroll
   for (int k = 0; k < KEYW / 8; ++k) kd[k] = ks[k];
 }
 
 __global__ void k_merge_scatter_delta(
-    const uint8_t* keysA, const uint64_t* revA, int64_t n,
-    const uint8_t* keysD, const uint64_t* metaD, const uint64_t* revD,
-    const uint64_t* voD, int64_t m, const uint64_t* __restrict__ dropx,
-    uint8_t* keysB, uint64_t* metaB, uint64_t* revB, uint64_t* voB) {
+    Run a, int64_t n, Run dnew, int64_t m, const uint8_t* __restrict__ spill,
+    const uint64_t* __restrict__ dropx,
+    uint8_t* keysB, uint64_t* metaB, uint64_t* revB, uint64_t* voB,
+    uint64_t* koB) {
   int64_t j = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
   if (j >= m) return;
-  const uint8_t* k = keysD + j * KEYW;
-  int64_t lb = d_lower_bound(keysA, revA, n, k, revD[j]);
+  QKey k = row_qk(dnew, spill, j);
+  int64_t lb = d_lb_range(a, spill, 0, n, k, dnew.rev[j]);
   int64_t out = lb - (int64_t)dropx[lb] + j;
-  revB[out] = revD[j];
-  metaB[out] = metaD[j];
-  voB[out] = voD[j];
-  const uint64_t* ks = (const uint64_t*)k;
+  revB[out] = dnew.rev[j];
+  metaB[out] = dnew.meta[j];
+  voB[out] = dnew.vo[j];
+  koB[out] = dnew.ko[j];
+  const uint64_t* ks = (const uint64_t*)(dnew.keys + j * KEYW);
   uint64_t* kd = (uint64_t*)(keysB + out * KEYW);
 #pragma unroll
   for (int kk = 0; kk < KEYW / 8; ++kk) kd[kk] = ks[kk];
@@ -752,18 +772,13 @@ __global__ void k_merge_scatter_delta(
 // single-workgroup merge for small runs (n+m <= 8192): rank, drop-scan,
 // scatter and same_next in ONE launch — the multi-kernel path costs ~10
 // launches plus two host syncs, which dominates small per-step delta merges.
-__global__ void k_merge_small(const uint8_t* __restrict__ skeys,
-                              const uint64_t* __restrict__ smeta,
-                              const uint64_t* __restrict__ srev,
-                              const uint64_t* __restrict__ svo, int64_t n,
-                              const uint8_t* __restrict__ nk,
-                              const uint64_t* __restrict__ nm,
-                              const uint64_t* __restrict__ nr,
-                              const uint64_t* __restrict__ nv, int64_t m,
+__global__ void k_merge_small(Run src, int64_t n, Run dnew, int64_t m,
+                              const uint8_t* __restrict__ spill,
                               uint8_t* __restrict__ okeys,
                               uint64_t* __restrict__ ometa,
                               uint64_t* __restrict__ orev,
                               uint64_t* __restrict__ ovo,
+                              uint64_t* __restrict__ oko,
                               int64_t* __restrict__ out_n) {
   __shared__ uint16_t dropx[8193];  // exclusive drop counts (n <= 8192)
   __shared__ uint16_t partial[257];
@@ -775,9 +790,10 @@ __global__ void k_merge_small(const uint8_t* __restrict__ skeys,
   int i1 = min((int64_t)i0 + chunk, n);
   uint16_t cnt = 0;
   for (int i = i0; i < i1; ++i) {
-    int64_t lb = d_lb_range(nk, nr, 0, m, skeys + (int64_t)i * KEYW, srev[i]);
-    bool drop = lb < m && nr[lb] == srev[i] &&
-                keycmp96(nk + lb * KEYW, skeys + (int64_t)i * KEYW) == 0;
+    QKey k = row_qk(src, spill, i);
+    int64_t lb = d_lb_range(dnew, spill, 0, m, k, src.rev[i]);
+    bool drop = lb < m && dnew.rev[lb] == src.rev[i] &&
+                rowcmp_q(dnew, spill, lb, k) == 0;
     dropx[i] = drop ? 1 : 0;
     cnt += drop;
   }
@@ -800,44 +816,41 @@ __global__ void k_merge_small(const uint8_t* __restrict__ skeys,
   if (t == 0) *out_n = new_n;
   // scatter src rows
   for (int64_t i = t; i < n; i += T) {
-    int64_t lb = d_lb_range(nk, nr, 0, m, skeys + i * KEYW, srev[i]);
-    bool drop = lb < m && nr[lb] == srev[i] &&
-                keycmp96(nk + lb * KEYW, skeys + i * KEYW) == 0;
+    QKey k = row_qk(src, spill, i);
+    int64_t lb = d_lb_range(dnew, spill, 0, m, k, src.rev[i]);
+    bool drop = lb < m && dnew.rev[lb] == src.rev[i] &&
+                rowcmp_q(dnew, spill, lb, k) == 0;
     if (drop) continue;
     int64_t j = i - (int64_t)dropx[i] + lb;
-    orev[j] = srev[i];
-    ometa[j] = smeta[i];
-    ovo[j] = svo[i];
-    const uint64_t* ks = (const uint64_t*)(skeys + i * KEYW);
+    orev[j] = src.rev[i];
+    ometa[j] = src.meta[i];
+    ovo[j] = src.vo[i];
+    oko[j] = src.ko[i];
+    const uint64_t* ks = (const uint64_t*)(src.keys + i * KEYW);
     uint64_t* kd = (uint64_t*)(okeys + j * KEYW);
 #pragma unroll
-    for (int k = 0; k < KEYW / 8; ++k) kd[k] = ks[k];
+    for (int k2 = 0; k2 < KEYW / 8; ++k2) kd[k2] = ks[k2];
   }
   // scatter new rows
   for (int64_t j = t; j < m; j += T) {
-    int64_t lb = d_lb_range(skeys, srev, 0, n, nk + j * KEYW, nr[j]);
+    QKey k = row_qk(dnew, spill, j);
+    int64_t lb = d_lb_range(src, spill, 0, n, k, dnew.rev[j]);
     int64_t dpx = lb < n ? dropx[lb] : (int64_t)dropped;
     int64_t o = lb - dpx + j;
-    orev[o] = nr[j];
-    ometa[o] = nm[j];
-    ovo[o] = nv[j];
-    const uint64_t* ks = (const uint64_t*)(nk + j * KEYW);
+    orev[o] = dnew.rev[j];
+    ometa[o] = dnew.meta[j];
+    ovo[o] = dnew.vo[j];
+    oko[o] = dnew.ko[j];
+    const uint64_t* ks = (const uint64_t*)(dnew.keys + j * KEYW);
     uint64_t* kd = (uint64_t*)(okeys + o * KEYW);
 #pragma unroll
-    for (int k = 0; k < KEYW / 8; ++k) kd[k] = ks[k];
+    for (int k2 = 0; k2 < KEYW / 8; ++k2) kd[k2] = ks[k2];
   }
   __syncthreads();
-  // same_next over the merged rows
+  // same_next over the merged rows (full-key equality incl. spill tails)
   for (int64_t i = t; i < new_n; i += T) {
-    bool same = false;
-    if (i + 1 < new_n) {
-      const uint64_t* a = (const uint64_t*)(okeys + i * KEYW);
-      const uint64_t* b = (const uint64_t*)(okeys + (i + 1) * KEYW);
-      same = true;
-#pragma unroll
-      for (int k = 0; k < KEYW / 8; ++k)
-        if (a[k] != b[k]) { same = false; break; }
-    }
+    bool same = i + 1 < new_n &&
+                rows_same_key(okeys, ometa, oko, spill, i, i + 1);
     ometa[i] = same ? (ometa[i] | M_SAME_NEXT) : (ometa[i] & ~M_SAME_NEXT);
   }
 }
@@ -892,15 +905,39 @@ struct Slab::Impl {
   int64_t max_rows = 0, heap_cap = 0;
   int64_t n = 0, heap_used_ = 0;
 
-  struct Col { uint8_t* keys = nullptr; uint64_t *meta = nullptr, *rev = nullptr, *vo = nullptr; };
+  struct Col {
+    uint8_t* keys = nullptr;
+    uint64_t *meta = nullptr, *rev = nullptr, *vo = nullptr, *ko = nullptr;
+    Run run() const { return Run{keys, meta, rev, vo, ko}; }
+  };
   Col A, B;
   Col DA, DB;            // delta-run ping-pong
   int64_t dn = 0;        // delta-run rows
   int64_t delta_cap = 0; // rows per delta buffer
   uint8_t *heapA = nullptr, *heapB = nullptr;
+  // key-spill heap (tails of keys > KEYW), shared by base+delta; compacted
+  // with the rows in Compact
+  uint8_t *spillA = nullptr, *spillB = nullptr;
+  int64_t spill_cap = 0, spill_used_ = 0;
+  // per-batch query-tail buffer (bounds/get keys > KEYW)
+  uint8_t* d_qtails = nullptr;
+  int64_t qt_cap = 0;
+  bool ensure_qtails(const std::string& qt, std::string* err) {
+    if ((int64_t)qt.size() > qt_cap) {
+      if (d_qtails) (void)hipFree(d_qtails);
+      int64_t cap = (int64_t)qt.size() * 2 + 4096;
+      HIP_CHECK(hipMalloc(&d_qtails, cap));
+      qt_cap = cap;
+    }
+    if (!qt.empty())
+      HIP_CHECK(hipMemcpyAsync(d_qtails, qt.data(), qt.size(),
+                               hipMemcpyHostToDevice, stream));
+    return true;
+  }
 
   // scan scratch (u64, shared across ops)
-  uint64_t *s_a = nullptr, *s_b = nullptr, *s_c = nullptr, *s_d = nullptr;  // max_rows+2
+  uint64_t *s_a = nullptr, *s_b = nullptr, *s_c = nullptr, *s_d = nullptr,
+           *s_e = nullptr;  // max_rows+2
   uint64_t *lv1 = nullptr, *lv1o = nullptr, *lv2 = nullptr, *lv2o = nullptr,
            *lv3 = nullptr, *lv3o = nullptr;
 
@@ -911,9 +948,6 @@ struct Slab::Impl {
   int64_t arena_bytes = 384ll << 20;
   DevRangeQ* d_qs = nullptr;
   DevGetQ* d_gq = nullptr;
-  uint8_t* d_fkeys = nullptr;   // fence index over the base run
-  uint64_t* d_frev = nullptr;
-  int64_t nf = 0;
   uint64_t* d_rows = nullptr;   // max_q*max_cap (base winners)
   uint64_t* d_rows2 = nullptr;  // delta winners
   uint64_t* d_rowsm = nullptr;  // merged winners
@@ -926,13 +960,13 @@ struct Slab::Impl {
   uint8_t* d_pack = nullptr;    // arena
   int64_t* d_goffs = nullptr;   // max_q+1
   int64_t* d_bounds = nullptr;  // compact bounds
-  int64_t* d_qb = nullptr;      // range query bounds [4*max_q]
   uint8_t* d_bkeys = nullptr;
   uint64_t* d_brevs = nullptr;
 
   // delta upload scratch (grown on demand)
   uint8_t* d_dkeys = nullptr;
-  uint64_t *d_dmeta = nullptr, *d_drev = nullptr, *d_dvo = nullptr;
+  uint64_t *d_dmeta = nullptr, *d_drev = nullptr, *d_dvo = nullptr,
+           *d_dko = nullptr;
   int64_t upload_cap = 0;
 
   // watcher table
@@ -1019,17 +1053,20 @@ struct Slab::Impl {
                     (void*)B.keys, (void*)B.meta, (void*)B.rev, (void*)B.vo,
                     (void*)DA.keys, (void*)DA.meta, (void*)DA.rev, (void*)DA.vo,
                     (void*)DB.keys, (void*)DB.meta, (void*)DB.rev, (void*)DB.vo,
-                    (void*)d_rows2, (void*)d_rowsm, (void*)d_fkeys, (void*)d_frev,
+                    (void*)d_rows2, (void*)d_rowsm,
                     (void*)heapA, (void*)heapB, (void*)s_a, (void*)s_b,
-                    (void*)s_c, (void*)s_d, (void*)lv1, (void*)lv1o, (void*)lv2,
+                    (void*)s_c, (void*)s_d, (void*)s_e, (void*)lv1, (void*)lv1o, (void*)lv2,
                     (void*)lv2o, (void*)lv3, (void*)lv3o, (void*)d_qs,
                     (void*)d_gq, (void*)d_rows, (void*)d_offs, (void*)d_found,
                     (void*)d_total, (void*)d_gbytes, (void*)d_ovf,
                     (void*)d_found32, (void*)d_orev, (void*)d_ometa,
                     (void*)d_scanned, (void*)d_bytes, (void*)d_gbuf,
-                    (void*)d_pack, (void*)d_goffs, (void*)d_bounds, (void*)d_qb,
+                    (void*)d_pack, (void*)d_goffs, (void*)d_bounds,
                     (void*)d_bkeys, (void*)d_brevs, (void*)d_dkeys,
-                    (void*)d_dmeta, (void*)d_drev, (void*)d_dvo, (void*)d_wpfx,
+                    (void*)d_dmeta, (void*)d_drev, (void*)d_dvo, (void*)d_dko,
+                    (void*)spillA, (void*)spillB, (void*)d_qtails,
+                    (void*)A.ko, (void*)B.ko, (void*)DA.ko, (void*)DB.ko,
+                    (void*)d_wpfx,
                     (void*)d_wplen, (void*)d_wlive, (void*)d_wfrom,
                     (void*)d_ekeys, (void*)d_erev, (void*)d_bitmap}) {
       if (p) (void)hipFree(p);
@@ -1089,34 +1126,27 @@ struct Slab::Impl {
   bool ensure_delta(int64_t m, std::string* err) {
     if (m <= upload_cap) return true;
     int64_t cap = m + m / 2 + 1024;
-    for (void* p : {(void*)d_dkeys, (void*)d_dmeta, (void*)d_drev, (void*)d_dvo})
+    for (void* p : {(void*)d_dkeys, (void*)d_dmeta, (void*)d_drev,
+                    (void*)d_dvo, (void*)d_dko})
       if (p) (void)hipFree(p);
     HIP_CHECK(hipMalloc(&d_dkeys, cap * KEYW));
     HIP_CHECK(hipMalloc(&d_dmeta, cap * 8));
     HIP_CHECK(hipMalloc(&d_drev, cap * 8));
     HIP_CHECK(hipMalloc(&d_dvo, cap * 8));
+    HIP_CHECK(hipMalloc(&d_dko, cap * 8));
     upload_cap = cap;
     return true;
   }
 
-  void rebuildFence(const Col& base, int64_t n) {
-    nf = n / FENCE_STRIDE + (n % FENCE_STRIDE ? 1 : 0);
-    if (nf == 0) return;
-    hipLaunchKernelGGL(k_build_fence, dim3((uint32_t)ceil_div(nf, 256)),
-                       dim3(256), 0, stream, base.keys, base.rev, n, d_fkeys,
-                       d_frev, nf);
-  }
-
   // generic two-run merge: src (n rows) + newer (m rows, device arrays,
   // rev-rows replace) -> dst; recomputes same_next. dst must hold n+m rows.
-  bool mergeRuns(const Col& src, int64_t n, const uint8_t* nk,
-                 const uint64_t* nm, const uint64_t* nr, const uint64_t* nv,
+  bool mergeRuns(const Col& src, int64_t n, const Run& dnew,
                  int64_t m, Col& dst, int64_t* out_n, std::string* err,
                  bool device_newn_ok = false) {
     if (n + m <= 2048) {  // single-launch path only where one CU wins
       hipLaunchKernelGGL(k_merge_small, dim3(1), dim3(256), 0, stream,
-                         src.keys, src.meta, src.rev, src.vo, n, nk, nm, nr,
-                         nv, m, dst.keys, dst.meta, dst.rev, dst.vo,
+                         src.run(), n, dnew, m, spillA,
+                         dst.keys, dst.meta, dst.rev, dst.vo, dst.ko,
                          (int64_t*)s_d);
       if (device_newn_ok) return true;  // caller knows new_n; stay async
       int64_t nn = 0;
@@ -1128,20 +1158,20 @@ struct Slab::Impl {
     if (n > 0) {
       int64_t nb = ceil_div(n, 256);
       hipLaunchKernelGGL(k_merge_rank, dim3((uint32_t)nb), dim3(256), 0, stream,
-                         src.keys, src.rev, n, nk, nr, m, s_a, s_b);
+                         src.run(), n, dnew, m, spillA, s_a, s_b);
       HIP_CHECK(hipMemsetAsync(s_b + n, 0, 8, stream));
       if (!scan(s_b, s_c, n + 1, nullptr, err)) return false;  // no host sync
       hipLaunchKernelGGL(k_merge_scatter_base, dim3((uint32_t)nb), dim3(256), 0,
-                         stream, src.keys, src.meta, src.rev, src.vo, s_a, s_b,
-                         s_c, dst.keys, dst.meta, dst.rev, dst.vo, n);
+                         stream, src.run(), s_a, s_b,
+                         s_c, dst.keys, dst.meta, dst.rev, dst.vo, dst.ko, n);
     } else {
       HIP_CHECK(hipMemsetAsync(s_c, 0, 8, stream));
     }
     if (m > 0) {
       int64_t mb = ceil_div(m, 256);
       hipLaunchKernelGGL(k_merge_scatter_delta, dim3((uint32_t)mb), dim3(256),
-                         0, stream, src.keys, src.rev, n, nk, nm, nr, nv, m,
-                         s_c, dst.keys, dst.meta, dst.rev, dst.vo);
+                         0, stream, src.run(), n, dnew, m, spillA,
+                         s_c, dst.keys, dst.meta, dst.rev, dst.vo, dst.ko);
     }
     // new_n on device (s_d[0]) to avoid an extra host round trip; same_next
     // must cover the full possible extent (n+m), which is safe: rows beyond
@@ -1162,7 +1192,7 @@ struct Slab::Impl {
     if (new_n > 0) {
       int64_t sb = ceil_div(new_n, 256);
       hipLaunchKernelGGL(k_same_next, dim3((uint32_t)sb), dim3(256), 0, stream,
-                         dst.keys, dst.meta, new_n);
+                         dst.keys, dst.meta, dst.ko, spillA, new_n);
     }
     *out_n = new_n;
     return true;
@@ -1213,7 +1243,11 @@ Slab* Slab::Create(int64_t max_rows, int64_t heap_cap, int device,
     HIP_CHECK_NULL(hipMalloc(&c->meta, max_rows * 8));
     HIP_CHECK_NULL(hipMalloc(&c->rev, max_rows * 8));
     HIP_CHECK_NULL(hipMalloc(&c->vo, max_rows * 8));
+    HIP_CHECK_NULL(hipMalloc(&c->ko, max_rows * 8));
   }
+  I->spill_cap = env_i64("KB_SPILL_BYTES", 256ll << 20);
+  HIP_CHECK_NULL(hipMalloc(&I->spillA, I->spill_cap));
+  HIP_CHECK_NULL(hipMalloc(&I->spillB, I->spill_cap));
   I->delta_cap = env_i64("KB_DELTA_CAP", 1 << 19);
   if (I->delta_cap > max_rows) I->delta_cap = max_rows;
   for (Impl::Col* c : {&I->DA, &I->DB}) {
@@ -1221,6 +1255,7 @@ Slab* Slab::Create(int64_t max_rows, int64_t heap_cap, int device,
     HIP_CHECK_NULL(hipMalloc(&c->meta, I->delta_cap * 8));
     HIP_CHECK_NULL(hipMalloc(&c->rev, I->delta_cap * 8));
     HIP_CHECK_NULL(hipMalloc(&c->vo, I->delta_cap * 8));
+    HIP_CHECK_NULL(hipMalloc(&c->ko, I->delta_cap * 8));
   }
   HIP_CHECK_NULL(hipMalloc(&I->heapA, heap_cap));
   HIP_CHECK_NULL(hipMalloc(&I->heapB, heap_cap));
@@ -1228,6 +1263,7 @@ Slab* Slab::Create(int64_t max_rows, int64_t heap_cap, int device,
   HIP_CHECK_NULL(hipMalloc(&I->s_b, (max_rows + 2) * 8));
   HIP_CHECK_NULL(hipMalloc(&I->s_c, (max_rows + 2) * 8));
   HIP_CHECK_NULL(hipMalloc(&I->s_d, (max_rows + 2) * 8));
+  HIP_CHECK_NULL(hipMalloc(&I->s_e, (max_rows + 2) * 8));
   int64_t nb1 = ceil_div(max_rows + 2, 256) + 1;
   int64_t nb2 = ceil_div(nb1, 256) + 1;
   HIP_CHECK_NULL(hipMalloc(&I->lv1, nb1 * 8));
@@ -1239,8 +1275,6 @@ Slab* Slab::Create(int64_t max_rows, int64_t heap_cap, int device,
   HIP_CHECK_NULL(hipMalloc(&I->d_qs, sizeof(DevRangeQ) * I->max_q));
   HIP_CHECK_NULL(hipMalloc(&I->d_gq, sizeof(DevGetQ) * I->max_q));
   HIP_CHECK_NULL(hipMalloc(&I->d_rows, (int64_t)I->max_q * I->max_cap * 8));
-  HIP_CHECK_NULL(hipMalloc(&I->d_fkeys, (max_rows / FENCE_STRIDE + 2) * KEYW));
-  HIP_CHECK_NULL(hipMalloc(&I->d_frev, (max_rows / FENCE_STRIDE + 2) * 8));
   HIP_CHECK_NULL(hipMalloc(&I->d_rows2, (int64_t)I->max_q * I->max_cap * 8));
   HIP_CHECK_NULL(hipMalloc(&I->d_rowsm, (int64_t)I->max_q * I->max_cap * 8));
   HIP_CHECK_NULL(hipMalloc(&I->d_offs, (int64_t)I->max_q * I->max_cap * 8));
@@ -1257,7 +1291,6 @@ Slab* Slab::Create(int64_t max_rows, int64_t heap_cap, int device,
   HIP_CHECK_NULL(hipMalloc(&I->d_pack, I->arena_bytes));
   HIP_CHECK_NULL(hipMalloc(&I->d_goffs, (I->max_q + 1) * 8));
   HIP_CHECK_NULL(hipMalloc(&I->d_bounds, 256 * 8));
-  HIP_CHECK_NULL(hipMalloc(&I->d_qb, (int64_t)I->max_q * 4 * 8));
   HIP_CHECK_NULL(hipMalloc(&I->d_bkeys, 256 * KEYW));
   HIP_CHECK_NULL(hipMalloc(&I->d_brevs, 256 * 8));
   (void)fail;
@@ -1270,6 +1303,26 @@ int64_t Slab::delta_rows() const { return p->dn; }
 int64_t Slab::delta_capacity() const { return p->delta_cap; }
 int64_t Slab::heap_used() const { return p->heap_used_; }
 int64_t Slab::max_winner_cap() const { return p->max_cap; }
+
+int64_t Slab::spill_used() const { return p->spill_used_; }
+
+bool Slab::SpillAppend(const void* src, int64_t len, int64_t* off,
+                       std::string* err) {
+  Impl* I = p;
+  if (len == 0) { *off = I->spill_used_; return true; }
+  if (I->spill_used_ + len > I->spill_cap) {
+    if (err) *err = "key-spill heap full (KB_SPILL_BYTES)";
+    return false;
+  }
+  HIP_CHECK(hipMemcpyAsync(I->spillA + I->spill_used_, src, len,
+                           hipMemcpyHostToDevice, I->stream));
+  HIP_CHECK(hipMemcpyAsync(I->spillB + I->spill_used_, src, len,
+                           hipMemcpyHostToDevice, I->stream));
+  HIP_CHECK(hipStreamSynchronize(I->stream));
+  *off = I->spill_used_;
+  I->spill_used_ += len;
+  return true;
+}
 
 bool Slab::HeapAppend(const void* src, int64_t len, int64_t* off, std::string* err) {
   Impl* I = p;
@@ -1289,7 +1342,8 @@ bool Slab::HeapAppend(const void* src, int64_t len, int64_t* off, std::string* e
 }
 
 bool Slab::AppendRows(const uint8_t* keys, const uint64_t* meta,
-                      const uint64_t* rev, const uint64_t* vo, int64_t m,
+                      const uint64_t* rev, const uint64_t* vo,
+                      const uint64_t* ko, int64_t m,
                       std::string* err, int64_t known_new_dn) {
   Impl* I = p;
   if (m == 0) return true;
@@ -1308,9 +1362,11 @@ bool Slab::AppendRows(const uint8_t* keys, const uint64_t* meta,
   HIP_CHECK(hipMemcpyAsync(I->d_dmeta, meta, m * 8, hipMemcpyHostToDevice, I->stream));
   HIP_CHECK(hipMemcpyAsync(I->d_drev, rev, m * 8, hipMemcpyHostToDevice, I->stream));
   HIP_CHECK(hipMemcpyAsync(I->d_dvo, vo, m * 8, hipMemcpyHostToDevice, I->stream));
+  HIP_CHECK(hipMemcpyAsync(I->d_dko, ko, m * 8, hipMemcpyHostToDevice, I->stream));
   int64_t new_dn = async ? known_new_dn : 0;  // async: mergeRuns reads it
-  if (!I->mergeRuns(I->DA, I->dn, I->d_dkeys, I->d_dmeta, I->d_drev, I->d_dvo,
-                    m, I->DB, &new_dn, err, /*device_newn_ok=*/async))
+  Run up{I->d_dkeys, I->d_dmeta, I->d_drev, I->d_dvo, I->d_dko};
+  if (!I->mergeRuns(I->DA, I->dn, up, m, I->DB, &new_dn, err,
+                    /*device_newn_ok=*/async))
     return false;
   if (async) {
     // kernels queue on the stream; subsequent reads queue behind them.
@@ -1343,8 +1399,7 @@ bool Slab::Fold(std::string* err) {
   if (I->n + I->dn > I->max_rows) { if (err) *err = "slab full (KB_MAX_ROWS)"; return false; }
   HIP_CHECK(hipEventRecord(I->ev2, I->stream));
   int64_t new_n = 0;
-  if (!I->mergeRuns(I->A, I->n, I->DA.keys, I->DA.meta, I->DA.rev, I->DA.vo,
-                    I->dn, I->B, &new_n, err))
+  if (!I->mergeRuns(I->A, I->n, I->DA.run(), I->dn, I->B, &new_n, err))
     return false;
   HIP_CHECK(hipEventRecord(I->ev3, I->stream));
   HIP_CHECK(hipStreamSynchronize(I->stream));
@@ -1355,7 +1410,6 @@ bool Slab::Fold(std::string* err) {
   std::swap(I->A, I->B);
   I->n = new_n;
   I->dn = 0;
-  I->rebuildFence(I->A, I->n);
   HIP_CHECK(hipStreamSynchronize(I->stream));
   return true;
 }
@@ -1387,13 +1441,18 @@ bool Slab::Merge(const DeltaRows& d, std::string* err) {
                              hipMemcpyHostToDevice, I->stream));
     HIP_CHECK(hipMemcpyAsync(I->d_dvo, d.vo.data(), d.m * 8,
                              hipMemcpyHostToDevice, I->stream));
+    if ((int64_t)d.ko.size() == d.m) {
+      HIP_CHECK(hipMemcpyAsync(I->d_dko, d.ko.data(), d.m * 8,
+                               hipMemcpyHostToDevice, I->stream));
+    } else {
+      HIP_CHECK(hipMemsetAsync(I->d_dko, 0, d.m * 8, I->stream));
+    }
     int64_t new_n = 0;
-    if (!I->mergeRuns(I->A, I->n, I->d_dkeys, I->d_dmeta, I->d_drev, I->d_dvo,
-                      d.m, I->B, &new_n, err))
+    Run up{I->d_dkeys, I->d_dmeta, I->d_drev, I->d_dvo, I->d_dko};
+    if (!I->mergeRuns(I->A, I->n, up, d.m, I->B, &new_n, err))
       return false;
     std::swap(I->A, I->B);
     I->n = new_n;
-    I->rebuildFence(I->A, I->n);
   }
   HIP_CHECK(hipEventRecord(I->ev1, I->stream));
   HIP_CHECK(hipStreamSynchronize(I->stream));
@@ -1405,40 +1464,37 @@ bool Slab::Merge(const DeltaRows& d, std::string* err) {
 }
 
 bool Slab::RangeBatch(const std::vector<DevRangeQ>& qs, bool d2h,
-                      std::vector<RangeResult>* outs, std::string* err) {
-  return RangeBatchEx(qs, d2h, true, outs, err);
+                      std::vector<RangeResult>* outs, std::string* err,
+                      const std::string& qtails) {
+  return RangeBatchEx(qs, d2h, true, outs, err, qtails);
 }
 
-bool Slab::RangeBatchStart(const std::vector<DevRangeQ>& qs, std::string* err) {
+bool Slab::RangeBatchStart(const std::vector<DevRangeQ>& qs, std::string* err,
+                           const std::string& qtails) {
   Impl* I = p;
   int nq = (int)qs.size();
   if (nq == 0) return true;
   if (nq > I->max_q) { if (err) *err = "too many queries per batch (KB_MAX_Q)"; return false; }
   int64_t qcap = I->arena_bytes / nq;
   qcap &= ~15ll;
+  if (!I->ensure_qtails(qtails, err)) return false;
   HIP_CHECK(hipMemcpyAsync(I->d_qs, qs.data(), sizeof(DevRangeQ) * nq,
                            hipMemcpyHostToDevice, I->stream));
   HIP_CHECK(hipMemsetAsync(I->d_scanned, 0, 8, I->stream));
   HIP_CHECK(hipMemsetAsync(I->d_bytes, 0, 8, I->stream));
   HIP_CHECK(hipEventRecord(I->ev0, I->stream));
-  hipLaunchKernelGGL(k_range_bounds, dim3((uint32_t)ceil_div(4 * nq, 256)),
-                     dim3(256), 0, I->stream, I->A.keys, I->A.rev, I->n,
-                     I->DA.keys, I->DA.rev, I->dn, I->d_fkeys, I->d_frev,
-                     I->nf, I->d_qs, nq, I->d_qb);
   hipLaunchKernelGGL(k_range_scan2, dim3(nq), dim3((uint32_t)I->scan_t), 0, I->stream,
-                     I->A.keys, I->A.meta, I->A.rev, I->n, I->DA.keys,
-                     I->DA.meta, I->DA.rev, I->dn, I->d_qb, I->d_qs, nq,
+                     I->A.run(), I->n, I->DA.run(), I->dn, I->spillA,
+                     I->d_qtails, I->d_qs, nq,
                      I->max_cap, I->d_rows, I->d_rows2, I->d_rowsm, I->d_found,
                      I->d_total, I->d_scanned);
   HIP_CHECK(hipEventRecord(I->ev1, I->stream));
-  hipLaunchKernelGGL(k_gather, dim3(nq), dim3(256), 0, I->stream, I->A.keys,
-                     I->A.meta, I->A.rev, I->A.vo, I->DA.keys, I->DA.meta,
-                     I->DA.rev, I->DA.vo, I->heapA, I->d_rowsm, I->max_cap,
+  hipLaunchKernelGGL(k_gather, dim3(nq), dim3(256), 0, I->stream, I->A.run(),
+                     I->DA.run(), I->heapA, I->d_rowsm, I->max_cap,
                      I->d_found, I->d_qs, nq, I->d_gbuf, qcap, I->d_offs,
                      I->d_gbytes, I->d_ovf, I->d_bytes);
   hipLaunchKernelGGL(k_gather_copy, dim3(nq), dim3(512), 0, I->stream,
-                     I->A.keys, I->A.meta, I->A.rev, I->A.vo, I->DA.keys,
-                     I->DA.meta, I->DA.rev, I->DA.vo, I->heapA, I->d_rowsm,
+                     I->A.run(), I->DA.run(), I->spillA, I->heapA, I->d_rowsm,
                      I->max_cap, I->d_found, I->d_qs, nq, I->d_gbuf, qcap,
                      I->d_offs, I->d_ovf);
   HIP_CHECK(hipEventRecord(I->ev2, I->stream));
@@ -1446,8 +1502,9 @@ bool Slab::RangeBatchStart(const std::vector<DevRangeQ>& qs, std::string* err) {
 }
 
 bool Slab::RangeBatchEx(const std::vector<DevRangeQ>& qs, bool d2h, bool parse,
-                        std::vector<RangeResult>* outs, std::string* err) {
-  if (!RangeBatchStart(qs, err)) return false;
+                        std::vector<RangeResult>* outs, std::string* err,
+                        const std::string& qtails) {
+  if (!RangeBatchStart(qs, err, qtails)) return false;
   return RangeBatchFinish((int)qs.size(), d2h, parse, outs, err);
 }
 
@@ -1570,23 +1627,24 @@ bool Slab::DrainD2H(std::string* err) {
 }
 
 bool Slab::GetBatch(const std::vector<DevGetQ>& qs, std::vector<GetResult>* outs,
-                    std::string* err) {
-  return GetBatchEx(qs, true, outs, err);
+                    std::string* err, const std::string& qtails) {
+  return GetBatchEx(qs, true, outs, err, qtails);
 }
 
-bool Slab::GetBatchStart(const std::vector<DevGetQ>& qs, std::string* err) {
+bool Slab::GetBatchStart(const std::vector<DevGetQ>& qs, std::string* err,
+                         const std::string& qtails) {
   Impl* I = p;
   int nq = (int)qs.size();
   if (nq == 0) return true;
   if (nq > I->max_q) { if (err) *err = "too many gets per batch"; return false; }
+  if (!I->ensure_qtails(qtails, err)) return false;
   HIP_CHECK(hipMemcpyAsync(I->d_gq, qs.data(), sizeof(DevGetQ) * nq,
                            hipMemcpyHostToDevice, I->stream));
   HIP_CHECK(hipEventRecord(I->ev_g0, I->stream));
   int blocks = (int)ceil_div(nq, 4);
-  hipLaunchKernelGGL(k_get2, dim3(blocks), dim3(256), 0, I->stream, I->A.keys,
-                     I->A.meta, I->A.rev, I->A.vo, I->DA.keys, I->DA.meta,
-                     I->DA.rev, I->DA.vo, I->n, I->dn, I->d_fkeys, I->d_frev,
-                     I->nf, I->heapA, I->d_gq, nq, 0,
+  hipLaunchKernelGGL(k_get2, dim3(blocks), dim3(256), 0, I->stream,
+                     I->A.run(), I->DA.run(), I->n, I->dn, I->spillA,
+                     I->d_qtails, I->heapA, I->d_gq, nq, 0,
                      I->d_gbuf, 0, I->d_orev, I->d_ometa, I->d_found32,
                      I->d_ovf);
   HIP_CHECK(hipMemcpyAsync(I->h_gmeta, I->d_orev, nq * 8, hipMemcpyDeviceToHost, I->stream));
@@ -1618,7 +1676,8 @@ bool Slab::GetBatchFinish(int nq, std::vector<GetResult>* outs, std::string* err
 }
 
 bool Slab::GetBatchEx(const std::vector<DevGetQ>& qs, bool values,
-                      std::vector<GetResult>* outs, std::string* err) {
+                      std::vector<GetResult>* outs, std::string* err,
+                      const std::string& qtails) {
   Impl* I = p;
   auto tt0 = std::chrono::steady_clock::now();
   auto lap = [&](double* acc) {
@@ -1638,10 +1697,9 @@ bool Slab::GetBatchEx(const std::vector<DevGetQ>& qs, bool values,
   lap(&perf.dbg_b);
   HIP_CHECK(hipEventRecord(I->ev0, I->stream));
   int blocks = (int)ceil_div(nq, 4);
-  hipLaunchKernelGGL(k_get2, dim3(blocks), dim3(256), 0, I->stream, I->A.keys,
-                     I->A.meta, I->A.rev, I->A.vo, I->DA.keys, I->DA.meta,
-                     I->DA.rev, I->DA.vo, I->n, I->dn, I->d_fkeys, I->d_frev,
-                     I->nf, I->heapA, I->d_gq, nq, values ? 1 : 0,
+  hipLaunchKernelGGL(k_get2, dim3(blocks), dim3(256), 0, I->stream,
+                     I->A.run(), I->DA.run(), I->n, I->dn, I->spillA,
+                     I->d_qtails, I->heapA, I->d_gq, nq, values ? 1 : 0,
                      I->d_gbuf, slot, I->d_orev, I->d_ometa, I->d_found32,
                      I->d_ovf);
   HIP_CHECK(hipEventRecord(I->ev1, I->stream));
@@ -1707,8 +1765,9 @@ bool Slab::Compact(const std::vector<std::pair<Bound, Bound>>& borders,
                            hipMemcpyHostToDevice, I->stream));
   HIP_CHECK(hipMemcpyAsync(I->d_brevs, brevs.data(), nb * 8,
                            hipMemcpyHostToDevice, I->stream));
-  hipLaunchKernelGGL(k_find_bounds, dim3(1), dim3(256), 0, I->stream, I->A.keys,
-                     I->A.rev, n, I->d_bkeys, I->d_brevs, nb, I->d_bounds);
+  hipLaunchKernelGGL(k_find_bounds, dim3(1), dim3(256), 0, I->stream,
+                     I->A.run(), n, I->spillA, I->d_bkeys, I->d_brevs, nb,
+                     I->d_bounds);
   std::vector<int64_t> hb(nb);
   HIP_CHECK(hipMemcpyAsync(hb.data(), I->d_bounds, nb * 8, hipMemcpyDeviceToHost,
                            I->stream));
@@ -1728,17 +1787,26 @@ bool Slab::Compact(const std::vector<std::pair<Bound, Bound>>& borders,
                      0, I->stream, I->s_a, I->A.rev, I->A.meta, I->s_c, n);
   uint64_t new_heap = 0;
   if (!I->scan(I->s_c, I->s_d, n, &new_heap, err)) return false;
+  // key-spill offsets (tails of surviving keys > KEYW)
+  hipLaunchKernelGGL(k_spill_sizes, dim3((uint32_t)ceil_div(n, 256)), dim3(256),
+                     0, I->stream, I->s_a, I->A.meta, I->s_c, n);
+  uint64_t new_spill = 0;
+  if (!I->scan(I->s_c, I->s_e, n, &new_spill, err)) return false;
   hipLaunchKernelGGL(k_compact_scatter, dim3((uint32_t)ceil_div(n, 256)),
                      dim3(256), 0, I->stream, I->A.keys, I->A.meta, I->A.rev,
-                     I->A.vo, I->s_a, I->s_b, I->s_d, I->B.keys, I->B.meta,
-                     I->B.rev, I->B.vo, n);
+                     I->A.vo, I->A.ko, I->s_a, I->s_b, I->s_d, I->s_e,
+                     I->B.keys, I->B.meta, I->B.rev, I->B.vo, I->B.ko, n);
   hipLaunchKernelGGL(k_heap_scatter, dim3((uint32_t)ceil_div(n, 4)), dim3(256),
                      0, I->stream, I->s_a, I->A.rev, I->A.meta, I->A.vo, I->s_d,
                      I->heapA, I->heapB, n);
+  if (new_spill > 0)
+    hipLaunchKernelGGL(k_spill_scatter, dim3((uint32_t)ceil_div(n, 4)),
+                       dim3(256), 0, I->stream, I->s_a, I->A.meta, I->A.ko,
+                       I->s_e, I->spillA, I->spillB, n);
   int64_t sb = ceil_div((int64_t)kept, 256);
   if (kept > 0)
     hipLaunchKernelGGL(k_same_next, dim3((uint32_t)sb), dim3(256), 0, I->stream,
-                       I->B.keys, I->B.meta, (int64_t)kept);
+                       I->B.keys, I->B.meta, I->B.ko, I->spillB, (int64_t)kept);
   HIP_CHECK(hipEventRecord(I->ev1, I->stream));
   HIP_CHECK(hipStreamSynchronize(I->stream));
   float ms = 0;
@@ -1747,9 +1815,10 @@ bool Slab::Compact(const std::vector<std::pair<Bound, Bound>>& borders,
   perf.compacts++;
   std::swap(I->A, I->B);
   std::swap(I->heapA, I->heapB);
+  std::swap(I->spillA, I->spillB);
   I->n = (int64_t)kept;
   I->heap_used_ = (int64_t)new_heap;
-  I->rebuildFence(I->A, I->n);
+  I->spill_used_ = (int64_t)new_spill;
   HIP_CHECK(hipStreamSynchronize(I->stream));
   return true;
 }
@@ -1761,20 +1830,27 @@ bool Slab::Dump(std::vector<DumpRow>* rows_out, std::string* err) {
   int64_t n = I->n;
   if (n == 0) return true;
   std::vector<uint8_t> keys((size_t)n * KEYW);
-  std::vector<uint64_t> meta(n), rev(n), vo(n);
+  std::vector<uint64_t> meta(n), rev(n), vo(n), ko(n);
   std::vector<uint8_t> heap(I->heap_used_);
+  std::vector<uint8_t> spill(I->spill_used_);
   HIP_CHECK(hipMemcpyAsync(keys.data(), I->A.keys, keys.size(), hipMemcpyDeviceToHost, I->stream));
   HIP_CHECK(hipMemcpyAsync(meta.data(), I->A.meta, n * 8, hipMemcpyDeviceToHost, I->stream));
   HIP_CHECK(hipMemcpyAsync(rev.data(), I->A.rev, n * 8, hipMemcpyDeviceToHost, I->stream));
   HIP_CHECK(hipMemcpyAsync(vo.data(), I->A.vo, n * 8, hipMemcpyDeviceToHost, I->stream));
+  HIP_CHECK(hipMemcpyAsync(ko.data(), I->A.ko, n * 8, hipMemcpyDeviceToHost, I->stream));
   if (I->heap_used_)
     HIP_CHECK(hipMemcpyAsync(heap.data(), I->heapA, I->heap_used_, hipMemcpyDeviceToHost, I->stream));
+  if (I->spill_used_)
+    HIP_CHECK(hipMemcpyAsync(spill.data(), I->spillA, I->spill_used_, hipMemcpyDeviceToHost, I->stream));
   HIP_CHECK(hipStreamSynchronize(I->stream));
   rows_out->reserve(n);
   for (int64_t i = 0; i < n; ++i) {
     DumpRow r;
     uint32_t klen = meta_klen(meta[i]);
-    r.key.assign((const char*)keys.data() + i * KEYW, klen);
+    uint32_t kin = klen > (uint32_t)KEYW ? (uint32_t)KEYW : klen;
+    r.key.assign((const char*)keys.data() + i * KEYW, kin);
+    if (klen > (uint32_t)KEYW)  // spill tail (keys > 96B)
+      r.key.append((const char*)spill.data() + ko[i], klen - KEYW);
     r.rev = rev[i];
     r.meta = meta[i];
     r.vo = vo[i];

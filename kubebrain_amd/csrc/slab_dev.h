@@ -15,6 +15,12 @@
 namespace kbslab {
 
 constexpr int KEYW = 96;  // fixed-width zero-padded key column
+// Keys LONGER than KEYW (SURVEY §7 hard part (a)): the column holds the
+// first 96 bytes; the tail lives in an append-only key-spill heap addressed
+// by the per-row `ko` column. Ordering: the zero-padded 96B prefix decides
+// every compare except exact 96-byte prefix ties (impossible unless both
+// keys are >= 96B, since key bytes are > 0x24 > 0x00), which compare tails.
+constexpr int KB_MAX_KEY = 4096;  // validated bound (meta klen is 16-bit)
 
 // meta word bits (per row)
 constexpr uint64_t M_SAME_NEXT = 1ull << 0;  // next row has the same user key
@@ -39,6 +45,10 @@ KB_HD inline uint32_t meta_vlen(uint64_t m) { return (uint32_t)(m >> 32); }
 struct DevRangeQ {
   uint8_t start[KEYW];
   uint8_t end[KEYW];
+  // true bound lengths; tails of bounds longer than KEYW live in the
+  // per-batch query-tail buffer at start_ko/end_ko
+  uint32_t start_klen, end_klen;
+  uint64_t start_ko, end_ko;
   uint64_t read_rev;
   // scan begins at the first row with (key,rev) >= (start, start_rev).
   // 0 = inclusive start-of-key (the normal case); UINT64_MAX = strictly
@@ -55,6 +65,8 @@ struct DevRangeQ {
 };
 struct DevGetQ {
   uint8_t key[KEYW];
+  uint32_t klen, _pad;  // true key length; tail at ko when > KEYW
+  uint64_t ko;
   uint64_t read_rev;  // UINT64_MAX for "latest"
 };
 #pragma pack(pop)
@@ -80,8 +92,9 @@ struct GetResult {
 // sorted delta (memtable flush) — vo entries already ABSOLUTE heap offsets
 // (host adds the pre-merge heap_used base) or objRev for rev rows.
 struct DeltaRows {
-  std::vector<uint8_t> keys;  // m * KEYW
+  std::vector<uint8_t> keys;  // m * KEYW (96B zero-padded prefixes)
   std::vector<uint64_t> meta, rev, vo;
+  std::vector<uint64_t> ko;   // spill offsets for keys > KEYW (empty = none)
   std::vector<uint8_t> heap;  // new value bytes (4B-aligned records)
   int64_t m = 0;
 };
@@ -134,31 +147,38 @@ class Slab {
 
   // append value bytes to the device heap; *off = absolute offset
   bool HeapAppend(const void* p, int64_t len, int64_t* off, std::string* err);
+  // append key-tail bytes to the key-spill heap (keys > KEYW); *off absolute
+  bool SpillAppend(const void* p, int64_t len, int64_t* off, std::string* err);
+  int64_t spill_used() const;
   // merge sorted new rows (vo already absolute) into the DELTA run.
   // known_new_dn >= 0: the caller knows the exact post-merge row count
   // (drops = rev-row replacements it tracked), so the merge runs fully
   // async — no host sync; the next kernel on the stream queues behind it.
   // KB_VALIDATE_DN=1 re-checks the prediction (enabled by the test env).
   bool AppendRows(const uint8_t* keys, const uint64_t* meta, const uint64_t* rev,
-                  const uint64_t* vo, int64_t m, std::string* err,
-                  int64_t known_new_dn = -1);
+                  const uint64_t* vo, const uint64_t* ko, int64_t m,
+                  std::string* err, int64_t known_new_dn = -1);
   // fold the delta run into the base run; delta becomes empty
   bool Fold(std::string* err);
 
   // batched Range (the north-star kernel; scanner worker.run semantics,
   // scanner.go:389-516). d2h=false leaves records in the device arena
   // (bench "value" mode); d2h=true packs + copies + parses them.
+  // qtails = concatenated tails of bounds longer than KEYW (may be empty)
   bool RangeBatch(const std::vector<DevRangeQ>& qs, bool d2h,
-                  std::vector<RangeResult>* outs, std::string* err);
+                  std::vector<RangeResult>* outs, std::string* err,
+                  const std::string& qtails = std::string());
   // parse=false: D2H into pinned memory without materializing records
   bool RangeBatchEx(const std::vector<DevRangeQ>& qs, bool d2h, bool parse,
-                    std::vector<RangeResult>* outs, std::string* err);
+                    std::vector<RangeResult>* outs, std::string* err,
+                    const std::string& qtails = std::string());
   // async split: Start launches the scan+gather without syncing, so host
   // work (e.g. the txn leg) overlaps the in-flight kernels; Finish collects.
   // d2h with parse=false runs PIPELINED: the payload copy lands in pinned
   // host memory on a copy stream, overlapping the next batch's kernels —
   // call DrainD2H() before reading wall-clock results.
-  bool RangeBatchStart(const std::vector<DevRangeQ>& qs, std::string* err);
+  bool RangeBatchStart(const std::vector<DevRangeQ>& qs, std::string* err,
+                       const std::string& qtails = std::string());
   bool RangeBatchFinish(int nq, bool d2h, bool parse,
                         std::vector<RangeResult>* outs, std::string* err);
   bool DrainD2H(std::string* err);
@@ -167,14 +187,17 @@ class Slab {
   // GetBatchEx(values=false) skips the value copy (meta-only: found/rev/
   // tomb) — the device-side revIndex lookup of the batched txn path.
   bool GetBatch(const std::vector<DevGetQ>& qs, std::vector<GetResult>* outs,
-                std::string* err);
+                std::string* err,
+                const std::string& qtails = std::string());
   bool GetBatchEx(const std::vector<DevGetQ>& qs, bool values,
-                  std::vector<GetResult>* outs, std::string* err);
+                  std::vector<GetResult>* outs, std::string* err,
+                  const std::string& qtails = std::string());
   // async split: Start launches the lookup; Finish waits ONLY on the
   // lookup's completion event, so kernels launched on the stream AFTER
   // Start (e.g. the range batch) keep running while the host consumes the
   // results. values=false only (meta-only lookups).
-  bool GetBatchStart(const std::vector<DevGetQ>& qs, std::string* err);
+  bool GetBatchStart(const std::vector<DevGetQ>& qs, std::string* err,
+                     const std::string& qtails = std::string());
   bool GetBatchFinish(int nq, std::vector<GetResult>* outs, std::string* err);
 
   // compaction mark+sweep over encoded borders (compact.go:55-68 +

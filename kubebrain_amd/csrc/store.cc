@@ -57,6 +57,19 @@ static void pad96(const Bytes& k, uint8_t out[KEYW]) {
   memcpy(out, k.data(), std::min(k.size(), (size_t)KEYW));
 }
 
+// fill a query bound: 96B padded prefix + tail into the per-batch qtails
+// buffer when the bound is longer than the key column
+static void setBound(uint8_t dst96[KEYW], uint32_t* klen, uint64_t* ko,
+                     const Bytes& b, std::string* qtails) {
+  pad96(b, dst96);
+  *klen = (uint32_t)b.size();
+  *ko = 0;
+  if (b.size() > (size_t)KEYW) {
+    *ko = qtails->size();
+    qtails->append(b.data() + KEYW, b.size() - KEYW);
+  }
+}
+
 static int64_t env_i64(const char* name, int64_t dflt) {
   const char* v = getenv(name);
   return v && *v ? atoll(v) : dflt;
@@ -82,7 +95,9 @@ Store* Store::Open(const Config& cfg_in, std::string* err) {
 Store::~Store() { delete slab_; }
 
 Status Store::validateKey(const Bytes& key) const {
-  if (key.size() > (size_t)KEYW) return KEYTOOLONG;  // DESIGN.md §4 round-1 limit
+  // keys > 96B spill their tail to the key-spill heap; the
+  // reference codec imposes no length limit (coder/normal.go:42-50)
+  if (key.size() > (size_t)kbslab::KB_MAX_KEY) return KEYTOOLONG;
   for (char c : key)
     if ((uint8_t)c <= 0x24) return BADKEY;  // coder/normal.go:29-31 constraint
   return OK;
@@ -94,7 +109,7 @@ Status Store::validateKey(const Bytes& key) const {
 // inside a bound would order differently against '$'-terminated internal keys
 // (coder/normal.go:29-31) and is rejected loudly.
 static Status validateBound(const Bytes& b) {
-  if (b.size() > (size_t)KEYW) return KEYTOOLONG;
+  if (b.size() > (size_t)kbslab::KB_MAX_KEY) return KEYTOOLONG;
   size_t end = b.size();
   while (end > 0 && b[end - 1] == '\x00') end--;
   for (size_t i = 0; i < end; ++i)
@@ -125,6 +140,17 @@ uint64_t Store::mustDeal(uint64_t prevRevision) {  // txn.go:139-142
   return deal(prevRevision, &st);
 }
 
+uint64_t Store::stageSpill(const Bytes& key) {
+  // tail of a key > 96B goes to the key-spill heap (absolute offset
+  // pre-assigned, uploaded by syncReads like the value heap)
+  if (key.size() <= (size_t)KEYW) return 0;
+  if (spill_base_ < 0) spill_base_ = slab_->spill_used();
+  uint64_t off = (uint64_t)(spill_base_ + (int64_t)spill_pending_.size());
+  spill_pending_.append(key.data() + KEYW, key.size() - KEYW);
+  spill_pending_.resize((spill_pending_.size() + 15) & ~15ull, '\0');
+  return off;
+}
+
 void Store::putRow(const Bytes& key, uint64_t rev, const Bytes& val) {
   NewRow r;
   r.key = key;
@@ -135,6 +161,7 @@ void Store::putRow(const Bytes& key, uint64_t rev, const Bytes& val) {
                              (uint32_t)val.size());
   if (heap_base_ < 0) heap_base_ = slab_->heap_used();
   r.vo = (uint64_t)(heap_base_ + (int64_t)heap_pending_.size());
+  r.ko = stageSpill(key);
   heap_pending_ += val;
   heap_pending_.resize((heap_pending_.size() + 15) & ~15ull, '\0');
   newrows_.push_back(std::move(r));
@@ -147,6 +174,7 @@ void Store::putRevRow(const Bytes& key, uint64_t objrev, bool flag9) {
   bool ev = key.find(kEvents) != Bytes::npos;
   r.meta = kbslab::meta_make(false, flag9, ev, (uint32_t)key.size(), flag9 ? 9 : 8);
   r.vo = objrev;
+  r.ko = stageSpill(key);
   auto it = nr_revrow_.find(key);
   if (it != nr_revrow_.end()) {
     newrows_[it->second] = std::move(r);  // replace in place (unique key@0)
@@ -172,6 +200,15 @@ bool Store::syncReads(std::string* err) {
     heap_pending_.clear();
     heap_base_ = -1;
   }
+  if (!spill_pending_.empty()) {
+    int64_t off = 0;
+    if (!slab_->SpillAppend(spill_pending_.data(),
+                            (int64_t)spill_pending_.size(), &off, err))
+      return false;
+    if (off != spill_base_) { if (err) *err = "spill offset drift"; return false; }
+    spill_pending_.clear();
+    spill_base_ = -1;
+  }
   if (!newrows_.empty()) {
     std::sort(newrows_.begin(), newrows_.end(),
               [](const NewRow& a, const NewRow& b) {
@@ -181,12 +218,13 @@ bool Store::syncReads(std::string* err) {
               });
     size_t m = newrows_.size();
     std::vector<uint8_t> keys(m * KEYW);
-    std::vector<uint64_t> meta(m), rev(m), vo(m);
+    std::vector<uint64_t> meta(m), rev(m), vo(m), ko(m);
     for (size_t i = 0; i < m; ++i) {
       pad96(newrows_[i].key, keys.data() + i * KEYW);
       meta[i] = newrows_[i].meta;
       rev[i] = newrows_[i].rev;
       vo[i] = newrows_[i].vo;
+      ko[i] = newrows_[i].ko;
     }
     // delta-merge drops happen exactly where an uploaded rev-row's key
     // already has a rev-row in the delta run — tracked host-side, so the
@@ -203,7 +241,7 @@ bool Store::syncReads(std::string* err) {
     }
     int64_t predicted = slab_->delta_rows() + (int64_t)m - drops;
     if (!slab_->AppendRows(keys.data(), meta.data(), rev.data(), vo.data(),
-                           (int64_t)m, err, predicted))
+                           ko.data(), (int64_t)m, err, predicted))
       return false;
     newrows_.clear();
     nr_revrow_.clear();
@@ -275,13 +313,20 @@ void Store::pumpEvents() {
     if (!slab_->WatchFilter(batch, &bitmap, &W, &err)) { pending_.clear(); return; }
     int64_t words = (e + 63) / 64;
     for (auto& [id, w] : watchers_) {
-      if (w.dropped || w.slot >= W) continue;
+      if (w.dropped || w.slot >= W || w.slot < 0) continue;
+      // prefixes longer than the 96B filter column: the device test on the
+      // first 96 bytes is a superset; re-check the full prefix here
+      const bool long_pfx = w.prefix.size() > (size_t)KEYW;
       for (int64_t c = 0; c < words; ++c) {
         uint64_t bits = bitmap[(size_t)(w.slot * words + c)];
         while (bits) {
           int j = __builtin_ctzll(bits);
           bits &= bits - 1;
-          w.queue.push_back(pending_[b0 + c * 64 + j]);
+          const Event& ev2 = pending_[b0 + c * 64 + j];
+          if (long_pfx &&
+              ev2.kv_key.compare(0, w.prefix.size(), w.prefix) != 0)
+            continue;
+          w.queue.push_back(ev2);
           delivered_++;
         }
       }
@@ -337,14 +382,14 @@ Status Store::StreamNext(int64_t sid, std::vector<KeyValue>* kvs) {
   std::vector<kbslab::RangeResult> outs;
   while ((int64_t)kvs->size() < 300) {
     DevRangeQ q{};
-    memset(q.start, 0, KEYW);
-    memcpy(q.start, ss.frontier.data(), std::min(ss.frontier.size(), (size_t)KEYW));
-    pad96(ss.end, q.end);
+    std::string qtails;
+    setBound(q.start, &q.start_klen, &q.start_ko, ss.frontier, &qtails);
+    setBound(q.end, &q.end_klen, &q.end_ko, ss.end, &qtails);
     q.read_rev = ss.read_rev;
     q.start_rev = ss.started ? UINT64_MAX : 0;
     q.cap = std::min<int64_t>(300 - (int64_t)kvs->size(), max_cap);
     q.count_only = 0;
-    if (!slab_->RangeBatch({q}, true, &outs, &err)) return INTERNAL;
+    if (!slab_->RangeBatch({q}, true, &outs, &err, qtails)) return INTERNAL;
     kbslab::RangeResult& r = outs[0];
     if (r.overflow) return NOBUF;
     for (auto& rec : r.recs) kvs->push_back(KeyValue{rec.key, rec.val, rec.rev});
@@ -422,7 +467,9 @@ int64_t Store::Watch(const Bytes& prefix, uint64_t revision, Status* st) {
   uint8_t p96[KEYW];
   pad96(prefix, p96);
   std::string err;
-  if (!slab_->WatcherSet(slot, p96, (uint32_t)prefix.size(), w.from_rev, &err)) {
+  uint32_t plen96 = prefix.size() > (size_t)KEYW ? (uint32_t)KEYW
+                                                 : (uint32_t)prefix.size();
+  if (!slab_->WatcherSet(slot, p96, plen96, w.from_rev, &err)) {
     free_slots_.push_back(slot);  // the reserved slot must not leak
     *st = INTERNAL;
     return -1;
@@ -538,11 +585,17 @@ Status Store::get(const Bytes& key, uint64_t revision, Bytes* val, uint64_t* mod
   std::string err;
   if (!syncReads(&err)) return INTERNAL;
   uint64_t R = revision == 0 ? UINT64_MAX : revision;
-  DevGetQ q;
+  DevGetQ q{};
+  std::string qtails;
   pad96(key, q.key);
+  q.klen = (uint32_t)key.size();
+  q.ko = 0;
+  if (key.size() > (size_t)KEYW) {
+    qtails.assign(key.data() + KEYW, key.size() - KEYW);
+  }
   q.read_rev = R;
   std::vector<kbslab::GetResult> outs;
-  if (!slab_->GetBatch({q}, &outs, &err)) return INTERNAL;
+  if (!slab_->GetBatch({q}, &outs, &err, qtails)) return INTERNAL;
   if (!outs[0].found) return NOTFOUND;
   *modRev = outs[0].rev;
   if (outs[0].tomb) return NOTFOUND;
@@ -742,16 +795,16 @@ RangeResponse Store::List(const Bytes& start, const Bytes& end,
   int64_t chunk_cap = 0;  // 0 = unbounded; halved on arena overflow
   while (need() && dev_more) {
     DevRangeQ q{};
-    memset(q.start, 0, KEYW);
-    memcpy(q.start, dev_frontier.data(), std::min(dev_frontier.size(), (size_t)KEYW));
-    pad96(end, q.end);
+    std::string qtails;
+    setBound(q.start, &q.start_klen, &q.start_ko, dev_frontier, &qtails);
+    setBound(q.end, &q.end_klen, &q.end_ko, end, &qtails);
     q.read_rev = reqRevision;
     q.start_rev = frontier_rev;
     q.cap = lim > 0 ? lim - (int64_t)kvs.size() : chunk_cap;
     if (chunk_cap > 0 && (q.cap <= 0 || q.cap > chunk_cap)) q.cap = chunk_cap;
     if (q.cap <= 0 || q.cap > max_cap) q.cap = max_cap;
     q.count_only = 0;
-    if (!slab_->RangeBatch({q}, true, &outs, &err)) { *st = INTERNAL; return resp; }
+    if (!slab_->RangeBatch({q}, true, &outs, &err, qtails)) { *st = INTERNAL; return resp; }
     kbslab::RangeResult& r = outs[0];
     if (r.overflow) {
       // results exceed the device arena: halve the chunk and continue (large
@@ -798,13 +851,14 @@ CountResponse Store::Count(const Bytes& start, const Bytes& end, Status* st) {
   std::string err;
   if (!syncReads(&err)) { *st = INTERNAL; return resp; }
   DevRangeQ q{};
-  pad96(start, q.start);
-  pad96(end, q.end);
+  std::string qtails;
+  setBound(q.start, &q.start_klen, &q.start_ko, start, &qtails);
+  setBound(q.end, &q.end_klen, &q.end_ko, end, &qtails);
   q.read_rev = rev;
   q.cap = 0;
   q.count_only = 1;
   std::vector<kbslab::RangeResult> outs;
-  if (!slab_->RangeBatch({q}, false, &outs, &err)) { *st = INTERNAL; return resp; }
+  if (!slab_->RangeBatch({q}, false, &outs, &err, qtails)) { *st = INTERNAL; return resp; }
   resp.count = (uint64_t)outs[0].total;
   *st = OK;
   return resp;
@@ -980,7 +1034,8 @@ bool Store::BulkCreate(const uint8_t* keys, const uint32_t* klens,
 // straight into device query structs (no per-query allocations)
 static void parseBenchQueries(const uint8_t* qbuf, size_t nq, uint64_t cur_rev,
                               int mode, std::vector<DevRangeQ>* qall,
-                              std::vector<int64_t>* limits) {
+                              std::vector<int64_t>* limits,
+                              std::string* qtails) {
   qall->assign(nq, DevRangeQ{});
   limits->resize(nq);
   const int keys_only = (mode & 2) ? 1 : 0;
@@ -993,8 +1048,10 @@ static void parseBenchQueries(const uint8_t* qbuf, size_t nq, uint64_t cur_rev,
     memcpy(&rev, p, 8); p += 8;
     memcpy(&limit, p, 8); p += 8;
     DevRangeQ& q = (*qall)[i];
-    memcpy(q.start, p, std::min((size_t)slen, (size_t)KEYW)); p += slen;
-    memcpy(q.end, p, std::min((size_t)elen, (size_t)KEYW)); p += elen;
+    setBound(q.start, &q.start_klen, &q.start_ko,
+             Bytes((const char*)p, slen), qtails); p += slen;
+    setBound(q.end, &q.end_klen, &q.end_ko,
+             Bytes((const char*)p, elen), qtails); p += elen;
     q.read_rev = rev == 0 ? cur_rev : rev;
     q.start_rev = 0;
     q.cap = (int64_t)limit > 0 ? (int64_t)limit + 1 : 0;
@@ -1011,7 +1068,8 @@ bool Store::BenchRange(const uint8_t* qbuf, size_t nq, int mode,
   if (!syncReads(err)) return false;
   std::vector<DevRangeQ> qall;
   std::vector<int64_t> limits;
-  parseBenchQueries(qbuf, nq, committed_, mode, &qall, &limits);
+  std::string qtails;
+  parseBenchQueries(qbuf, nq, committed_, mode, &qall, &limits, &qtails);
   const bool d2h = (mode & 1) != 0;
   const int64_t kMax = 1024;
   unsigned long long tot = 0;
@@ -1020,7 +1078,7 @@ bool Store::BenchRange(const uint8_t* qbuf, size_t nq, int mode,
   for (size_t b0 = 0; b0 < (size_t)nq; b0 += kMax) {
     size_t bn = std::min((size_t)kMax, (size_t)nq - b0);
     std::vector<DevRangeQ> dq(qall.begin() + b0, qall.begin() + b0 + bn);
-    if (!slab_->RangeBatchEx(dq, d2h, /*parse=*/false, &outs, err)) return false;
+    if (!slab_->RangeBatchEx(dq, d2h, /*parse=*/false, &outs, err, qtails)) return false;
     for (size_t j = 0; j < bn; ++j) {
       int64_t lim = limits[b0 + j];
       int64_t w = outs[j].written;
@@ -1046,7 +1104,8 @@ bool Store::BenchStep(const uint8_t* qbuf, size_t nq, const uint8_t* tbuf,
   if (!syncReads(err)) return false;
   std::vector<DevRangeQ> qall;
   std::vector<int64_t> limits;
-  parseBenchQueries(qbuf, nq, committed_, mode, &qall, &limits);
+  std::string qtails;
+  parseBenchQueries(qbuf, nq, committed_, mode, &qall, &limits, &qtails);
   const bool d2h = (mode & 1) != 0;
   // txn ops parsed up front so the batched CAS lookup (f1) can launch BEFORE
   // the range batch: stream order runs the small lookup first, the host
@@ -1075,13 +1134,20 @@ bool Store::BenchStep(const uint8_t* qbuf, size_t nq, const uint8_t* tbuf,
   auto t0 = std::chrono::steady_clock::now();
   if (tuniq) {
     std::vector<DevGetQ> gq(ntx);
+    std::string gtails;
     for (size_t i = 0; i < ntx; ++i) {
+      gq[i] = DevGetQ{};
       pad96(tops[i].key, gq[i].key);
+      gq[i].klen = (uint32_t)tops[i].key.size();
+      if (tops[i].key.size() > (size_t)KEYW) {
+        gq[i].ko = gtails.size();
+        gtails.append(tops[i].key.data() + KEYW, tops[i].key.size() - KEYW);
+      }
       gq[i].read_rev = UINT64_MAX;
     }
-    if (!slab_->GetBatchStart(gq, err)) return false;
+    if (!slab_->GetBatchStart(gq, err, gtails)) return false;
   }
-  if (!slab_->RangeBatchStart(qall, err)) return false;
+  if (!slab_->RangeBatchStart(qall, err, qtails)) return false;
   if (ntx > 0) {
     if (tuniq) {
       std::vector<kbslab::GetResult> cur;
@@ -1160,14 +1226,22 @@ bool Store::BenchTxn(const uint8_t* tbuf, size_t n, uint64_t* out_revs,
     size_t cn = std::min(kChunk, n - c0);
     if (!syncReads(err)) return false;  // device revision state current
     std::vector<DevGetQ> qs(cn);
+    std::string gtails;
     for (size_t i = 0; i < cn; ++i) {
       Status v = validateKey(ops[c0 + i].key);
       if (v != OK) { if (err) *err = "bench txn bad key"; return false; }
+      qs[i] = DevGetQ{};
       pad96(ops[c0 + i].key, qs[i].key);
+      qs[i].klen = (uint32_t)ops[c0 + i].key.size();
+      if (ops[c0 + i].key.size() > (size_t)KEYW) {
+        qs[i].ko = gtails.size();
+        gtails.append(ops[c0 + i].key.data() + KEYW,
+                      ops[c0 + i].key.size() - KEYW);
+      }
       qs[i].read_rev = UINT64_MAX;
     }
     std::vector<kbslab::GetResult> cur;
-    if (!slab_->GetBatchEx(qs, /*values=*/false, &cur, err)) return false;
+    if (!slab_->GetBatchEx(qs, /*values=*/false, &cur, err, gtails)) return false;
     if (!applyTxnOps(&ops[c0], cn, cur, out_revs + c0, err)) return false;
   }
   return true;
@@ -1245,14 +1319,22 @@ bool Store::BenchDel(const uint8_t* dbuf, size_t n, uint64_t* out_revs,
     size_t cn = std::min(kChunk, n - c0);
     if (!syncReads(err)) return false;
     std::vector<DevGetQ> qs(cn);
+    std::string gtails;
     for (size_t i = 0; i < cn; ++i) {
       Status v = validateKey(ops[c0 + i].key);
       if (v != OK) { if (err) *err = "bench del bad key"; return false; }
+      qs[i] = DevGetQ{};
       pad96(ops[c0 + i].key, qs[i].key);
+      qs[i].klen = (uint32_t)ops[c0 + i].key.size();
+      if (ops[c0 + i].key.size() > (size_t)KEYW) {
+        qs[i].ko = gtails.size();
+        gtails.append(ops[c0 + i].key.data() + KEYW,
+                      ops[c0 + i].key.size() - KEYW);
+      }
       qs[i].read_rev = UINT64_MAX;
     }
     std::vector<kbslab::GetResult> cur;
-    if (!slab_->GetBatchEx(qs, /*values=*/true, &cur, err)) return false;
+    if (!slab_->GetBatchEx(qs, /*values=*/true, &cur, err, gtails)) return false;
     for (size_t i = 0; i < cn; ++i) {
       const Op& op = ops[c0 + i];
       ops_delete_++;

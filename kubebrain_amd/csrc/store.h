@@ -139,6 +139,7 @@ class Store {
   Status get(const Bytes& key, uint64_t revision, Bytes* val, uint64_t* modRev);
   Status checkCompactRace(uint64_t revision);
   void putRow(const Bytes& key, uint64_t rev, const Bytes& val);
+  uint64_t stageSpill(const Bytes& key);  // stage a long key's tail
   void putRevRow(const Bytes& key, uint64_t objrev, bool flag9);
   void pumpEvents();  // fan-out pending events via the GPU filter
   struct BatchOp { Bytes key, val; uint64_t prev; };
@@ -165,11 +166,13 @@ class Store {
 
   // rows written since the last device sync: values are
   // staged in heap_pending_ with pre-assigned absolute heap offsets
-  struct NewRow { Bytes key; uint64_t rev, meta, vo; };
+  struct NewRow { Bytes key; uint64_t rev, meta, vo, ko; };
   std::vector<NewRow> newrows_;
   std::unordered_map<Bytes, size_t> nr_revrow_;  // key -> index of rev-row
   Bytes heap_pending_;
   int64_t heap_base_ = -1;
+  Bytes spill_pending_;   // key tails (> 96B) staged for the spill heap
+  int64_t spill_base_ = -1;
 
   bool compact_cell_set_ = false;
   uint64_t compact_cell_ = 0;

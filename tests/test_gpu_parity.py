@@ -604,3 +604,107 @@ def test_bench_txn_batched_parity():
         d.diff_event_log()
     finally:
         d.close()
+
+
+def test_long_keys_spill():
+    """Keys longer than the 96B key column (up to KB_MAX_KEY): tails live in
+    the key-spill heap; ordering, MVCC winners, compaction and watch stay
+    bit-exact vs the oracle — including keys sharing an identical 96-byte
+    prefix, where only the spilled tails decide the order
+    (coder/normal.go:42-50 imposes no length limit)."""
+    d = parity.Dual()
+    try:
+        # P96 is exactly 96 bytes; all long keys share it as their column
+        # prefix, so every compare ties on the column and resolves in spill
+        P = b"/registry/pods/ns-long/" + b"x" * 73
+        assert len(P) == 96
+        keys = [P,                       # exactly-96B key
+                P + b"-aa", P + b"-ab", P + b"-b",
+                P + b"." * 100,          # 196B
+                P + b"~tail-" + b"z" * 198]  # 300B
+        revs = {}
+        w = d.watch(b"/registry/pods/ns-long/", 0)
+        wlong = d.watch(P + b"-a", 0)    # >96B watch prefix
+        for i, k in enumerate(keys):
+            r = d.create(k, b"lv-%d" % i)
+            assert r.succeeded
+            revs[k] = r.header_revision
+        # interleave short keys in the same namespace
+        for i in range(5):
+            d.create(b"/registry/pods/ns-long/short-%d" % i, b"s%d" % i)
+        lo, hi = b"/registry/pods/ns-long/", b"/registry/pods/ns-long0"
+        r = d.list(lo, hi, 0, 0)
+        assert len(r.kvs) == len(keys) + 5
+        d.count(lo, hi)
+        # updates + MVCC reads at old revisions
+        mid = d.p.current_rev()
+        for k in keys[:3]:
+            r = d.update(k, b"lv2", revs[k])
+            assert r.succeeded
+            revs[k] = r.header_revision
+        for k in keys:
+            d.get(k, 0)
+            d.get(k, mid)
+        d.list(lo, hi, mid, 0)
+        d.list(lo, hi, 0, 4)             # limit cut across tie groups
+        # delete one long key; tombstone + recreate
+        d.delete(keys[4], 0)
+        d.get(keys[4], 0)
+        d.create(keys[4], b"again")
+        # stream (300-batch protocol) over the long-key namespace
+        d.stream(lo, hi, 0)
+        d.poll(w)
+        d.poll(wlong)
+        d.diff_dump()
+        # compaction: old versions + tombstones die, spill heap is compacted
+        d.compact(0)
+        d.diff_dump()
+        r = d.list(lo, hi, 0, 0)
+        assert len(r.kvs) == len(keys) + 5
+        d.diff_event_log()
+    finally:
+        d.close()
+
+
+def test_long_keys_random_soak():
+    """Randomized mixed workload over a keyspace with ~30% long keys
+    (97-300B), diffed against the oracle every phase."""
+    rng = random.Random(0x10A6)
+    d = parity.Dual()
+    try:
+        P = b"/registry/cfg/ns-0/" + b"p" * 80  # 99B shared prefix
+        pool = []
+        for i in range(40):
+            if i % 3 == 0:
+                pool.append(b"/registry/cfg/ns-0/obj-%04d" % i)
+            elif i % 3 == 1:
+                pool.append(P[:96] + b"/t-%04d" % i + b"y" * rng.randrange(0, 150))
+            else:
+                pool.append(P + b"-%04d" % i)
+        live = {}
+        for step in range(600):
+            op = rng.random()
+            k = pool[rng.randrange(len(pool))]
+            if op < 0.4:
+                r = d.create(k, b"c%d" % step)
+                if r.succeeded:
+                    live[k] = r.header_revision
+            elif op < 0.6:
+                r = d.update(k, b"u%d" % step, live.get(k, 0))
+                if r.succeeded:
+                    live[k] = r.header_revision
+            elif op < 0.7:
+                r = d.delete(k, 0)
+                if r.succeeded:
+                    live.pop(k, None)
+            elif op < 0.9:
+                d.list(b"/registry/cfg/ns-0/", b"/registry/cfg/ns-00", 0,
+                       rng.choice([0, 3, 10]))
+            else:
+                d.get(k, 0)
+            if step in (250, 500):
+                d.compact(max(1, d.p.current_rev() - 30))
+                d.diff_dump()
+        d.diff_dump()
+    finally:
+        d.close()
