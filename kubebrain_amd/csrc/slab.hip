@@ -1221,7 +1221,8 @@ Slab* Slab::Create(int64_t max_rows, int64_t heap_cap, int device,
   I->heap_cap = heap_cap;
   I->max_q = (int)env_i64("KB_MAX_Q", 1024);
   I->scan_t = (int)env_i64("KB_SCAN_T", 256);
-  if (I->scan_t < 64) I->scan_t = 64;
+  // the scan prologue resolves 4 bounds with waves 0-3: >= 256 threads
+  if (I->scan_t < 256) I->scan_t = 256;
   if (I->scan_t > 1024) I->scan_t = 1024;
   I->scan_t &= ~63;
   I->max_cap = env_i64("KB_MAX_CAP", 4352);

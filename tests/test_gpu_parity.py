@@ -107,18 +107,20 @@ def test_limits_and_boundaries(dual):
 
 def test_key_validation(dual):
     import kbclient
-    # key too long (KB_EKEYTOOLONG)
-    long_key = b"/registry/" + b"x" * 96
-    r = dual.p.create(long_key, b"v")
+    # keys beyond the spill bound (KB_MAX_KEY = 4096) are rejected loudly
+    too_long = b"/registry/" + b"x" * 4096
+    r = dual.p.create(too_long, b"v")
     assert r.status == kbclient.KEYTOOLONG
     # key bytes <= '$' (KB_EBADKEY)
     r = dual.p.create(b"/registry/a\x01b", b"v")
     assert r.status == kbclient.BADKEY
-    # max allowed length (96B total)
+    # exactly 96B (full key column) and just past it (spill tail)
     k96 = b"/registry/" + b"y" * 86
     assert len(k96) == 96
     dual.create(k96, b"v96")
+    dual.create(k96 + b"z", b"v97")
     dual.get(k96, 0)
+    dual.get(k96 + b"z", 0)
     dual.list(b"/registry/", b"/registry0", 0, 0)
 
 
