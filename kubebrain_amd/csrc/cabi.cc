@@ -2,6 +2,8 @@
 // host store. Signatures intentionally mirror the oracle's okb_* ABI so the
 // test harness drives both and diffs (tests/kbclient.py).
 
+#include <cstdio>
+#include <cstdlib>
 #include <cstring>
 #include <string>
 #include <vector>
@@ -21,6 +23,8 @@ thread_local std::string g_last_msg;
 void set_err(int st, const std::string& msg) {
   g_last_status = st;
   g_last_msg = msg;
+  static const bool dbg = getenv("KB_DEBUG") && atoi(getenv("KB_DEBUG")) != 0;
+  if (dbg) fprintf(stderr, "[kb_slab] error %d: %s\n", st, msg.c_str());
 }
 
 struct Writer {

@@ -1690,6 +1690,7 @@ bool Slab::GetBatchEx(const std::vector<DevGetQ>& qs, bool values,
   outs->assign(nq, GetResult());
   if (nq == 0) return true;
   if (nq > I->max_q) { if (err) *err = "too many gets per batch"; return false; }
+  if (!I->ensure_qtails(qtails, err)) return false;
   int64_t slot = I->arena_bytes / nq;
   slot &= ~15ll;
   lap(&perf.dbg_a);
