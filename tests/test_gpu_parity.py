@@ -560,9 +560,11 @@ def test_bench_txn_batched_parity():
         for (k, pr, v), nr in zip(ops, out):
             ro = d.o.update(k, v, pr)
             assert (ro.succeeded and ro.header_revision or 0) == int(nr), (k, ro, nr)
-        # batch 2: deletes — hit, miss, wrong prev, tombstone (unique keys)
+        # batch 2: deletes — hit, miss, stale prev (CAS fail), tombstone
+        # (unique keys; a FUTURE prev_rev would hit the reference's
+        # revision-drift guard, an error on every path — txn.go:139-142)
         dels = [(ns + b"/o-000", 0), (ns + b"/o-003", int(out[3]) or revs[3]),
-                (ns + b"/none", 0), (ns + b"/o-006", 12345)]
+                (ns + b"/none", 0), (ns + b"/o-006", 1)]
         parts = []
         for k, pr in dels:
             parts.append(S.pack("<IQ", len(k), pr))
