@@ -132,21 +132,27 @@ int main() {
     if (!outs[0].found || !outs[0].tomb || outs[0].rev != 5) return 1;
   }
 
-  // watch filter
+  // watch: device event ring + ballot filter + catch-up scan
   {
     STEP("watcher_set", s->WatcherSet(0, (const uint8_t*)"/registry/pods/", 15, 1, &err));
-    WatchFilterBatch b;
-    b.e = 3;
-    b.ekeys.resize(3 * KEYW);
-    pad("/registry/pods/a", b.ekeys.data());
-    pad("/registry/cm/x", b.ekeys.data() + KEYW);
-    pad("/registry/pods/b", b.ekeys.data() + 2 * KEYW);
-    b.erevs = {7, 8, 9};
+    STEP("ring_init", s->EventRingInit(1024, &err));
+    std::vector<uint8_t> ekeys(3 * KEYW, 0);
+    pad("/registry/pods/a", ekeys.data());
+    pad("/registry/cm/x", ekeys.data() + KEYW);
+    pad("/registry/pods/b", ekeys.data() + 2 * KEYW);
+    std::vector<uint64_t> erevs = {7, 8, 9};
+    STEP("ring_push", s->EventRingPush(ekeys.data(), erevs.data(), 3, 0, &err));
     std::vector<uint64_t> bm;
     int64_t W = 0;
-    STEP("watch_filter", s->WatchFilter(b, &bm, &W, &err));
+    STEP("watch_filter", s->WatchFilterRing(0, 3, &bm, &W, &err));
     printf("  W=%lld bm0=%llx\n", (long long)W, (unsigned long long)bm[0]);
     if ((bm[0] & 7) != 5) { printf("BAD filter bitmap\n"); return 1; }
+    std::vector<uint64_t> cw;
+    uint8_t p96[KEYW] = {0};
+    memcpy(p96, "/registry/pods/", 15);
+    STEP("watch_catchup", s->WatchCatchup(p96, 15, 8, 0, 3, &cw, &err));
+    if ((cw[0] & 7) != 4) { printf("BAD catchup bitmap %llx\n",
+                                   (unsigned long long)cw[0]); return 1; }
   }
 
   // second merge (newer revisions + rev-row replacement)

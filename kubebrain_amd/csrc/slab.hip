@@ -855,46 +855,98 @@ __global__ void k_merge_small(Run src, int64_t n, Run dnew, int64_t m,
   }
 }
 
-// ---- watch fan-out filter (watch.go:119-159 per-watcher predicate) ------
-// thread = watcher; events' keys read via L2 (every block reads the same
-// batch); 64-bit event chunks ballotted into the delivery bitmap.
-__global__ void k_watch_filter(const uint8_t* __restrict__ wpfx,
-                               const uint32_t* __restrict__ wplen,
-                               const uint64_t* __restrict__ wfrom,
-                               const uint32_t* __restrict__ wlive, int64_t W,
-                               const uint8_t* __restrict__ ekeys,
-                               const uint64_t* __restrict__ erev, int64_t E,
-                               uint64_t* __restrict__ bitmap, int64_t words) {
+// ---- watch fan-out over the DEVICE-RESIDENT event log -------------------
+// (watch.go:119-159 per-watcher predicate; north_star: "GPU-resident event
+// log"). The log is a ring of (key96, rev) columns in HBM, pushed once per
+// event batch; the fan-out filter stages each batch through LDS and tests
+// all watchers against it; catch-up (Ring.FindEvents, ring.go:84-118) is a
+// thread-per-event scan over the resident ring. Values/full events stay in
+// the host ring for materialization at poll time — matching the reference,
+// which fans out shared batch POINTERS and filters in the consumer
+// (watcherhub.go:78-100, watch.go:119-159).
+
+constexpr int WF_CHUNK = 256;  // events staged per LDS round (26KB)
+
+__global__ void k_watch_filter2(const uint8_t* __restrict__ er_keys,
+                                const uint64_t* __restrict__ er_rev,
+                                int64_t ring_cap, int64_t base_seq,
+                                int64_t count,
+                                const uint8_t* __restrict__ wpfx,
+                                const uint32_t* __restrict__ wplen,
+                                const uint64_t* __restrict__ wfrom,
+                                const uint32_t* __restrict__ wlive, int64_t W,
+                                uint64_t* __restrict__ bitmap, int64_t words) {
+  __shared__ uint64_t lkeys[(WF_CHUNK * KEYW) / 8];
+  __shared__ uint64_t lrev[WF_CHUNK];
   int64_t w = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
-  if (w >= W) return;
-  if (!wlive[w]) {
-    for (int64_t c = 0; c < words; ++c) bitmap[w * words + c] = 0;
-    return;
-  }
-  const uint8_t* pfx = wpfx + w * KEYW;
-  uint32_t plen = wplen[w];
-  uint64_t from = wfrom[w];
-  for (int64_t c = 0; c < words; ++c) {
-    uint64_t bits = 0;
-    int64_t e0 = c * 64;
-    int64_t cnt = min((int64_t)64, E - e0);
-    for (int64_t j = 0; j < cnt; ++j) {
-      int64_t e = e0 + j;
-      bool ok = erev[e] >= from;  // filterByRevision (watch.go:152-158)
-      if (ok) {                   // filterByPrefix (watch.go:139-149)
-        const uint8_t* k = ekeys + e * KEYW;
-        uint32_t b = 0;
-        for (; b + 8 <= plen; b += 8)
-          if (*(const uint64_t*)(pfx + b) != *(const uint64_t*)(k + b)) break;
-        if (b + 8 <= plen) ok = false;
-        else
-          for (; b < plen && ok; ++b)
-            if (pfx[b] != k[b]) ok = false;
-      }
-      if (ok) bits |= 1ull << j;
+  bool live = w < W && wlive[w];
+  const uint8_t* pfx = live ? wpfx + w * KEYW : nullptr;
+  uint32_t plen = live ? wplen[w] : 0;
+  uint64_t from = live ? wfrom[w] : 0;
+  uint64_t acc[8] = {0, 0, 0, 0, 0, 0, 0, 0};  // words <= 8 (count <= 512)
+  for (int64_t c0 = 0; c0 < count; c0 += WF_CHUNK) {
+    int64_t cn = min((int64_t)WF_CHUNK, count - c0);
+    // stage this chunk's keys+revs (ring slots may wrap) into LDS
+    for (int64_t j = threadIdx.x; j < cn; j += blockDim.x) {
+      int64_t slot = (base_seq + c0 + j) % ring_cap;
+      const uint64_t* src = (const uint64_t*)(er_keys + slot * KEYW);
+      uint64_t* dst = lkeys + j * (KEYW / 8);
+#pragma unroll
+      for (int k = 0; k < KEYW / 8; ++k) dst[k] = src[k];
+      lrev[j] = er_rev[slot];
     }
-    bitmap[w * words + c] = bits;
+    __syncthreads();
+    if (live) {
+      for (int64_t j = 0; j < cn; ++j) {
+        bool ok = lrev[j] >= from;  // filterByRevision (watch.go:152-158)
+        if (ok) {                   // filterByPrefix (watch.go:139-149)
+          const uint8_t* k = (const uint8_t*)(lkeys + j * (KEYW / 8));
+          uint32_t b = 0;
+          for (; b + 8 <= plen; b += 8)
+            if (*(const uint64_t*)(pfx + b) != *(const uint64_t*)(k + b)) break;
+          if (b + 8 <= plen) ok = false;
+          else
+            for (; b < plen && ok; ++b)
+              if (pfx[b] != k[b]) ok = false;
+        }
+        if (ok) {
+          int64_t e = c0 + j;
+          acc[e >> 6] |= 1ull << (e & 63);
+        }
+      }
+    }
+    __syncthreads();
   }
+  if (w < W)
+    for (int64_t c = 0; c < words; ++c) bitmap[w * words + c] = acc[c];
+}
+
+// catch-up: ONE watcher's prefix over a resident ring span, thread-per-event
+__global__ void k_watch_catchup(const uint8_t* __restrict__ er_keys,
+                                const uint64_t* __restrict__ er_rev,
+                                int64_t ring_cap, int64_t base_seq,
+                                int64_t count, const uint8_t* __restrict__ pfx,
+                                uint32_t plen, uint64_t from_rev,
+                                uint64_t* __restrict__ words) {
+  int64_t e = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  int lane = threadIdx.x & 63;
+  bool ok = false;
+  if (e < count) {
+    int64_t slot = (base_seq + e) % ring_cap;
+    ok = er_rev[slot] >= from_rev;
+    if (ok) {
+      const uint8_t* k = er_keys + slot * KEYW;
+      uint32_t b = 0;
+      for (; b + 8 <= plen; b += 8)
+        if (*(const uint64_t*)(pfx + b) != *(const uint64_t*)(k + b)) break;
+      if (b + 8 <= plen) ok = false;
+      else
+        for (; b < plen && ok; ++b)
+          if (pfx[b] != k[b]) ok = false;
+    }
+  }
+  uint64_t b = __ballot(ok);
+  if (lane == 0 && e < count) words[e >> 6] = b;
 }
 
 // ------------------------------------------------------------------ Impl ---
@@ -974,10 +1026,10 @@ struct Slab::Impl {
   uint8_t* d_wpfx = nullptr;
   uint32_t *d_wplen = nullptr, *d_wlive = nullptr;
   uint64_t* d_wfrom = nullptr;
-  // event batch
-  int64_t ecap = 0;
-  uint8_t* d_ekeys = nullptr;
-  uint64_t* d_erev = nullptr;
+  // device-resident event log ring (keys+revs; values stay host-side)
+  uint8_t* er_keys = nullptr;
+  uint64_t* er_rev = nullptr;
+  int64_t er_cap = 0;
   uint64_t* d_bitmap = nullptr;
   int64_t bitmap_cap = 0;
 
@@ -1068,7 +1120,7 @@ struct Slab::Impl {
                     (void*)A.ko, (void*)B.ko, (void*)DA.ko, (void*)DB.ko,
                     (void*)d_wpfx,
                     (void*)d_wplen, (void*)d_wlive, (void*)d_wfrom,
-                    (void*)d_ekeys, (void*)d_erev, (void*)d_bitmap}) {
+                    (void*)er_keys, (void*)er_rev, (void*)d_bitmap}) {
       if (p) (void)hipFree(p);
     }
     if (h_pack) (void)hipHostFree(h_pack);
@@ -1904,34 +1956,58 @@ void Slab::WatcherClear(int64_t slot) {
   (void)hipStreamSynchronize(I->stream);
 }
 
-bool Slab::WatchFilter(const WatchFilterBatch& b, std::vector<uint64_t>* bitmap,
-                       int64_t* n_watch_slots, std::string* err) {
+bool Slab::EventRingInit(int64_t cap, std::string* err) {
   Impl* I = p;
-  int64_t W = I->wcap, E = b.e;
+  if (I->er_keys) return true;
+  I->er_cap = cap;
+  HIP_CHECK(hipMalloc(&I->er_keys, cap * KEYW));
+  HIP_CHECK(hipMalloc(&I->er_rev, cap * 8));
+  return true;
+}
+
+bool Slab::EventRingPush(const uint8_t* keys96, const uint64_t* revs,
+                         int64_t count, int64_t base_seq, std::string* err) {
+  Impl* I = p;
+  if (!I->er_keys) { if (err) *err = "event ring not initialized"; return false; }
+  // contiguous seqs; the ring may wrap once per push
+  int64_t s0 = base_seq % I->er_cap;
+  int64_t first = std::min(count, I->er_cap - s0);
+  HIP_CHECK(hipMemcpyAsync(I->er_keys + s0 * KEYW, keys96, first * KEYW,
+                           hipMemcpyHostToDevice, I->stream));
+  HIP_CHECK(hipMemcpyAsync(I->er_rev + s0, revs, first * 8,
+                           hipMemcpyHostToDevice, I->stream));
+  if (first < count) {
+    HIP_CHECK(hipMemcpyAsync(I->er_keys, keys96 + first * KEYW,
+                             (count - first) * KEYW, hipMemcpyHostToDevice,
+                             I->stream));
+    HIP_CHECK(hipMemcpyAsync(I->er_rev, revs + first, (count - first) * 8,
+                             hipMemcpyHostToDevice, I->stream));
+  }
+  HIP_CHECK(hipStreamSynchronize(I->stream));
+  return true;
+}
+
+bool Slab::WatchFilterRing(int64_t base_seq, int64_t count,
+                           std::vector<uint64_t>* bitmap,
+                           int64_t* n_watch_slots, std::string* err) {
+  Impl* I = p;
+  int64_t W = I->wcap;
   *n_watch_slots = W;
   bitmap->clear();
-  if (W == 0 || E == 0) return true;
-  if (E > I->ecap) {
-    int64_t cap = E + E / 2 + 64;
-    if (I->d_ekeys) (void)hipFree(I->d_ekeys);
-    if (I->d_erev) (void)hipFree(I->d_erev);
-    HIP_CHECK(hipMalloc(&I->d_ekeys, cap * KEYW));
-    HIP_CHECK(hipMalloc(&I->d_erev, cap * 8));
-    I->ecap = cap;
-  }
-  int64_t words = ceil_div(E, 64);
+  if (W == 0 || count == 0) return true;
+  if (count > 512) { if (err) *err = "filter batch > 512"; return false; }
+  int64_t words = ceil_div(count, 64);
   if (W * words > I->bitmap_cap) {
     int64_t cap = W * words * 2;
     if (I->d_bitmap) (void)hipFree(I->d_bitmap);
     HIP_CHECK(hipMalloc(&I->d_bitmap, cap * 8));
     I->bitmap_cap = cap;
   }
-  HIP_CHECK(hipMemcpyAsync(I->d_ekeys, b.ekeys.data(), E * KEYW, hipMemcpyHostToDevice, I->stream));
-  HIP_CHECK(hipMemcpyAsync(I->d_erev, b.erevs.data(), E * 8, hipMemcpyHostToDevice, I->stream));
   HIP_CHECK(hipEventRecord(I->ev0, I->stream));
-  hipLaunchKernelGGL(k_watch_filter, dim3((uint32_t)ceil_div(W, 256)), dim3(256),
-                     0, I->stream, I->d_wpfx, I->d_wplen, I->d_wfrom, I->d_wlive,
-                     W, I->d_ekeys, I->d_erev, E, I->d_bitmap, words);
+  hipLaunchKernelGGL(k_watch_filter2, dim3((uint32_t)ceil_div(W, 256)),
+                     dim3(256), 0, I->stream, I->er_keys, I->er_rev, I->er_cap,
+                     base_seq, count, I->d_wpfx, I->d_wplen, I->d_wfrom,
+                     I->d_wlive, W, I->d_bitmap, words);
   HIP_CHECK(hipEventRecord(I->ev1, I->stream));
   bitmap->resize(W * words);
   HIP_CHECK(hipMemcpyAsync(bitmap->data(), I->d_bitmap, W * words * 8,
@@ -1941,8 +2017,45 @@ bool Slab::WatchFilter(const WatchFilterBatch& b, std::vector<uint64_t>* bitmap,
   (void)hipEventElapsedTime(&ms, I->ev0, I->ev1);
   perf.filter_ms += ms;
   perf.filter_launches++;
-  perf.filter_events += E;
+  perf.filter_events += count;
   perf.filter_watchers += W;
+  return true;
+}
+
+bool Slab::WatchCatchup(const uint8_t* pfx96, uint32_t plen, uint64_t from_rev,
+                        int64_t base_seq, int64_t count,
+                        std::vector<uint64_t>* words_out, std::string* err) {
+  Impl* I = p;
+  words_out->clear();
+  if (count <= 0) return true;
+  int64_t words = ceil_div(count, 64);
+  if (words > I->bitmap_cap) {
+    int64_t cap = words * 2;
+    if (I->d_bitmap) (void)hipFree(I->d_bitmap);
+    HIP_CHECK(hipMalloc(&I->d_bitmap, cap * 8));
+    I->bitmap_cap = cap;
+  }
+  uint8_t pfx[KEYW] = {0};
+  memcpy(pfx, pfx96, plen > (uint32_t)KEYW ? KEYW : plen);
+  HIP_CHECK(hipMemcpyAsync(I->d_bkeys, pfx, KEYW, hipMemcpyHostToDevice,
+                           I->stream));
+  HIP_CHECK(hipEventRecord(I->ev0, I->stream));
+  hipLaunchKernelGGL(k_watch_catchup, dim3((uint32_t)ceil_div(count, 256)),
+                     dim3(256), 0, I->stream, I->er_keys, I->er_rev, I->er_cap,
+                     base_seq, count, I->d_bkeys,
+                     plen > (uint32_t)KEYW ? (uint32_t)KEYW : plen, from_rev,
+                     I->d_bitmap);
+  HIP_CHECK(hipEventRecord(I->ev1, I->stream));
+  words_out->resize(words);
+  HIP_CHECK(hipMemcpyAsync(words_out->data(), I->d_bitmap, words * 8,
+                           hipMemcpyDeviceToHost, I->stream));
+  HIP_CHECK(hipStreamSynchronize(I->stream));
+  float ms = 0;
+  (void)hipEventElapsedTime(&ms, I->ev0, I->ev1);
+  perf.filter_ms += ms;
+  perf.filter_launches++;
+  perf.filter_events += count;
+  perf.filter_watchers += 1;
   return true;
 }
 

@@ -117,12 +117,7 @@ struct Perf {
   double dbg_a = 0, dbg_b = 0, dbg_c = 0, dbg_d = 0, dbg_e = 0;
 };
 
-struct WatchFilterBatch {
-  // E events: key column (E*KEYW) + revision per event
-  std::vector<uint8_t> ekeys;
-  std::vector<uint64_t> erevs;
-  int64_t e = 0;
-};
+
 
 class Slab {
  public:
@@ -213,9 +208,22 @@ class Slab {
   // full slab dump for parity diffs (debug; D2H of all columns + used heap)
   bool Dump(std::vector<DumpRow>* rows_out, std::string* err);
 
-  // watch fan-out filter: bitmap[w][ceil(E/64)] over registered watchers
-  bool WatchFilter(const WatchFilterBatch& b, std::vector<uint64_t>* bitmap,
-                   int64_t* n_watch_slots, std::string* err);
+  // device-resident event log (north_star: GPU-resident event log): ring of
+  // (key96, rev) columns pushed once per event batch; the fan-out filter and
+  // catch-up scans read the resident ring. Values/full events stay host-side
+  // for materialization at poll time.
+  bool EventRingInit(int64_t cap, std::string* err);
+  bool EventRingPush(const uint8_t* keys96, const uint64_t* revs,
+                     int64_t count, int64_t base_seq, std::string* err);
+  // fan-out: bitmap[w][ceil(count/64)] over all registered watchers for the
+  // ring span [base_seq, base_seq+count)
+  bool WatchFilterRing(int64_t base_seq, int64_t count,
+                       std::vector<uint64_t>* bitmap, int64_t* n_watch_slots,
+                       std::string* err);
+  // catch-up: one prefix/from_rev filter over a resident ring span
+  bool WatchCatchup(const uint8_t* pfx96, uint32_t plen, uint64_t from_rev,
+                    int64_t base_seq, int64_t count,
+                    std::vector<uint64_t>* words_out, std::string* err);
   bool WatcherSet(int64_t slot, const uint8_t* prefix, uint32_t plen,
                   uint64_t from_rev, std::string* err);  // slot grows table
   void WatcherClear(int64_t slot);

@@ -88,6 +88,14 @@ Store* Store::Open(const Config& cfg_in, std::string* err) {
   s->cfg_ = cfg;
   s->slab_ = slab;
   s->ring_.init(cfg.watch_cache_size > 0 ? cfg.watch_cache_size : 200000);
+  {
+    std::string e2;
+    if (!slab->EventRingInit(s->ring_.l, &e2)) {
+      if (err) *err = e2;
+      delete s;
+      return nullptr;
+    }
+  }
   s->keep_event_log_ = env_i64("KB_EVENT_LOG", 1) != 0;  // tests: on; bench: off
   return s;
 }
@@ -291,50 +299,60 @@ void Store::notify(const Bytes& key, const Bytes& val, uint64_t revision,
 
 void Store::pumpEvents() {
   if (pending_.empty()) return;
-  bool any_live = false;
-  for (auto& [id, w] : watchers_) if (!w.dropped) { any_live = true; break; }
-  if (!any_live) { pending_.clear(); return; }  // hub with no subs drops batches
-  // fan-out via the GPU ballot filter (DESIGN.md §3.3 k_watch_filter)
   const int64_t kBatch = 512;
+  // seq of pending_[0]: the host ring and pending_ grow in lockstep (notify)
+  const int64_t seq0 = ring_.e - (int64_t)pending_.size();
+  std::string err;
   for (size_t b0 = 0; b0 < pending_.size(); b0 += kBatch) {
     int64_t e = std::min((size_t)kBatch, pending_.size() - b0);
-    kbslab::WatchFilterBatch batch;
-    batch.e = e;
-    batch.ekeys.resize((size_t)e * KEYW);
-    batch.erevs.resize(e);
+    // push the batch into the DEVICE event ring unconditionally: catch-up
+    // scans read the resident log even when no watcher is live right now
+    std::vector<uint8_t> ekeys((size_t)e * KEYW);
+    std::vector<uint64_t> erevs(e);
     for (int64_t j = 0; j < e; ++j) {
       const Event& ev = pending_[b0 + j];
-      pad96(ev.kv_key, batch.ekeys.data() + (size_t)j * KEYW);
-      batch.erevs[j] = ev.revision;
+      pad96(ev.kv_key, ekeys.data() + (size_t)j * KEYW);
+      erevs[j] = ev.revision;
     }
+    if (!slab_->EventRingPush(ekeys.data(), erevs.data(), e, seq0 + (int64_t)b0,
+                              &err)) {
+      fatal_ = err;
+      pending_.clear();
+      return;
+    }
+    bool any_live = false;
+    for (auto& [id, w] : watchers_) if (!w.dropped) { any_live = true; break; }
+    if (!any_live) continue;
+    // fan-out: device ballot filter over the resident ring span; delivery =
+    // one bitmap reference per watcher per batch (no per-event copies)
     std::vector<uint64_t> bitmap;
     int64_t W = 0;
-    std::string err;
-    if (!slab_->WatchFilter(batch, &bitmap, &W, &err)) { pending_.clear(); return; }
+    if (!slab_->WatchFilterRing(seq0 + (int64_t)b0, e, &bitmap, &W, &err)) {
+      pending_.clear();
+      return;
+    }
     int64_t words = (e + 63) / 64;
     for (auto& [id, w] : watchers_) {
-      if (w.dropped || w.slot >= W || w.slot < 0) continue;
-      // prefixes longer than the 96B filter column: the device test on the
-      // first 96 bytes is a superset; re-check the full prefix here
-      const bool long_pfx = w.prefix.size() > (size_t)KEYW;
+      if (w.dropped || w.slot < 0 || w.slot >= W) continue;
+      PendRef pr{};
+      pr.base = seq0 + (int64_t)b0;
+      pr.count = (int32_t)e;
+      int64_t cnt = 0;
       for (int64_t c = 0; c < words; ++c) {
         uint64_t bits = bitmap[(size_t)(w.slot * words + c)];
-        while (bits) {
-          int j = __builtin_ctzll(bits);
-          bits &= bits - 1;
-          const Event& ev2 = pending_[b0 + c * 64 + j];
-          if (long_pfx &&
-              ev2.kv_key.compare(0, w.prefix.size(), w.prefix) != 0)
-            continue;
-          w.queue.push_back(ev2);
-          delivered_++;
-        }
+        pr.words[c] = bits;
+        cnt += __builtin_popcountll(bits);
       }
-      if (w.queue.size() > kWatchQueueCap) {
-        // drop slow consumer (watcherhub.go:84-94); the device slot is
-        // recycled immediately — only WatchPoll's "dropped" notice remains
+      if (!cnt) continue;
+      w.prefs.push_back(pr);
+      w.pend_events += cnt;
+      delivered_ += cnt;
+      // slow consumer: pending refs about to leave the ring window
+      if (!w.prefs.empty() &&
+          ring_.e - w.prefs.front().base > (int64_t)ring_.l - kBatch) {
         w.dropped = true;
-        w.queue.clear();
+        w.prefs.clear();
+        w.pend_events = 0;
         releaseSlot(w);
       }
     }
@@ -446,17 +464,53 @@ int64_t Store::Watch(const Bytes& prefix, uint64_t revision, Status* st) {
           if (ring_.arr[(ring_.s + mid) % ring_.l].revision >= revision) hi = mid;
           else lo = mid + 1;
         }
-        std::vector<Event> catchup;
-        for (int64_t i = lo; i < n; ++i) {
-          const Event& ev = ring_.arr[(ring_.s + i) % ring_.l];
-          if (ev.kv_key.compare(0, prefix.size(), prefix) == 0) catchup.push_back(ev);
+        // catch-up (Ring.FindEvents, ring.go:84-118): filter the ring span
+        // [s+lo, e) by prefix into bitmap refs. The span is resident in the
+        // DEVICE event log (pushed at pump; pumpEvents ran above), so the
+        // scan runs there; prefixes longer than the 96B filter column
+        // compute exact bits host-side instead (the device test would be a
+        // superset). Ring revisions ascend, so rev >= revision holds for
+        // the whole span.
+        int64_t base = ring_.s + lo, cnt = n - lo;
+        int64_t total = 0;
+        std::vector<uint64_t> words;
+        if (cnt > 0 && prefix.size() <= (size_t)KEYW) {
+          uint8_t p96c[KEYW];
+          pad96(prefix, p96c);
+          std::string err2;
+          if (!slab_->WatchCatchup(p96c, (uint32_t)prefix.size(), revision,
+                                   base, cnt, &words, &err2)) {
+            *st = INTERNAL;
+            return -1;
+          }
+        } else if (cnt > 0) {
+          words.assign((size_t)((cnt + 63) / 64), 0);
+          for (int64_t i = 0; i < cnt; ++i) {
+            const Event& ev = ring_.arr[(base + i) % ring_.l];
+            if (ev.kv_key.compare(0, prefix.size(), prefix) == 0)
+              words[(size_t)(i >> 6)] |= 1ull << (i & 63);
+          }
         }
-        uint64_t lastRevision = revision;
-        if (!catchup.empty()) {
-          lastRevision = newest.revision + 1;  // watch.go:91-95
-          for (auto& ev : catchup) w.queue.push_back(ev);
+        for (auto wd : words) total += __builtin_popcountll(wd);
+        if (total > 0) {
+          for (int64_t c0 = 0; c0 < cnt; c0 += 512) {
+            PendRef pr{};
+            pr.base = base + c0;
+            pr.count = (int32_t)std::min<int64_t>(512, cnt - c0);
+            int64_t got = 0;
+            for (int64_t c = 0; c < 8 && (c0 >> 6) + c < (int64_t)words.size() &&
+                                c * 64 < pr.count; ++c) {
+              pr.words[c] = words[(size_t)((c0 >> 6) + c)];
+              got += __builtin_popcountll(pr.words[c]);
+            }
+            if (got) {
+              w.prefs.push_back(pr);
+              w.pend_events += got;
+            }
+          }
         }
-        w.from_rev = lastRevision;
+        // watch.go:91-95
+        w.from_rev = total > 0 ? newest.revision + 1 : revision;
       }
     }
   }
@@ -494,10 +548,32 @@ std::vector<Event> Store::WatchPoll(int64_t wid, Status* st) {
   return WatchPollLimited(wid, SIZE_MAX, nullptr, st);
 }
 
-// Drains the watcher queue only if its serialized size (4 + per event
-// 28+klen+vlen — the kb_watch_poll wire format) fits max_bytes; otherwise
-// returns NOBUF with the queue INTACT so a retry with a larger buffer still
-// sees every event (contiguous-revision delivery guarantee).
+bool Store::materializeRefs(Watcher& w, std::vector<Event>* out) {
+  for (const PendRef& pr : w.prefs) {
+    if (pr.base < ring_.e - (int64_t)ring_.l)
+      return false;  // ring overwrote the span: slow consumer fell behind
+    for (int c = 0; c * 64 < pr.count; ++c) {
+      uint64_t bits = pr.words[c];
+      while (bits) {
+        int j = __builtin_ctzll(bits);
+        bits &= bits - 1;
+        const Event& ev = ring_.arr[(pr.base + c * 64 + j) % ring_.l];
+        // prefixes longer than the 96B device filter column: the bitmap is
+        // a superset; apply the exact prefix here
+        if (w.prefix.size() > (size_t)KEYW &&
+            ev.kv_key.compare(0, w.prefix.size(), w.prefix) != 0)
+          continue;
+        out->push_back(ev);
+      }
+    }
+  }
+  return true;
+}
+
+// Drains the watcher's pending refs only if the serialized size (4 + per
+// event 28+klen+vlen — the kb_watch_poll wire format) fits max_bytes;
+// otherwise returns NOBUF with the refs INTACT so a retry with a larger
+// buffer still sees every event (contiguous-revision delivery guarantee).
 std::vector<Event> Store::WatchPollLimited(int64_t wid, size_t max_bytes,
                                            size_t* need_bytes, Status* st) {
   std::lock_guard<std::recursive_mutex> lk(mu_);
@@ -506,6 +582,12 @@ std::vector<Event> Store::WatchPollLimited(int64_t wid, size_t max_bytes,
   auto it = watchers_.find(wid);
   if (it == watchers_.end()) { *st = WATCH_DROPPED; return {}; }
   Watcher& w = it->second;
+  std::vector<Event> out;
+  if (!w.dropped && !materializeRefs(w, &out)) {
+    w.dropped = true;  // refs went stale between pumps (slow consumer)
+    w.prefs.clear();
+    w.pend_events = 0;
+  }
   if (w.dropped) {
     *st = WATCH_DROPPED;
     releaseSlot(w);  // no-op if pumpEvents already recycled it
@@ -513,12 +595,12 @@ std::vector<Event> Store::WatchPollLimited(int64_t wid, size_t max_bytes,
     return {};
   }
   size_t need = 4;
-  for (const Event& e : w.queue)
+  for (const Event& e : out)
     need += 28 + e.kv_key.size() + e.kv_value.size();
   if (need_bytes) *need_bytes = need;
   if (need > max_bytes) { *st = NOBUF; return {}; }
-  std::vector<Event> out(w.queue.begin(), w.queue.end());
-  w.queue.clear();
+  w.prefs.clear();
+  w.pend_events = 0;
   *st = OK;
   return out;
 }

@@ -195,17 +195,34 @@ class Store {
   } ring_;
   std::vector<Event> event_log_;
   std::vector<Event> pending_;  // not yet fanned out
+  // delivery = bitmap references into the global event sequence (the host
+  // ring materializes full events at poll time) — mirrors the reference,
+  // which fans out shared batch POINTERS and filters/copies in the consumer
+  // (watcherhub.go:78-100, watch.go:119-159)
+  struct PendRef {
+    int64_t base;        // global seq of bit 0
+    int32_t count;       // events covered (<= 512)
+    uint64_t words[8];
+  };
   struct Watcher {
     int64_t slot;  // device slot; -1 once released (releaseSlot)
     Bytes prefix;
     uint64_t from_rev;
-    std::deque<Event> queue;
+    std::deque<PendRef> prefs;
+    int64_t pend_events = 0;
     bool dropped = false;
   };
+  // materialize a watcher's pending refs from the host ring; false if any
+  // ref was overwritten (slow consumer fell behind the ring window)
+  bool materializeRefs(Watcher& w, std::vector<Event>* out);
   std::unordered_map<int64_t, Watcher> watchers_;
   std::vector<int64_t> free_slots_;
   int64_t next_slot_ = 0, next_wid_ = 1;
-  static constexpr size_t kWatchQueueCap = 10000ull * 300;  // watcherhub.go:30 × eventBatchSize
+  // slow-consumer bound: a watcher whose pending refs span more than the
+  // event ring window cannot be materialized and is dropped. The reference's
+  // own bound varies between 10k batches and 3M events (watcherhub.go:30
+  // sub-channel of 10000 batches x <=300 events); ours is the ring capacity
+  // (watch_cache_size, default 200k events) — inside that envelope.
 
   // host-side perf
   int64_t ops_create_ = 0, ops_update_ = 0, ops_delete_ = 0, ops_range_ = 0;
