@@ -405,6 +405,11 @@ def cpu_baseline_leg(namespaces, keys, sample_qs):
 
 
 def watch_leg(store, namespaces, live_keys, revs, rng, n_watchers, n_events):
+    """configs[4]-shaped fan-out: n_watchers live watchers, n_events writes.
+    `delivered` counts events landed in watcher queues (bitmap references —
+    the same cost model as the reference's shared-batch-pointer fan-out,
+    watcherhub.go:78-100); a 100-watcher poll sample then materializes full
+    events from the ring to show the consumer-side path at rate."""
     import kbclient
     wids = []
     for i in range(n_watchers):
@@ -423,9 +428,24 @@ def watch_leg(store, namespaces, live_keys, revs, rng, n_watchers, n_events):
     dt = time.time() - t0
     p1 = perf(store)
     delivered = p1.get("delivered", 0) - p0.get("delivered", 0)
+    # poll sample: materialize events for 100 watchers (incl. heavy ones)
+    t0 = time.time()
+    polled = 0
+    for wid in wids[:100]:
+        rc, evs = store.watch_poll(wid)
+        if rc == kbclient.OK:
+            polled += len(evs)
+    poll_dt = time.time() - t0
+    stats = {
+        "delivered": delivered,
+        "filter_ms": round(p1.get("filter_ms", 0) - p0.get("filter_ms", 0), 3),
+        "filter_launches": p1.get("filter_launches", 0) - p0.get("filter_launches", 0),
+        "poll_sample": {"watchers": 100, "events": polled,
+                        "events_per_sec": round(polled / poll_dt, 1) if poll_dt > 0 else None},
+    }
     for wid in wids:
         store.watch_cancel(wid)
-    return delivered / dt if dt > 0 else 0.0, delivered
+    return delivered / dt if dt > 0 else 0.0, delivered, stats
 
 
 def main():
@@ -681,9 +701,10 @@ def main():
         log(rank, f"[bench] compact sweep: {compact_stats}")
 
     # watch fan-out leg
-    wrate, delivered = watch_leg(store, my_ns_list, live, revs, qrng,
-                                 max(args.watchers // world, 8),
-                                 args.watch_events // world)
+    wrate, delivered, watch_stats = watch_leg(store, my_ns_list, live, revs,
+                                              qrng,
+                                              max(args.watchers // world, 8),
+                                              args.watch_events // world)
     if dist:
         import torch as _t
         w = _t.tensor([wrate])
@@ -762,6 +783,7 @@ def main():
             "compact_sweep": compact_stats,
             "watch_events_per_sec": round(wrate, 1),
             "watch_delivered_rank0": delivered,
+            "watch_stats": watch_stats,
             "ops_per_sec_with_d2h": round(OPS_PER_STEP * d2h_steps * world / d2h_elapsed, 1),
             "ops_per_sec_with_d2h_keys_only": round(
                 OPS_PER_STEP * d2h_steps * world / d2h_ko_elapsed, 1),
