@@ -522,11 +522,12 @@ __global__ void k_gather_copy(Run rb, Run rd,
       ((uint32_t*)dst)[3] = vlen;
     }
     const uint8_t* ks = r.keys + row * KEYW;
-    uint8_t* kd = dst + 16;
+    uint8_t* kd = dst + 16;  // 8-aligned (records are 16B-aligned)
     uint32_t kin = klen > (uint32_t)KEYW ? (uint32_t)KEYW : klen;
-    uint32_t kw = (kin + 3) & ~3u;  // the 96B column part (4-aligned reads)
-    for (uint32_t b = gl; b < kw / 4; b += 16)
-      ((uint32_t*)kd)[b] = ((const uint32_t*)ks)[b];
+    // column rows are 96B-strided => 8-aligned; copy whole u64s (may round
+    // up to 7B into the record's 16B key padding — always in-bounds)
+    for (uint32_t b = gl; b < (kin + 7) / 8; b += 16)
+      ((uint64_t*)kd)[b] = ((const uint64_t*)ks)[b];
     if (klen > (uint32_t)KEYW) {  // spill tail (keys > 96B)
       const uint8_t* ts = spill + r.ko[row];
       for (uint32_t t = gl; t < klen - (uint32_t)KEYW; t += 16)
