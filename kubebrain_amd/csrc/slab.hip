@@ -226,27 +226,44 @@ __device__ int64_t scan_run_winners(
     const int64_t wbase = t + (int64_t)w * C;
     int rounds = 0;
     if (wbase < hi) rounds = (int)((min(C, hi - wbase) + 63) >> 6);
-    // pass 1: predicate + winner flags (register-only, no syncs)
+    // pass 1a: streaming winner predicate — rev/meta 8B streams only, no
+    // probes and no divergent chains, so the loads pipeline across rounds.
+    // same_next set => row i+1 exists, shares the key, and (rows of one key
+    // being contiguous and its key < qend) lies below hi, so the rev[i+1]
+    // load is gated by i+1<hi alone, independent of the meta bits.
     uint64_t flags = 0;
-    uint32_t wcnt = 0;
+#pragma unroll 4
     for (int r = 0; r < rounds; ++r) {
       int64_t i = wbase + ((int64_t)r << 6) + lane;
       bool win = false;
       if (i < hi) {
         uint64_t rv = rev[i], m = meta[i];
+        uint64_t rvn = (i + 1 < hi) ? rev[i + 1] : 0;
         if (rv > 0 && rv <= R && !(m & M_TOMB))
-          // same_next set => row i+1 exists and shares the key
-          win = !(m & M_SAME_NEXT) || rev[i + 1] > R;
-        if (win && srun) {
+          win = !(m & M_SAME_NEXT) || rvn > R;
+      }
+      if (win) flags |= 1ull << r;
+    }
+    // pass 1b: delta-suppression probes on flagged rows only (~cap of the
+    // 6*cap-row tile), so the random binary searches no longer stall the
+    // 16 B/row stream of pass 1a
+    uint32_t wcnt = 0;
+    for (int r = 0; r < rounds; ++r) {
+      bool win = (flags >> r) & 1;
+      uint64_t b = __ballot(win);
+      if (srun && b) {
+        if (win) {
+          int64_t i = wbase + ((int64_t)r << 6) + lane;
           QKey kk = row_qk(run, spill, i);
           int64_t lb = d_lb_range(*srun, spill, slo, shi, kk, 1);
           if (lb < shi && srun->rev[lb] <= R &&
-              rowcmp_q(*srun, spill, lb, kk) == 0)
+              rowcmp_q(*srun, spill, lb, kk) == 0) {
             win = false;  // a newer (delta) row of this key wins instead
+            flags &= ~(1ull << r);
+          }
         }
+        b = __ballot(win);
       }
-      uint64_t b = __ballot(win);
-      if (win) flags |= 1ull << r;
       wcnt += (uint32_t)__popcll(b);
     }
     if (lane == 0) wave_cnt[w] = (int)wcnt;
@@ -866,8 +883,12 @@ __global__ void k_merge_small(Run src, int64_t n, Run dnew, int64_t m,
 // which fans out shared batch POINTERS and filters in the consumer
 // (watcherhub.go:78-100, watch.go:119-159).
 
-constexpr int WF_CHUNK = 256;  // events staged per LDS round (26KB)
-
+// one 64-thread block per (64-watcher slice x 64-event chunk): at 10k
+// watchers / 512 events that is 157x8 = 1256 single-wave blocks (the old
+// one-thread-per-watcher shape launched only W/256 = 40 blocks and left the
+// chip idle). Each block stages its event chunk (key96+rev, 6.5 KB) in LDS;
+// each thread (= one watcher) tests the 64 events and emits exactly one
+// bitmap word — no atomics, no cross-wave syncs, no ballot needed.
 __global__ void k_watch_filter2(const uint8_t* __restrict__ er_keys,
                                 const uint64_t* __restrict__ er_rev,
                                 int64_t ring_cap, int64_t base_seq,
@@ -877,49 +898,45 @@ __global__ void k_watch_filter2(const uint8_t* __restrict__ er_keys,
                                 const uint64_t* __restrict__ wfrom,
                                 const uint32_t* __restrict__ wlive, int64_t W,
                                 uint64_t* __restrict__ bitmap, int64_t words) {
-  __shared__ uint64_t lkeys[(WF_CHUNK * KEYW) / 8];
-  __shared__ uint64_t lrev[WF_CHUNK];
-  int64_t w = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
-  bool live = w < W && wlive[w];
-  const uint8_t* pfx = live ? wpfx + w * KEYW : nullptr;
-  uint32_t plen = live ? wplen[w] : 0;
-  uint64_t from = live ? wfrom[w] : 0;
-  uint64_t acc[8] = {0, 0, 0, 0, 0, 0, 0, 0};  // words <= 8 (count <= 512)
-  for (int64_t c0 = 0; c0 < count; c0 += WF_CHUNK) {
-    int64_t cn = min((int64_t)WF_CHUNK, count - c0);
-    // stage this chunk's keys+revs (ring slots may wrap) into LDS
-    for (int64_t j = threadIdx.x; j < cn; j += blockDim.x) {
-      int64_t slot = (base_seq + c0 + j) % ring_cap;
-      const uint64_t* src = (const uint64_t*)(er_keys + slot * KEYW);
-      uint64_t* dst = lkeys + j * (KEYW / 8);
-#pragma unroll
-      for (int k = 0; k < KEYW / 8; ++k) dst[k] = src[k];
-      lrev[j] = er_rev[slot];
-    }
-    __syncthreads();
-    if (live) {
-      for (int64_t j = 0; j < cn; ++j) {
-        bool ok = lrev[j] >= from;  // filterByRevision (watch.go:152-158)
-        if (ok) {                   // filterByPrefix (watch.go:139-149)
-          const uint8_t* k = (const uint8_t*)(lkeys + j * (KEYW / 8));
-          uint32_t b = 0;
-          for (; b + 8 <= plen; b += 8)
-            if (*(const uint64_t*)(pfx + b) != *(const uint64_t*)(k + b)) break;
-          if (b + 8 <= plen) ok = false;
-          else
-            for (; b < plen && ok; ++b)
-              if (pfx[b] != k[b]) ok = false;
-        }
-        if (ok) {
-          int64_t e = c0 + j;
-          acc[e >> 6] |= 1ull << (e & 63);
-        }
-      }
-    }
-    __syncthreads();
+  __shared__ uint64_t lkeys[64 * (KEYW / 8)];
+  __shared__ uint64_t lrev[64];
+  const int64_t w = (int64_t)blockIdx.x * 64 + threadIdx.x;
+  const int64_t c0 = (int64_t)blockIdx.y << 6;
+  const int64_t cn = min((int64_t)64, count - c0);
+  // stage the chunk: consecutive threads load consecutive u64s (ring slots
+  // may wrap, so the slot is recomputed per element)
+  for (int64_t v = threadIdx.x; v < cn * (KEYW / 8); v += 64) {
+    int64_t j = v / (KEYW / 8), k = v % (KEYW / 8);
+    int64_t slot = (base_seq + c0 + j) % ring_cap;
+    lkeys[v] = ((const uint64_t*)(er_keys + slot * KEYW))[k];
   }
-  if (w < W)
-    for (int64_t c = 0; c < words; ++c) bitmap[w * words + c] = acc[c];
+  if (threadIdx.x < cn) {
+    int64_t slot = (base_seq + c0 + threadIdx.x) % ring_cap;
+    lrev[threadIdx.x] = er_rev[slot];
+  }
+  __syncthreads();
+  if (w >= W) return;
+  uint64_t acc = 0;
+  if (wlive[w]) {
+    const uint8_t* pfx = wpfx + w * KEYW;
+    const uint32_t plen = wplen[w];
+    const uint64_t from = wfrom[w];
+    for (int64_t j = 0; j < cn; ++j) {
+      bool ok = lrev[j] >= from;  // filterByRevision (watch.go:152-158)
+      if (ok) {                   // filterByPrefix (watch.go:139-149)
+        const uint8_t* k = (const uint8_t*)(lkeys + j * (KEYW / 8));
+        uint32_t b = 0;
+        for (; b + 8 <= plen; b += 8)
+          if (*(const uint64_t*)(pfx + b) != *(const uint64_t*)(k + b)) break;
+        if (b + 8 <= plen) ok = false;
+        else
+          for (; b < plen && ok; ++b)
+            if (pfx[b] != k[b]) ok = false;
+      }
+      if (ok) acc |= 1ull << j;
+    }
+  }
+  bitmap[w * words + blockIdx.y] = acc;
 }
 
 // catch-up: ONE watcher's prefix over a resident ring span, thread-per-event
@@ -2005,8 +2022,9 @@ bool Slab::WatchFilterRing(int64_t base_seq, int64_t count,
     I->bitmap_cap = cap;
   }
   HIP_CHECK(hipEventRecord(I->ev0, I->stream));
-  hipLaunchKernelGGL(k_watch_filter2, dim3((uint32_t)ceil_div(W, 256)),
-                     dim3(256), 0, I->stream, I->er_keys, I->er_rev, I->er_cap,
+  hipLaunchKernelGGL(k_watch_filter2,
+                     dim3((uint32_t)ceil_div(W, 64), (uint32_t)words),
+                     dim3(64), 0, I->stream, I->er_keys, I->er_rev, I->er_cap,
                      base_seq, count, I->d_wpfx, I->d_wplen, I->d_wfrom,
                      I->d_wlive, W, I->d_bitmap, words);
   HIP_CHECK(hipEventRecord(I->ev1, I->stream));
