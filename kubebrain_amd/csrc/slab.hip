@@ -204,7 +204,8 @@ __device__ int64_t scan_run_winners(
     const Run& run, const uint8_t* spill, int64_t lo, int64_t hi, uint64_t R,
     int64_t cap, uint64_t* out, int64_t out_cap, uint64_t tagbit,
     const Run* srun, int64_t slo, int64_t shi,  // suppression run (null => none)
-    int64_t* total_out, int64_t* scanned_accum, int* wave_cnt) {
+    int64_t* total_out, int64_t* scanned_accum, int* wave_cnt,
+    unsigned long long* dbg) {  // KB_SCAN_DBG phase cycles (null in prod)
   const uint64_t* __restrict__ rev = run.rev;
   const uint64_t* __restrict__ meta = run.meta;
   const int lane = threadIdx.x & 63, w = threadIdx.x >> 6;
@@ -222,6 +223,8 @@ __device__ int64_t scan_run_winners(
   const int64_t TILE = (int64_t)NW * C;
   int64_t scanned = 0;
   int64_t cnt = 0;  // winners before this tile (uniform across the block)
+  const bool dbg0 = dbg && threadIdx.x == 0;
+  unsigned long long tprev = dbg0 ? wall_clock64() : 0;
   for (int64_t t = lo; t < hi; t += TILE) {
     const int64_t wbase = t + (int64_t)w * C;
     int rounds = 0;
@@ -244,6 +247,7 @@ __device__ int64_t scan_run_winners(
       }
       if (win) flags |= 1ull << r;
     }
+    if (dbg0) { unsigned long long t1 = wall_clock64(); atomicAdd(&dbg[1], t1 - tprev); tprev = t1; }
     // pass 1b: delta-suppression probes on flagged rows only (~cap of the
     // 6*cap-row tile), so the random binary searches no longer stall the
     // 16 B/row stream of pass 1a
@@ -266,6 +270,7 @@ __device__ int64_t scan_run_winners(
       }
       wcnt += (uint32_t)__popcll(b);
     }
+    if (dbg0) { unsigned long long t2 = wall_clock64(); atomicAdd(&dbg[2], t2 - tprev); tprev = t2; }
     if (lane == 0) wave_cnt[w] = (int)wcnt;
     __syncthreads();
     int64_t waveoff = 0, tile_total = 0;
@@ -287,6 +292,7 @@ __device__ int64_t scan_run_winners(
         done += (uint32_t)__popcll(b);
       }
     }
+    if (dbg0) { unsigned long long t3 = wall_clock64(); atomicAdd(&dbg[3], t3 - tprev); tprev = t3; }
     cnt += tile_total;
     scanned += min(TILE, hi - t);
     if (cnt >= cap) break;
@@ -320,7 +326,8 @@ __global__ void k_range_scan2(
     int nq, int64_t max_cap,
     uint64_t* __restrict__ rows_b, uint64_t* __restrict__ rows_d,
     uint64_t* __restrict__ rows_m, int64_t* __restrict__ found_out,
-    int64_t* __restrict__ total_out, unsigned long long* __restrict__ scanned_out) {
+    int64_t* __restrict__ total_out, unsigned long long* __restrict__ scanned_out,
+    unsigned long long* dbg) {  // KB_SCAN_DBG phase cycles (null in prod)
   int q = blockIdx.x;
   if (q >= nq) return;
   __shared__ int wave_cnt[SCAN_T_MAX / 64];
@@ -328,6 +335,8 @@ __global__ void k_range_scan2(
   const DevRangeQ& Q = qs[q];
   const QKey qstart{Q.start, Q.start_klen, Q.start_ko, qtails};
   const QKey qend{Q.end, Q.end_klen, Q.end_ko, qtails};
+  const bool dbg0 = dbg && threadIdx.x == 0;
+  unsigned long long tk0 = dbg0 ? wall_clock64() : 0;
   // prologue: waves 0-3 resolve the 4 bounds concurrently, each via the
   // wave-cooperative 64-ary search (blockDim >= 256 always)
   {
@@ -340,6 +349,7 @@ __global__ void k_range_scan2(
     if (w < 4 && (threadIdx.x & 63) == 0) bounds_s[w] = r;
   }
   __syncthreads();
+  if (dbg0) { unsigned long long tkb = wall_clock64(); atomicAdd(&dbg[0], tkb - tk0); }
   const int64_t lo_s = bounds_s[0], hi_s = bounds_s[1];
   const int64_t dlo_s = bounds_s[2], dhi_s = bounds_s[3];
   const int64_t cap = Q.cap > 0 ? Q.cap : INT64_MAX;
@@ -349,12 +359,13 @@ __global__ void k_range_scan2(
   uint64_t* outb = Q.count_only ? nullptr : rows_b + (int64_t)q * max_cap;
   int64_t nB = scan_run_winners(d, spill, dlo_s, dhi_s, Q.read_rev,
                                 cap, outd, max_cap, ROW_TAG_DELTA, nullptr,
-                                0, 0, &dtotal, &scanned, wave_cnt);
+                                0, 0, &dtotal, &scanned, wave_cnt, dbg);
   __syncthreads();  // wave_cnt handoff between the two runs
   int64_t nA = scan_run_winners(b, spill, lo_s, hi_s, Q.read_rev, cap,
                                 outb, max_cap, 0, dn ? &d : nullptr,
-                                dlo_s, dhi_s, &btotal, &scanned, wave_cnt);
+                                dlo_s, dhi_s, &btotal, &scanned, wave_cnt, dbg);
   __syncthreads();  // winner lists complete before the merge reads them
+  unsigned long long tkm = dbg0 ? wall_clock64() : 0;
   // merge by rank into rows_m (keys are disjoint across the two lists)
   int64_t cap_m = nA + nB;
   if (cap_m > cap) cap_m = cap;
@@ -376,6 +387,7 @@ __global__ void k_range_scan2(
       if (pos < cap_m) outm[pos] = rt;
     }
   }
+  if (dbg0) { unsigned long long tke = wall_clock64(); atomicAdd(&dbg[4], tke - tkm); atomicAdd(&dbg[5], tke - tk0); }
   if (threadIdx.x == 0) {
     total_out[q] = btotal + dtotal;
     found_out[q] = Q.count_only ? 0 : cap_m;
@@ -1051,6 +1063,11 @@ struct Slab::Impl {
   uint64_t* d_bitmap = nullptr;
   int64_t bitmap_cap = 0;
 
+  // KB_SCAN_DBG=1: per-phase wall_clock64 cycle sums from k_range_scan2
+  // ([0]=bounds [1]=pass1a [2]=pass1b [3]=scatter [4]=merge [5]=block total)
+  unsigned long long* d_dbg = nullptr;
+  double wall_khz = 0;
+
   hipEvent_t ev0 = nullptr, ev1 = nullptr, ev2 = nullptr, ev3 = nullptr;
   hipEvent_t ev_g0 = nullptr, ev_g1 = nullptr;  // async get lookup
   uint64_t* h_gmeta = nullptr;   // pinned [2*max_q]: orev | ometa
@@ -1316,6 +1333,13 @@ Slab* Slab::Create(int64_t max_rows, int64_t heap_cap, int device,
     HIP_CHECK_NULL(hipMalloc(&c->vo, max_rows * 8));
     HIP_CHECK_NULL(hipMalloc(&c->ko, max_rows * 8));
   }
+  if (env_i64("KB_SCAN_DBG", 0)) {
+    HIP_CHECK_NULL(hipMalloc(&I->d_dbg, 8 * 8));
+    HIP_CHECK_NULL(hipMemset(I->d_dbg, 0, 8 * 8));
+    int khz = 0;
+    (void)hipDeviceGetAttribute(&khz, hipDeviceAttributeWallClockRate, I->device);
+    I->wall_khz = khz > 0 ? (double)khz : 100000.0;
+  }
   I->spill_cap = env_i64("KB_SPILL_BYTES", 256ll << 20);
   HIP_CHECK_NULL(hipMalloc(&I->spillA, I->spill_cap));
   HIP_CHECK_NULL(hipMalloc(&I->spillB, I->spill_cap));
@@ -1558,7 +1582,7 @@ bool Slab::RangeBatchStart(const std::vector<DevRangeQ>& qs, std::string* err,
                      I->A.run(), I->n, I->DA.run(), I->dn, I->spillA,
                      I->d_qtails, I->d_qs, nq,
                      I->max_cap, I->d_rows, I->d_rows2, I->d_rowsm, I->d_found,
-                     I->d_total, I->d_scanned);
+                     I->d_total, I->d_scanned, I->d_dbg);
   HIP_CHECK(hipEventRecord(I->ev1, I->stream));
   hipLaunchKernelGGL(k_gather, dim3(nq), dim3(256), 0, I->stream, I->A.run(),
                      I->DA.run(), I->heapA, I->d_rowsm, I->max_cap,
@@ -1612,6 +1636,16 @@ bool Slab::RangeBatchFinish(int nq, bool d2h, bool parse,
   perf.scan_ms += ms;
   (void)hipEventElapsedTime(&ms, I->ev1, I->ev2);
   perf.gather_ms += ms;
+  if (I->d_dbg) {  // KB_SCAN_DBG: block-serial ms per phase, summed over blocks
+    unsigned long long c[8] = {0};
+    HIP_CHECK(hipMemcpy(c, I->d_dbg, 8 * 8, hipMemcpyDeviceToHost));
+    HIP_CHECK(hipMemset(I->d_dbg, 0, 8 * 8));
+    perf.dbg_a += (double)c[0] / I->wall_khz;
+    perf.dbg_b += (double)c[1] / I->wall_khz;
+    perf.dbg_c += (double)c[2] / I->wall_khz;
+    perf.dbg_d += (double)c[3] / I->wall_khz;
+    perf.dbg_e += (double)c[4] / I->wall_khz;
+  }
   perf.scan_launches++;
   perf.gather_launches++;
   perf.rows_scanned += (int64_t)scanned;
