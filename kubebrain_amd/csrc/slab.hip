@@ -192,6 +192,20 @@ constexpr uint64_t ROW_MASK = ROW_TAG_DELTA - 1;
 // scan block width: runtime-tunable (KB_SCAN_T in {256,512,1024}); the
 // kernel reads blockDim.x, only the LDS wave-count array is sized for the max
 constexpr int SCAN_T_MAX = 1024;
+// max delta-suppression span staged in LDS per query block (14 KB of LDS)
+constexpr int SUPP_MAX = 128;
+
+// compare LDS-staged suppression row j against query key q (sign as
+// rowcmp_q); exact 96B-prefix ties where either side exceeds KEYW fall back
+// to the global-memory compare (spill tails are never staged)
+__device__ __forceinline__ int supp_cmp(const uint64_t* sl_keys,
+                                        const uint32_t* sl_klen, int j,
+                                        const Run& srun, const uint8_t* spill,
+                                        int64_t slo, const QKey& q) {
+  int c = keycmp96((const uint8_t*)(sl_keys + (int64_t)j * (KEYW / 8)), q.k96);
+  if (c || (sl_klen[j] <= (uint32_t)KEYW && q.len <= (uint32_t)KEYW)) return c;
+  return rowcmp_q(srun, spill, slo + j, q);
+}
 
 // one run's winner scan (ordered append); returns written, *total = seen.
 // Two-pass per tile: (1) each wave reads a CONTIGUOUS chunk of rows with
@@ -205,6 +219,10 @@ __device__ int64_t scan_run_winners(
     int64_t cap, uint64_t* out, int64_t out_cap, uint64_t tagbit,
     const Run* srun, int64_t slo, int64_t shi,  // suppression run (null => none)
     int64_t* total_out, int64_t* scanned_accum, int* wave_cnt,
+    // LDS-staged copy of the suppression span (sl_n = -1 => not staged):
+    // probes binary-search LDS instead of paying ~9 HBM round trips each
+    const uint64_t* sl_keys, const uint64_t* sl_rev, const uint32_t* sl_klen,
+    int sl_n,
     unsigned long long* dbg) {  // KB_SCAN_DBG phase cycles (null in prod)
   const uint64_t* __restrict__ rev = run.rev;
   const uint64_t* __restrict__ meta = run.meta;
@@ -259,9 +277,24 @@ __device__ int64_t scan_run_winners(
         if (win) {
           int64_t i = wbase + ((int64_t)r << 6) + lane;
           QKey kk = row_qk(run, spill, i);
-          int64_t lb = d_lb_range(*srun, spill, slo, shi, kk, 1);
-          if (lb < shi && srun->rev[lb] <= R &&
-              rowcmp_q(*srun, spill, lb, kk) == 0) {
+          bool supp;
+          if (sl_n >= 0) {  // LDS-staged span: search without HBM latency
+            int lo2 = 0, hi2 = sl_n;
+            while (lo2 < hi2) {
+              int mid = (lo2 + hi2) >> 1;
+              if (supp_cmp(sl_keys, sl_klen, mid, *srun, spill, slo, kk) < 0)
+                lo2 = mid + 1;
+              else
+                hi2 = mid;
+            }
+            supp = lo2 < sl_n && sl_rev[lo2] <= R &&
+                   supp_cmp(sl_keys, sl_klen, lo2, *srun, spill, slo, kk) == 0;
+          } else {
+            int64_t lb = d_lb_range(*srun, spill, slo, shi, kk, 1);
+            supp = lb < shi && srun->rev[lb] <= R &&
+                   rowcmp_q(*srun, spill, lb, kk) == 0;
+          }
+          if (supp) {
             win = false;  // a newer (delta) row of this key wins instead
             flags &= ~(1ull << r);
           }
@@ -359,11 +392,31 @@ __global__ void k_range_scan2(
   uint64_t* outb = Q.count_only ? nullptr : rows_b + (int64_t)q * max_cap;
   int64_t nB = scan_run_winners(d, spill, dlo_s, dhi_s, Q.read_rev,
                                 cap, outd, max_cap, ROW_TAG_DELTA, nullptr,
-                                0, 0, &dtotal, &scanned, wave_cnt, dbg);
-  __syncthreads();  // wave_cnt handoff between the two runs
+                                0, 0, &dtotal, &scanned, wave_cnt,
+                                nullptr, nullptr, nullptr, -1, dbg);
+  // stage the delta-suppression span in LDS when it is small (the common
+  // case: the delta run folds long before any namespace accrues >SUPP_MAX
+  // rows), so base-scan probes binary-search LDS instead of HBM
+  __shared__ uint64_t sl_keys[SUPP_MAX * (KEYW / 8)];
+  __shared__ uint64_t sl_rev[SUPP_MAX];
+  __shared__ uint32_t sl_klen[SUPP_MAX];
+  int sl_n = -1;
+  if (dn && dhi_s - dlo_s <= (int64_t)SUPP_MAX) {
+    sl_n = (int)(dhi_s - dlo_s);
+    for (int64_t v = threadIdx.x; v < (int64_t)sl_n * (KEYW / 8);
+         v += blockDim.x)
+      sl_keys[v] =
+          ((const uint64_t*)(d.keys + dlo_s * KEYW))[v];
+    for (int64_t j = threadIdx.x; j < sl_n; j += blockDim.x) {
+      sl_rev[j] = d.rev[dlo_s + j];
+      sl_klen[j] = meta_klen(d.meta[dlo_s + j]);
+    }
+  }
+  __syncthreads();  // wave_cnt handoff between the two runs + staged span
   int64_t nA = scan_run_winners(b, spill, lo_s, hi_s, Q.read_rev, cap,
                                 outb, max_cap, 0, dn ? &d : nullptr,
-                                dlo_s, dhi_s, &btotal, &scanned, wave_cnt, dbg);
+                                dlo_s, dhi_s, &btotal, &scanned, wave_cnt,
+                                sl_keys, sl_rev, sl_klen, sl_n, dbg);
   __syncthreads();  // winner lists complete before the merge reads them
   unsigned long long tkm = dbg0 ? wall_clock64() : 0;
   // merge by rank into rows_m (keys are disjoint across the two lists)
@@ -1025,7 +1078,7 @@ struct Slab::Impl {
 
   // range/get scratch
   int max_q = 1024;
-  int scan_t = 256;             // KB_SCAN_T: threads per scan block
+  int scan_t = 1024;            // KB_SCAN_T: threads per scan block
   int64_t max_cap = 4352;       // winners per query cap (>= limit+1 for etcd's 500)
   int64_t arena_bytes = 384ll << 20;
   DevRangeQ* d_qs = nullptr;
@@ -1307,7 +1360,7 @@ Slab* Slab::Create(int64_t max_rows, int64_t heap_cap, int device,
   I->max_rows = max_rows;
   I->heap_cap = heap_cap;
   I->max_q = (int)env_i64("KB_MAX_Q", 1024);
-  I->scan_t = (int)env_i64("KB_SCAN_T", 256);
+  I->scan_t = (int)env_i64("KB_SCAN_T", 1024);
   // the scan prologue resolves 4 bounds with waves 0-3: >= 256 threads
   if (I->scan_t < 256) I->scan_t = 256;
   if (I->scan_t > 1024) I->scan_t = 1024;
