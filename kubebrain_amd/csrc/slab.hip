@@ -280,9 +280,10 @@ __device__ int64_t scan_run_winners(
           bool supp;
           if (sl_n >= 0) {  // LDS-staged span: search without HBM latency
             int lo2 = 0, hi2 = sl_n;
-            while (lo2 < hi2) {
+            while (lo2 < hi2) {  // lower bound of (kk, rev>=1), as d_lb_range
               int mid = (lo2 + hi2) >> 1;
-              if (supp_cmp(sl_keys, sl_klen, mid, *srun, spill, slo, kk) < 0)
+              int c = supp_cmp(sl_keys, sl_klen, mid, *srun, spill, slo, kk);
+              if (c < 0 || (c == 0 && sl_rev[mid] < 1))
                 lo2 = mid + 1;
               else
                 hi2 = mid;
