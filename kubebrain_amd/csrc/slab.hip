@@ -192,8 +192,10 @@ constexpr uint64_t ROW_MASK = ROW_TAG_DELTA - 1;
 // scan block width: runtime-tunable (KB_SCAN_T in {256,512,1024}); the
 // kernel reads blockDim.x, only the LDS wave-count array is sized for the max
 constexpr int SCAN_T_MAX = 1024;
-// max delta-suppression span staged in LDS per query block (14 KB of LDS)
-constexpr int SUPP_MAX = 128;
+// max delta-suppression span staged in LDS per query block (~55 KB of LDS:
+// still 2 blocks/CU of 160 KB; the stage is a coalesced span-sized read that
+// replaces ~cap random binary searches over HBM)
+constexpr int SUPP_MAX = 512;
 
 // compare LDS-staged suppression row j against query key q (sign as
 // rowcmp_q); exact 96B-prefix ties where either side exceeds KEYW fall back
@@ -579,19 +581,21 @@ __global__ void k_gather_copy(Run rb, Run rd,
                               const DevRangeQ* __restrict__ qs, int nq,
                               uint8_t* __restrict__ gbuf, int64_t qcap,
                               const int64_t* __restrict__ offs,
-                              const int32_t* __restrict__ overflow) {
+                              const int32_t* __restrict__ overflow,
+                              int gwl) {  // log2(record-group lanes), KB_GATHER_GW
   int q = blockIdx.x;
   if (q >= nq || overflow[q]) return;
   int64_t nwin = found_out[q];
   const bool konly = qs[q].keys_only != 0;
   const uint64_t* rows = rows_out + (int64_t)q * max_cap;
   const int64_t* qoffs = offs + (int64_t)q * max_cap;
-  // 16-lane record groups — independent load chains across records
+  // record groups of 2^gwl lanes — independent load chains across records
+  const int gw = 1 << gwl, gpw = 64 >> gwl;  // lanes per group, groups per wave
   int lane = threadIdx.x & 63, w = threadIdx.x >> 6;
-  int grp = lane >> 4, gl = lane & 15;
+  int grp = lane >> gwl, gl = lane & (gw - 1);
   uint8_t* qb = gbuf + (int64_t)q * qcap;
-  int64_t stride = (blockDim.x / 64) * 4;
-  for (int64_t j = (int64_t)w * 4 + grp; j < nwin; j += stride) {
+  int64_t stride = (blockDim.x / 64) * gpw;
+  for (int64_t j = (int64_t)w * gpw + grp; j < nwin; j += stride) {
     uint64_t rt = rows[j];
     bool isd = (rt & ROW_TAG_DELTA) != 0;
     int64_t row = (int64_t)(rt & ROW_MASK);
@@ -609,18 +613,18 @@ __global__ void k_gather_copy(Run rb, Run rd,
     uint32_t kin = klen > (uint32_t)KEYW ? (uint32_t)KEYW : klen;
     // column rows are 96B-strided => 8-aligned; copy whole u64s (may round
     // up to 7B into the record's 16B key padding — always in-bounds)
-    for (uint32_t b = gl; b < (kin + 7) / 8; b += 16)
+    for (uint32_t b = gl; b < (kin + 7) / 8; b += gw)
       ((uint64_t*)kd)[b] = ((const uint64_t*)ks)[b];
     if (klen > (uint32_t)KEYW) {  // spill tail (keys > 96B)
       const uint8_t* ts = spill + r.ko[row];
-      for (uint32_t t = gl; t < klen - (uint32_t)KEYW; t += 16)
+      for (uint32_t t = gl; t < klen - (uint32_t)KEYW; t += gw)
         kd[KEYW + t] = ts[t];
     }
     if (konly) continue;
     const uint8_t* vs = heap + r.vo[row];
     uint8_t* vd = dst + 16 + ((klen + 15) & ~15u);
     uint32_t w16 = vlen >> 4;
-    for (uint32_t b = gl; b < w16; b += 16)
+    for (uint32_t b = gl; b < w16; b += gw)
       ((uint4*)vd)[b] = ((const uint4*)vs)[b];
     if (gl == 0)
       for (uint32_t b = w16 * 16; b < vlen; ++b) vd[b] = vs[b];
@@ -1080,6 +1084,7 @@ struct Slab::Impl {
   // range/get scratch
   int max_q = 1024;
   int scan_t = 1024;            // KB_SCAN_T: threads per scan block
+  int gather_gwl = 4;           // KB_GATHER_GW: log2 lanes per record group
   int64_t max_cap = 4352;       // winners per query cap (>= limit+1 for etcd's 500)
   int64_t arena_bytes = 384ll << 20;
   DevRangeQ* d_qs = nullptr;
@@ -1362,6 +1367,12 @@ Slab* Slab::Create(int64_t max_rows, int64_t heap_cap, int device,
   I->heap_cap = heap_cap;
   I->max_q = (int)env_i64("KB_MAX_Q", 1024);
   I->scan_t = (int)env_i64("KB_SCAN_T", 1024);
+  {
+    int64_t gwv = env_i64("KB_GATHER_GW", 16);
+    int l = 0;
+    while ((1 << l) < gwv && l < 6) ++l;
+    I->gather_gwl = l;
+  }
   // the scan prologue resolves 4 bounds with waves 0-3: >= 256 threads
   if (I->scan_t < 256) I->scan_t = 256;
   if (I->scan_t > 1024) I->scan_t = 1024;
@@ -1645,7 +1656,7 @@ bool Slab::RangeBatchStart(const std::vector<DevRangeQ>& qs, std::string* err,
   hipLaunchKernelGGL(k_gather_copy, dim3(nq), dim3(512), 0, I->stream,
                      I->A.run(), I->DA.run(), I->spillA, I->heapA, I->d_rowsm,
                      I->max_cap, I->d_found, I->d_qs, nq, I->d_gbuf, qcap,
-                     I->d_offs, I->d_ovf);
+                     I->d_offs, I->d_ovf, I->gather_gwl);
   HIP_CHECK(hipEventRecord(I->ev2, I->stream));
   return true;
 }
