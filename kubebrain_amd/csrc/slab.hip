@@ -356,6 +356,29 @@ __device__ int64_t d_lb_winlist(const uint64_t* rows, int64_t n, const Run& b,
   return lo;
 }
 
+// bounds pre-pass: one 256-thread block per query, its 4 waves resolving the
+// 4 run bounds concurrently via the wave-cooperative 64-ary search. As a
+// standalone launch all 4*nq searches overlap chip-wide; inside the scan
+// kernel they sat on every block's critical path (~7 us serial per block).
+__global__ void k_range_bounds(Run b, int64_t n, Run d, int64_t dn,
+                               const uint8_t* __restrict__ spill,
+                               const uint8_t* __restrict__ qtails,
+                               const DevRangeQ* __restrict__ qs, int nq,
+                               int64_t* __restrict__ bounds_g /*[4*nq]*/) {
+  int q = blockIdx.x;
+  if (q >= nq) return;
+  const DevRangeQ& Q = qs[q];
+  const QKey qstart{Q.start, Q.start_klen, Q.start_ko, qtails};
+  const QKey qend{Q.end, Q.end_klen, Q.end_ko, qtails};
+  int w = threadIdx.x >> 6;
+  int64_t r = -1;
+  if (w == 0) r = d_lb_wave(b, spill, n, qstart, Q.start_rev);
+  else if (w == 1) r = d_lb_wave(b, spill, n, qend, 0);
+  else if (w == 2) r = d_lb_wave(d, spill, dn, qstart, Q.start_rev);
+  else if (w == 3) r = d_lb_wave(d, spill, dn, qend, 0);
+  if (w < 4 && (threadIdx.x & 63) == 0) bounds_g[(int64_t)q * 4 + w] = r;
+}
+
 __global__ void k_range_scan2(
     Run b, int64_t n, Run d, int64_t dn, const uint8_t* __restrict__ spill,
     const uint8_t* __restrict__ qtails, const DevRangeQ* __restrict__ qs,
@@ -363,31 +386,22 @@ __global__ void k_range_scan2(
     uint64_t* __restrict__ rows_b, uint64_t* __restrict__ rows_d,
     uint64_t* __restrict__ rows_m, int64_t* __restrict__ found_out,
     int64_t* __restrict__ total_out, unsigned long long* __restrict__ scanned_out,
+    const int64_t* __restrict__ bounds_g,  // from k_range_bounds
     unsigned long long* dbg) {  // KB_SCAN_DBG phase cycles (null in prod)
   int q = blockIdx.x;
   if (q >= nq) return;
   __shared__ int wave_cnt[SCAN_T_MAX / 64];
-  __shared__ int64_t bounds_s[4];
   const DevRangeQ& Q = qs[q];
   const QKey qstart{Q.start, Q.start_klen, Q.start_ko, qtails};
   const QKey qend{Q.end, Q.end_klen, Q.end_ko, qtails};
   const bool dbg0 = dbg && threadIdx.x == 0;
   unsigned long long tk0 = dbg0 ? wall_clock64() : 0;
-  // prologue: waves 0-3 resolve the 4 bounds concurrently, each via the
-  // wave-cooperative 64-ary search (blockDim >= 256 always)
-  {
-    int w = threadIdx.x >> 6;
-    int64_t r = -1;
-    if (w == 0) r = d_lb_wave(b, spill, n, qstart, Q.start_rev);
-    else if (w == 1) r = d_lb_wave(b, spill, n, qend, 0);
-    else if (w == 2) r = d_lb_wave(d, spill, dn, qstart, Q.start_rev);
-    else if (w == 3) r = d_lb_wave(d, spill, dn, qend, 0);
-    if (w < 4 && (threadIdx.x & 63) == 0) bounds_s[w] = r;
-  }
-  __syncthreads();
+  // bounds resolved by the k_range_bounds pre-pass
+  const int64_t lo_s = bounds_g[(int64_t)q * 4 + 0];
+  const int64_t hi_s = bounds_g[(int64_t)q * 4 + 1];
+  const int64_t dlo_s = bounds_g[(int64_t)q * 4 + 2];
+  const int64_t dhi_s = bounds_g[(int64_t)q * 4 + 3];
   if (dbg0) { unsigned long long tkb = wall_clock64(); atomicAdd(&dbg[0], tkb - tk0); }
-  const int64_t lo_s = bounds_s[0], hi_s = bounds_s[1];
-  const int64_t dlo_s = bounds_s[2], dhi_s = bounds_s[3];
   const int64_t cap = Q.cap > 0 ? Q.cap : INT64_MAX;
   int64_t scanned = 0;
   int64_t dtotal = 0, btotal = 0;
@@ -414,6 +428,10 @@ __global__ void k_range_scan2(
       sl_rev[j] = d.rev[dlo_s + j];
       sl_klen[j] = meta_klen(d.meta[dlo_s + j]);
     }
+  }
+  if (dbg0) {
+    atomicAdd(&dbg[6], (unsigned long long)(sl_n >= 0 ? 1 : 0));
+    atomicAdd(&dbg[7], (unsigned long long)(dhi_s - dlo_s));
   }
   __syncthreads();  // wave_cnt handoff between the two runs + staged span
   int64_t nA = scan_run_winners(b, spill, lo_s, hi_s, Q.read_rev, cap,
@@ -1092,6 +1110,7 @@ struct Slab::Impl {
   uint64_t* d_rows = nullptr;   // max_q*max_cap (base winners)
   uint64_t* d_rows2 = nullptr;  // delta winners
   uint64_t* d_rowsm = nullptr;  // merged winners
+  int64_t* d_bounds4 = nullptr;  // [4*max_q] k_range_bounds results
   int64_t* d_offs = nullptr;    // max_q*max_cap
   int64_t *d_found = nullptr, *d_total = nullptr, *d_gbytes = nullptr;
   int32_t *d_ovf = nullptr, *d_found32 = nullptr;
@@ -1437,6 +1456,7 @@ Slab* Slab::Create(int64_t max_rows, int64_t heap_cap, int device,
   HIP_CHECK_NULL(hipMalloc(&I->d_rows, (int64_t)I->max_q * I->max_cap * 8));
   HIP_CHECK_NULL(hipMalloc(&I->d_rows2, (int64_t)I->max_q * I->max_cap * 8));
   HIP_CHECK_NULL(hipMalloc(&I->d_rowsm, (int64_t)I->max_q * I->max_cap * 8));
+  HIP_CHECK_NULL(hipMalloc(&I->d_bounds4, (int64_t)I->max_q * 4 * 8));
   HIP_CHECK_NULL(hipMalloc(&I->d_offs, (int64_t)I->max_q * I->max_cap * 8));
   HIP_CHECK_NULL(hipMalloc(&I->d_found, I->max_q * 8));
   HIP_CHECK_NULL(hipMalloc(&I->d_total, I->max_q * 8));
@@ -1643,11 +1663,14 @@ bool Slab::RangeBatchStart(const std::vector<DevRangeQ>& qs, std::string* err,
   HIP_CHECK(hipMemsetAsync(I->d_scanned, 0, 8, I->stream));
   HIP_CHECK(hipMemsetAsync(I->d_bytes, 0, 8, I->stream));
   HIP_CHECK(hipEventRecord(I->ev0, I->stream));
+  hipLaunchKernelGGL(k_range_bounds, dim3(nq), dim3(256), 0, I->stream,
+                     I->A.run(), I->n, I->DA.run(), I->dn, I->spillA,
+                     I->d_qtails, I->d_qs, nq, I->d_bounds4);
   hipLaunchKernelGGL(k_range_scan2, dim3(nq), dim3((uint32_t)I->scan_t), 0, I->stream,
                      I->A.run(), I->n, I->DA.run(), I->dn, I->spillA,
                      I->d_qtails, I->d_qs, nq,
                      I->max_cap, I->d_rows, I->d_rows2, I->d_rowsm, I->d_found,
-                     I->d_total, I->d_scanned, I->d_dbg);
+                     I->d_total, I->d_scanned, I->d_bounds4, I->d_dbg);
   HIP_CHECK(hipEventRecord(I->ev1, I->stream));
   hipLaunchKernelGGL(k_gather, dim3(nq), dim3(256), 0, I->stream, I->A.run(),
                      I->DA.run(), I->heapA, I->d_rowsm, I->max_cap,
@@ -1710,6 +1733,8 @@ bool Slab::RangeBatchFinish(int nq, bool d2h, bool parse,
     perf.dbg_c += (double)c[2] / I->wall_khz;
     perf.dbg_d += (double)c[3] / I->wall_khz;
     perf.dbg_e += (double)c[4] / I->wall_khz;
+    fprintf(stderr, "[scan_dbg] staged=%llu/%d span_sum=%llu\n", c[6], nq,
+            c[7]);
   }
   perf.scan_launches++;
   perf.gather_launches++;
