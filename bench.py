@@ -595,16 +595,18 @@ def main():
     t0 = time.time()
     wsteps = 0
     while wsteps < args.warmup or time.time() - t0 < 2.0:
-        one_step()
+        one_step(mode=4)  # pipelined: previous batch collected next step
         wsteps += 1
         if wsteps > 10000:
             break
+    drain()
     barrier()
     store._f("perf_reset")(ctypes.c_void_p(store.h))
     t_split["range_s"] = t_split["txn_s"] = 0.0
     t0 = time.time()
     for _ in range(args.steps):
-        one_step()
+        one_step(mode=4)
+    drain()  # finish the in-flight final batch inside the timed bracket
     barrier()
     elapsed = time.time() - t0
     split_snapshot = dict(t_split)  # before the untimed d2h/watch phases

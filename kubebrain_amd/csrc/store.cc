@@ -1173,8 +1173,26 @@ bool Store::BenchRange(const uint8_t* qbuf, size_t nq, int mode,
   return true;
 }
 
+bool Store::finishPendingBench(unsigned long long* total, std::string* err) {
+  unsigned long long tot = 0;
+  if (bench_pending_nq_ >= 0) {
+    std::vector<kbslab::RangeResult> outs;
+    int nq = bench_pending_nq_;
+    bench_pending_nq_ = -1;
+    if (!slab_->RangeBatchFinish(nq, false, false, &outs, err)) return false;
+    for (int j = 0; j < nq; ++j) {
+      int64_t lim = bench_pending_limits_[j];
+      int64_t w = outs[j].written;
+      tot += (unsigned long long)(lim > 0 && w > lim ? lim : w);
+    }
+  }
+  if (total) *total = tot;
+  return true;
+}
+
 bool Store::Sync(std::string* err) {
   std::lock_guard<std::recursive_mutex> lk(mu_);
+  if (!finishPendingBench(nullptr, err)) return false;
   return slab_->DrainD2H(err);
 }
 
@@ -1183,12 +1201,21 @@ bool Store::BenchStep(const uint8_t* qbuf, size_t nq, const uint8_t* tbuf,
                       unsigned long long* total, double* secs,
                       std::string* err) {
   std::lock_guard<std::recursive_mutex> lk(mu_);
+  const bool pipe = (mode & 4) != 0;
+  const bool d2h = (mode & 1) != 0;
+  if (pipe && d2h) { if (err) *err = "pipelined bench step excludes d2h"; return false; }
   if (!syncReads(err)) return false;
   std::vector<DevRangeQ> qall;
   std::vector<int64_t> limits;
   std::string qtails;
   parseBenchQueries(qbuf, nq, committed_, mode, &qall, &limits, &qtails);
-  const bool d2h = (mode & 1) != 0;
+  // pipelined (mode bit2): the previous step's range kernels ran while the
+  // host applied its txns and parsed this step; collect them now, return
+  // THEIR totals, and leave this step's batch in flight (range reads are
+  // snapshot-exact at their read_rev, so deferring collection never changes
+  // results)
+  unsigned long long prev_tot = 0;
+  if (pipe && !finishPendingBench(&prev_tot, err)) return false;
   // txn ops parsed up front so the batched CAS lookup (f1) can launch BEFORE
   // the range batch: stream order runs the small lookup first, the host
   // applies the conditional-update protocol against the device results while
@@ -1247,6 +1274,14 @@ bool Store::BenchStep(const uint8_t* qbuf, size_t nq, const uint8_t* tbuf,
         out_revs[i] = r.succeeded ? r.header_revision : 0;
       }
     }
+  }
+  if (pipe) {  // leave this step's batch in flight; report the previous one
+    bench_pending_nq_ = (int)nq;
+    bench_pending_limits_ = std::move(limits);
+    *secs = std::chrono::duration<double>(std::chrono::steady_clock::now() - t0).count();
+    *total = prev_tot;
+    ops_range_ += (int64_t)nq;
+    return true;
   }
   std::vector<kbslab::RangeResult> outs;
   if (!slab_->RangeBatchFinish((int)nq, d2h, false, &outs, err)) return false;

@@ -226,6 +226,13 @@ class Store {
 
   // host-side perf
   int64_t ops_create_ = 0, ops_update_ = 0, ops_delete_ = 0, ops_range_ = 0;
+  // one-step-deep bench pipeline (BenchStep mode bit2): the previous step's
+  // range batch is finished at the NEXT step (or Sync), so its kernels
+  // overlap the host's txn apply + next-step prep. Valid only between
+  // consecutive BenchStep calls; Sync() drains it.
+  int bench_pending_nq_ = -1;
+  std::vector<int64_t> bench_pending_limits_;
+  bool finishPendingBench(unsigned long long* total, std::string* err);
   int64_t delivered_ = 0;  // watch events enqueued to watchers (fan-out)
   bool keep_event_log_ = true;
   double sync_s_ = 0;
