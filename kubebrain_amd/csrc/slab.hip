@@ -837,6 +837,95 @@ __global__ void k_merge_rank(Run a, int64_t n, Run dnew, int64_t m,
   drop[i] = eq ? 1 : 0;
 }
 
+// inverted small-insert merge (m <= 1024 new rows into an n-row run): ONE
+// block searches each NEW row into the big run — m*log2(n) work instead of
+// k_merge_rank's n*log2(m) — writes the new rows at their final slots, and
+// leaves a difference array v[0..n+1] whose prefix sum is each base row's
+// slot shift (insertions +1 at ub_j, drops -1 after lb_j). drop[] marks
+// replaced base rows ((key,rev)-equal memtable revision-rows). new_n -> s_d.
+__global__ void k_merge_insert_inv(
+    Run a, int64_t n, Run dnew, int64_t m, const uint8_t* __restrict__ spill,
+    uint64_t* __restrict__ v /*[n+2] zeroed*/,
+    uint64_t* __restrict__ drop /*[n] zeroed*/,
+    uint64_t* __restrict__ outpos /*[m] delta dest slots (for the fixup)*/,
+    uint8_t* keysB, uint64_t* metaB, uint64_t* revB, uint64_t* voB,
+    uint64_t* koB, int64_t* __restrict__ newn_out) {
+  __shared__ uint16_t ldup[1024];
+  int64_t j = threadIdx.x;
+  bool has = j < m;
+  int64_t lb = n;
+  bool dup = false;
+  if (has) {
+    QKey k = row_qk(dnew, spill, j);
+    lb = d_lb_range(a, spill, 0, n, k, dnew.rev[j]);
+    dup = lb < n && a.rev[lb] == dnew.rev[j] &&
+          rowcmp_q(a, spill, lb, k) == 0;
+  }
+  ldup[threadIdx.x] = dup ? 1 : 0;
+  __syncthreads();
+  // exclusive prefix of dup flags over the block (m <= 1024)
+  int64_t dupx = 0, total = 0;
+  for (int t = 0; t < (int)m; ++t) {
+    if (t < (int)j) dupx += ldup[t];
+    total += ldup[t];
+  }
+  if (threadIdx.x == 0 && newn_out) *newn_out = n + m - total;
+  if (!has) return;
+  if (dup) {
+    drop[lb] = 1;
+    atomicAdd(&v[lb + 1], (uint64_t)-1ll);
+  }
+  atomicAdd(&v[lb + (dup ? 1 : 0)], 1ull);  // ub_j = lb + dup
+  int64_t out = lb - dupx + j;
+  outpos[j] = (uint64_t)out;
+  revB[out] = dnew.rev[j];
+  metaB[out] = dnew.meta[j];
+  voB[out] = dnew.vo[j];
+  koB[out] = dnew.ko[j];
+  const uint4* ks = (const uint4*)(dnew.keys + j * KEYW);
+  uint4* kd = (uint4*)(keysB + out * KEYW);
+#pragma unroll
+  for (int t = 0; t < KEYW / 16; ++t) kd[t] = ks[t];
+}
+
+// base-row scatter for the inverted path: slot = i + excl_scan(v)[i+1]
+__global__ void k_merge_scatter_base2(
+    Run a, const uint64_t* __restrict__ sv /*excl scan of v, [n+2]*/,
+    const uint64_t* __restrict__ drop, uint8_t* keysB, uint64_t* metaB,
+    uint64_t* revB, uint64_t* voB, uint64_t* koB, int64_t n) {
+  int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  if (i >= n || drop[i]) return;
+  int64_t j = i + (int64_t)sv[i + 1];
+  revB[j] = a.rev[i];
+  metaB[j] = a.meta[i];
+  voB[j] = a.vo[i];
+  koB[j] = a.ko[i];
+  const uint4* ks = (const uint4*)(a.keys + i * KEYW);
+  uint4* kd = (uint4*)(keysB + j * KEYW);
+#pragma unroll
+  for (int t = 0; t < KEYW / 16; ++t) kd[t] = ks[t];
+}
+
+// same_next fixup after an inverted insert: only rows adjacent to an
+// inserted slot can change (relative order of surviving base rows is
+// preserved and their meta bits travel with them); recompute at out_j-1 and
+// out_j for every inserted row
+__global__ void k_same_next_fixup(uint8_t* __restrict__ keys,
+                                  uint64_t* __restrict__ meta,
+                                  const uint64_t* __restrict__ ko,
+                                  const uint8_t* __restrict__ spill,
+                                  int64_t new_n,
+                                  const uint64_t* __restrict__ outpos,
+                                  int64_t m) {
+  int64_t t = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  if (t >= 2 * m) return;
+  int64_t i = (int64_t)outpos[t >> 1] - (t & 1);
+  if (i < 0 || i >= new_n) return;
+  uint64_t mm = meta[i];
+  bool same = i + 1 < new_n && rows_same_key(keys, meta, ko, spill, i, i + 1);
+  meta[i] = same ? (mm | M_SAME_NEXT) : (mm & ~M_SAME_NEXT);
+}
+
 __global__ void k_merge_scatter_base(
     Run a, const uint64_t* __restrict__ rank,
     const uint64_t* __restrict__ drop, const uint64_t* __restrict__ dropx,
@@ -1094,6 +1183,7 @@ struct Slab::Impl {
   }
 
   // scan scratch (u64, shared across ops)
+  uint64_t* d_outpos = nullptr;  // [1024] inverted-insert dest slots
   uint64_t *s_a = nullptr, *s_b = nullptr, *s_c = nullptr, *s_d = nullptr,
            *s_e = nullptr;  // max_rows+2
   uint64_t *lv1 = nullptr, *lv1o = nullptr, *lv2 = nullptr, *lv2o = nullptr,
@@ -1116,6 +1206,8 @@ struct Slab::Impl {
   int32_t *d_ovf = nullptr, *d_found32 = nullptr;
   uint64_t *d_orev = nullptr, *d_ometa = nullptr;
   unsigned long long *d_scanned = nullptr, *d_bytes = nullptr;
+  uint8_t* h_resmeta = nullptr;  // pinned mirror of the resmeta block
+  int64_t resmeta_bytes = 0;
   uint8_t* d_gbuf = nullptr;    // arena
   uint8_t* d_pack = nullptr;    // arena
   int64_t* d_goffs = nullptr;   // max_q+1
@@ -1157,6 +1249,7 @@ struct Slab::Impl {
   bool ensure_hpack(int64_t need, std::string* err) {
     if (need <= h_pack_cap) return true;
     if (h_pack) (void)hipHostFree(h_pack);
+    if (h_resmeta) (void)hipHostFree(h_resmeta);
     int64_t cap = need + need / 2;
     HIP_CHECK(hipHostMalloc(&h_pack, cap));
     h_pack_cap = cap;
@@ -1222,10 +1315,10 @@ struct Slab::Impl {
                     (void*)heapA, (void*)heapB, (void*)s_a, (void*)s_b,
                     (void*)s_c, (void*)s_d, (void*)s_e, (void*)lv1, (void*)lv1o, (void*)lv2,
                     (void*)lv2o, (void*)lv3, (void*)lv3o, (void*)d_qs,
-                    (void*)d_gq, (void*)d_rows, (void*)d_offs, (void*)d_found,
-                    (void*)d_total, (void*)d_gbytes, (void*)d_ovf,
+                    (void*)d_gq, (void*)d_rows, (void*)d_offs,
+                    (void*)d_found,  // base of the resmeta block
                     (void*)d_found32, (void*)d_orev, (void*)d_ometa,
-                    (void*)d_scanned, (void*)d_bytes, (void*)d_gbuf,
+                    (void*)d_outpos, (void*)d_gbuf,
                     (void*)d_pack, (void*)d_goffs, (void*)d_bounds,
                     (void*)d_bkeys, (void*)d_brevs, (void*)d_dkeys,
                     (void*)d_dmeta, (void*)d_drev, (void*)d_dvo, (void*)d_dko,
@@ -1237,6 +1330,7 @@ struct Slab::Impl {
       if (p) (void)hipFree(p);
     }
     if (h_pack) (void)hipHostFree(h_pack);
+    if (h_resmeta) (void)hipHostFree(h_resmeta);
     if (h_gmeta) (void)hipHostFree(h_gmeta);
     if (h_gfound) (void)hipHostFree(h_gfound);
     if (ev_g0) (void)hipEventDestroy(ev_g0);
@@ -1318,6 +1412,34 @@ struct Slab::Impl {
       HIP_CHECK(hipMemcpyAsync(&nn, s_d, 8, hipMemcpyDeviceToHost, stream));
       HIP_CHECK(hipStreamSynchronize(stream));
       *out_n = nn;
+      return true;
+    }
+    if (m > 0 && m <= 1024 && n > 0) {
+      // inverted small-insert path (the per-step txn merge): m searches into
+      // the n-row run, not n searches into the m new rows
+      if (!d_outpos) HIP_CHECK(hipMalloc(&d_outpos, 1024 * 8));
+      HIP_CHECK(hipMemsetAsync(s_a, 0, (n + 2) * 8, stream));
+      HIP_CHECK(hipMemsetAsync(s_b, 0, n * 8, stream));
+      hipLaunchKernelGGL(k_merge_insert_inv, dim3(1), dim3(1024), 0, stream,
+                         src.run(), n, dnew, m, spillA, s_a, s_b, d_outpos,
+                         dst.keys, dst.meta, dst.rev, dst.vo, dst.ko,
+                         (int64_t*)s_d);
+      if (!scan(s_a, s_c, n + 2, nullptr, err)) return false;  // no host sync
+      hipLaunchKernelGGL(k_merge_scatter_base2, dim3((uint32_t)ceil_div(n, 256)),
+                         dim3(256), 0, stream, src.run(), s_c, s_b, dst.keys,
+                         dst.meta, dst.rev, dst.vo, dst.ko, n);
+      int64_t new_n;
+      if (device_newn_ok) {
+        new_n = *out_n;
+      } else {
+        new_n = 0;
+        HIP_CHECK(hipMemcpyAsync(&new_n, s_d, 8, hipMemcpyDeviceToHost, stream));
+        HIP_CHECK(hipStreamSynchronize(stream));
+      }
+      hipLaunchKernelGGL(k_same_next_fixup, dim3((uint32_t)ceil_div(2 * m, 256)),
+                         dim3(256), 0, stream, dst.keys, dst.meta, dst.ko,
+                         spillA, new_n, d_outpos, m);
+      *out_n = new_n;
       return true;
     }
     if (n > 0) {
@@ -1458,15 +1580,24 @@ Slab* Slab::Create(int64_t max_rows, int64_t heap_cap, int device,
   HIP_CHECK_NULL(hipMalloc(&I->d_rowsm, (int64_t)I->max_q * I->max_cap * 8));
   HIP_CHECK_NULL(hipMalloc(&I->d_bounds4, (int64_t)I->max_q * 4 * 8));
   HIP_CHECK_NULL(hipMalloc(&I->d_offs, (int64_t)I->max_q * I->max_cap * 8));
-  HIP_CHECK_NULL(hipMalloc(&I->d_found, I->max_q * 8));
-  HIP_CHECK_NULL(hipMalloc(&I->d_total, I->max_q * 8));
-  HIP_CHECK_NULL(hipMalloc(&I->d_gbytes, I->max_q * 8));
-  HIP_CHECK_NULL(hipMalloc(&I->d_ovf, I->max_q * 4));
+  {
+    // found/total/gbytes/ovf/scanned/bytes live in ONE allocation so
+    // RangeBatchFinish reads them all back in ONE D2H (6 small copies cost
+    // ~20 us/step of copy-queue time otherwise)
+    uint8_t* base = nullptr;
+    I->resmeta_bytes = I->max_q * 8 * 3 + I->max_q * 4 + 16;
+    HIP_CHECK_NULL(hipMalloc(&base, I->resmeta_bytes));
+    I->d_found = (int64_t*)base;
+    I->d_total = (int64_t*)(base + I->max_q * 8);
+    I->d_gbytes = (int64_t*)(base + I->max_q * 16);
+    I->d_ovf = (int32_t*)(base + I->max_q * 24);
+    I->d_scanned = (unsigned long long*)(base + I->max_q * 24 + I->max_q * 4);
+    I->d_bytes = I->d_scanned + 1;
+    HIP_CHECK_NULL(hipHostMalloc(&I->h_resmeta, I->resmeta_bytes));
+  }
   HIP_CHECK_NULL(hipMalloc(&I->d_found32, I->max_q * 4));
   HIP_CHECK_NULL(hipMalloc(&I->d_orev, I->max_q * 8));
   HIP_CHECK_NULL(hipMalloc(&I->d_ometa, I->max_q * 8));
-  HIP_CHECK_NULL(hipMalloc(&I->d_scanned, 8));
-  HIP_CHECK_NULL(hipMalloc(&I->d_bytes, 8));
   HIP_CHECK_NULL(hipMalloc(&I->d_gbuf, I->arena_bytes));
   HIP_CHECK_NULL(hipMalloc(&I->d_pack, I->arena_bytes));
   HIP_CHECK_NULL(hipMalloc(&I->d_goffs, (I->max_q + 1) * 8));
@@ -1701,23 +1832,21 @@ bool Slab::RangeBatchFinish(int nq, bool d2h, bool parse,
   // small result metadata: d_found/d_total/d_gbytes are CONTIGUOUS slices of
   // one allocation (see Create), so one D2H covers them; ovf+counters ride
   // two more copies
+  // found/total/gbytes/ovf/scanned/bytes share one device block: one D2H
+  // into the pinned mirror instead of six small copies on the store stream
   std::vector<int64_t> found(nq), total(nq), gbytes(nq);
   std::vector<int32_t> ovf(nq);
   unsigned long long scanned = 0, bytes = 0;
   {
-    std::vector<int64_t> meta3(3 * nq);
-    unsigned long long cnt2[2];
-    HIP_CHECK(hipMemcpyAsync(meta3.data(), I->d_found, nq * 8, hipMemcpyDeviceToHost, I->stream));
-    HIP_CHECK(hipMemcpyAsync(meta3.data() + nq, I->d_total, nq * 8, hipMemcpyDeviceToHost, I->stream));
-    HIP_CHECK(hipMemcpyAsync(meta3.data() + 2 * nq, I->d_gbytes, nq * 8, hipMemcpyDeviceToHost, I->stream));
-    HIP_CHECK(hipMemcpyAsync(ovf.data(), I->d_ovf, nq * 4, hipMemcpyDeviceToHost, I->stream));
-    HIP_CHECK(hipMemcpyAsync(&scanned, I->d_scanned, 8, hipMemcpyDeviceToHost, I->stream));
-    HIP_CHECK(hipMemcpyAsync(&bytes, I->d_bytes, 8, hipMemcpyDeviceToHost, I->stream));
+    HIP_CHECK(hipMemcpyAsync(I->h_resmeta, I->d_found, I->resmeta_bytes,
+                             hipMemcpyDeviceToHost, I->stream));
     HIP_CHECK(hipStreamSynchronize(I->stream));
-    memcpy(found.data(), meta3.data(), nq * 8);
-    memcpy(total.data(), meta3.data() + nq, nq * 8);
-    memcpy(gbytes.data(), meta3.data() + 2 * nq, nq * 8);
-    (void)cnt2;
+    memcpy(found.data(), I->h_resmeta, nq * 8);
+    memcpy(total.data(), I->h_resmeta + I->max_q * 8, nq * 8);
+    memcpy(gbytes.data(), I->h_resmeta + I->max_q * 16, nq * 8);
+    memcpy(ovf.data(), I->h_resmeta + I->max_q * 24, nq * 4);
+    memcpy(&scanned, I->h_resmeta + I->max_q * 28, 8);
+    memcpy(&bytes, I->h_resmeta + I->max_q * 28 + 8, 8);
   }
   float ms = 0;
   (void)hipEventElapsedTime(&ms, I->ev0, I->ev1);
