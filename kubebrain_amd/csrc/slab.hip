@@ -387,6 +387,7 @@ __global__ void k_range_scan2(
     uint64_t* __restrict__ rows_m, int64_t* __restrict__ found_out,
     int64_t* __restrict__ total_out, unsigned long long* __restrict__ scanned_out,
     const int64_t* __restrict__ bounds_g,  // from k_range_bounds
+    int supp_max,  // LDS suppression-stage span cap (KB_SUPP_MAX; 0=off)
     unsigned long long* dbg) {  // KB_SCAN_DBG phase cycles (null in prod)
   int q = blockIdx.x;
   if (q >= nq) return;
@@ -418,7 +419,7 @@ __global__ void k_range_scan2(
   __shared__ uint64_t sl_rev[SUPP_MAX];
   __shared__ uint32_t sl_klen[SUPP_MAX];
   int sl_n = -1;
-  if (dn && dhi_s - dlo_s <= (int64_t)SUPP_MAX) {
+  if (dn && supp_max > 0 && dhi_s - dlo_s <= (int64_t)supp_max) {
     sl_n = (int)(dhi_s - dlo_s);
     for (int64_t v = threadIdx.x; v < (int64_t)sl_n * (KEYW / 8);
          v += blockDim.x)
@@ -1193,6 +1194,7 @@ struct Slab::Impl {
   int max_q = 1024;
   int scan_t = 1024;            // KB_SCAN_T: threads per scan block
   int gather_gwl = 4;           // KB_GATHER_GW: log2 lanes per record group
+  int supp_max = SUPP_MAX;      // KB_SUPP_MAX: LDS suppression stage cap
   int64_t max_cap = 4352;       // winners per query cap (>= limit+1 for etcd's 500)
   int64_t arena_bytes = 384ll << 20;
   DevRangeQ* d_qs = nullptr;
@@ -1508,6 +1510,8 @@ Slab* Slab::Create(int64_t max_rows, int64_t heap_cap, int device,
   I->heap_cap = heap_cap;
   I->max_q = (int)env_i64("KB_MAX_Q", 1024);
   I->scan_t = (int)env_i64("KB_SCAN_T", 1024);
+  I->supp_max = (int)env_i64("KB_SUPP_MAX", SUPP_MAX);
+  if (I->supp_max > SUPP_MAX) I->supp_max = SUPP_MAX;
   {
     int64_t gwv = env_i64("KB_GATHER_GW", 16);
     int l = 0;
@@ -1801,7 +1805,8 @@ bool Slab::RangeBatchStart(const std::vector<DevRangeQ>& qs, std::string* err,
                      I->A.run(), I->n, I->DA.run(), I->dn, I->spillA,
                      I->d_qtails, I->d_qs, nq,
                      I->max_cap, I->d_rows, I->d_rows2, I->d_rowsm, I->d_found,
-                     I->d_total, I->d_scanned, I->d_bounds4, I->d_dbg);
+                     I->d_total, I->d_scanned, I->d_bounds4, I->supp_max,
+                     I->d_dbg);
   HIP_CHECK(hipEventRecord(I->ev1, I->stream));
   hipLaunchKernelGGL(k_gather, dim3(nq), dim3(256), 0, I->stream, I->A.run(),
                      I->DA.run(), I->heapA, I->d_rowsm, I->max_cap,
