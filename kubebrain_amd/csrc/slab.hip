@@ -1251,7 +1251,6 @@ struct Slab::Impl {
   bool ensure_hpack(int64_t need, std::string* err) {
     if (need <= h_pack_cap) return true;
     if (h_pack) (void)hipHostFree(h_pack);
-    if (h_resmeta) (void)hipHostFree(h_resmeta);
     int64_t cap = need + need / 2;
     HIP_CHECK(hipHostMalloc(&h_pack, cap));
     h_pack_cap = cap;
@@ -1784,10 +1783,13 @@ bool Slab::RangeBatch(const std::vector<DevRangeQ>& qs, bool d2h,
   return RangeBatchEx(qs, d2h, true, outs, err, qtails);
 }
 
+static int kb_trace() { static int t = env_i64("KB_TRACE", 0) ? 1 : 0; return t; }
+
 bool Slab::RangeBatchStart(const std::vector<DevRangeQ>& qs, std::string* err,
                            const std::string& qtails) {
   Impl* I = p;
   int nq = (int)qs.size();
+  if (kb_trace()) fprintf(stderr, "[trace] RBStart nq=%d cap0=%lld\n", nq, nq?(long long)qs[0].cap:-1);
   if (nq == 0) return true;
   if (nq > I->max_q) { if (err) *err = "too many queries per batch (KB_MAX_Q)"; return false; }
   int64_t qcap = I->arena_bytes / nq;
@@ -1843,15 +1845,33 @@ bool Slab::RangeBatchFinish(int nq, bool d2h, bool parse,
   std::vector<int32_t> ovf(nq);
   unsigned long long scanned = 0, bytes = 0;
   {
-    HIP_CHECK(hipMemcpyAsync(I->h_resmeta, I->d_found, I->resmeta_bytes,
-                             hipMemcpyDeviceToHost, I->stream));
-    HIP_CHECK(hipStreamSynchronize(I->stream));
+    if (kb_trace()) {
+      fprintf(stderr, "[trace] RBFinish nq=%d readback %lld B h=%p d=%p\n", nq, (long long)I->resmeta_bytes, (void*)I->h_resmeta, (void*)I->d_found);
+      hipPointerAttribute_t pa{};
+      hipError_t pe = hipPointerGetAttributes(&pa, I->d_found);
+      fprintf(stderr, "[trace] dev attr rc=%d type=%d\n", (int)pe, (int)pa.type);
+      pe = hipPointerGetAttributes(&pa, I->h_resmeta);
+      fprintf(stderr, "[trace] host attr rc=%d type=%d\n", (int)pe, (int)pa.type);
+    }
+    if (env_i64("KB_SYNC_READBACK", 0)) {
+      HIP_CHECK(hipStreamSynchronize(I->stream));
+      if (kb_trace()) fprintf(stderr, "[trace] f0 pre-synced\n");
+      HIP_CHECK(hipMemcpy(I->h_resmeta, I->d_found, I->resmeta_bytes,
+                          hipMemcpyDeviceToHost));
+    } else {
+      HIP_CHECK(hipMemcpyAsync(I->h_resmeta, I->d_found, I->resmeta_bytes,
+                               hipMemcpyDeviceToHost, I->stream));
+      if (kb_trace()) fprintf(stderr, "[trace] f1 copy enq\n");
+      HIP_CHECK(hipStreamSynchronize(I->stream));
+    }
+    if (kb_trace()) fprintf(stderr, "[trace] f2 synced\n");
     memcpy(found.data(), I->h_resmeta, nq * 8);
     memcpy(total.data(), I->h_resmeta + I->max_q * 8, nq * 8);
     memcpy(gbytes.data(), I->h_resmeta + I->max_q * 16, nq * 8);
     memcpy(ovf.data(), I->h_resmeta + I->max_q * 24, nq * 4);
     memcpy(&scanned, I->h_resmeta + I->max_q * 28, 8);
     memcpy(&bytes, I->h_resmeta + I->max_q * 28 + 8, 8);
+    if (kb_trace()) fprintf(stderr, "[trace] f3 memcpys done w0=%lld ovf0=%d\n", (long long)found[0], (int)ovf[0]);
   }
   float ms = 0;
   (void)hipEventElapsedTime(&ms, I->ev0, I->ev1);
@@ -1912,6 +1932,7 @@ bool Slab::RangeBatchFinish(int nq, bool d2h, bool parse,
     return true;
   }
   // parse mode (List/Stream): synchronous pack + one D2H + parse
+  if (kb_trace()) fprintf(stderr, "[trace] parse acc=%lld w0=%lld\n", (long long)acc, (long long)(*outs)[0].written);
   I->wait_slot(0);  // d_pack aliases pipeline slot 0
   HIP_CHECK(hipEventRecord(I->ev0, I->stream));
   HIP_CHECK(hipMemcpyAsync(I->d_goffs, goffs.data(), (nq + 1) * 8,
@@ -1926,10 +1947,17 @@ bool Slab::RangeBatchFinish(int nq, bool d2h, bool parse,
   HIP_CHECK(hipStreamSynchronize(I->stream));
   (void)hipEventElapsedTime(&ms, I->ev0, I->ev1);
   perf.pack_d2h_ms += ms;
+  const uint8_t* hp_end = I->h_pack + acc;
   for (int q = 0; q < nq; ++q) {
     RangeResult& r = (*outs)[q];
     if (r.overflow) continue;
     const uint8_t* pp = I->h_pack + goffs[q];
+    if (r.written < 0 || r.written > I->max_cap) {
+      fprintf(stderr, "[parse_bug] q=%d written=%lld total=%lld gbytes=%lld\n",
+              q, (long long)r.written, (long long)r.total, (long long)r.bytes);
+      if (err) *err = "corrupt range result counters";
+      return false;
+    }
     r.recs.reserve(r.written);
     for (int64_t j = 0; j < r.written; ++j) {
       uint64_t rv;
@@ -1937,6 +1965,16 @@ bool Slab::RangeBatchFinish(int nq, bool d2h, bool parse,
       memcpy(&rv, pp, 8);
       memcpy(&klen, pp + 8, 4);
       memcpy(&vlen, pp + 12, 4);
+      if (klen > 8192 || pp + 16 + ((klen + 15) & ~15u) + vlen > hp_end) {
+        fprintf(stderr,
+                "[parse_bug] q=%d j=%lld/%lld rv=%llu klen=%u vlen=%u "
+                "goff=%lld acc=%lld gbytes=%lld\n",
+                q, (long long)j, (long long)r.written,
+                (unsigned long long)rv, klen, vlen, (long long)goffs[q],
+                (long long)acc, (long long)r.bytes);
+        if (err) *err = "corrupt packed record";
+        return false;
+      }
       RangeResult::Rec rec;
       rec.rev = rv;
       rec.key.assign((const char*)pp + 16, klen);
@@ -1945,6 +1983,7 @@ bool Slab::RangeBatchFinish(int nq, bool d2h, bool parse,
       pp += 16 + ((klen + 15) & ~15u) + ((vlen + 15) & ~15u);
     }
   }
+  if (kb_trace()) fprintf(stderr, "[trace] parse done\n");
   return true;
 }
 

@@ -397,6 +397,7 @@ Status Store::StreamNext(int64_t sid, std::vector<KeyValue>* kvs) {
   // (key, rev=+inf) bound past the last chunk's final key (exact for
   // full-width keys; see List).
   const int64_t max_cap = slab_->max_winner_cap();
+  const bool tr = getenv("KB_TRACE") && *getenv("KB_TRACE");
   std::vector<kbslab::RangeResult> outs;
   while ((int64_t)kvs->size() < 300) {
     DevRangeQ q{};
@@ -868,6 +869,7 @@ RangeResponse Store::List(const Bytes& start, const Bytes& end,
   // the next chunk resumes strictly after every row of the last winner's key
   // — exact even for keys of the full 96B width.
   const int64_t max_cap = slab_->max_winner_cap();
+  const bool tr = getenv("KB_TRACE") && *getenv("KB_TRACE");
   std::vector<kbslab::RangeResult> outs;
   std::vector<KeyValue> kvs;
   bool dev_more = true;
@@ -888,6 +890,8 @@ RangeResponse Store::List(const Bytes& start, const Bytes& end,
     q.count_only = 0;
     if (!slab_->RangeBatch({q}, true, &outs, &err, qtails)) { *st = INTERNAL; return resp; }
     kbslab::RangeResult& r = outs[0];
+    if (tr) fprintf(stderr, "[trace] List chunk cap=%lld written=%lld total=%lld recs=%zu kvs=%zu\n",
+                    (long long)q.cap, (long long)r.written, (long long)r.total, r.recs.size(), kvs.size());
     if (r.overflow) {
       // results exceed the device arena: halve the chunk and continue (large
       // values); a single record larger than the arena is a real limit
