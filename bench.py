@@ -547,13 +547,7 @@ def main():
             seen_tx.add(k)
             step_seen.add(k)
             ks.append(k)
-        parts = []
-        for k in ks:
-            parts.append(struct.pack("<IQI", len(k), revs[k], VAL_LEN))
-            parts.append(k)
-            parts.append(vbuf)
         tx_keys.append(ks)
-        tx_blobs.append(b"".join(parts))
     fstep = store._f("bench_step")
     tx_out = np.empty(ntx, dtype=np.uint64)
     tx_out_ptr = tx_out.ctypes.data_as(ctypes.POINTER(ctypes.c_uint64))
@@ -564,13 +558,22 @@ def main():
     def one_step(mode=0):
         # one call: range batch launched async, txn batch overlapped on the
         # host while the kernels are in flight (kb_bench_step). mode bit0 =
-        # d2h (pipelined payload copy), bit1 = keys_only.
+        # d2h (pipelined payload copy), bit1 = keys_only, bit2 = pipelined.
+        # The txn blob is packed HERE, per step, from the tracked revisions:
+        # pre-built blobs go stale once the key pool cycles and every CAS
+        # then fails, silently dropping the write half of the 90/10 mix.
         i = step_i[0] % n_pre
         step_i[0] += 1
+        parts = []
+        for k in tx_keys[i]:
+            parts.append(struct.pack("<IQI", len(k), revs[k], VAL_LEN))
+            parts.append(k)
+            parts.append(vbuf)
+        tx_blob = b"".join(parts)
         total = ctypes.c_ulonglong()
         secs = ctypes.c_double()
         rc = fstep(ctypes.c_void_p(store.h), pre_blobs[i], ctypes.c_size_t(nq),
-                   tx_blobs[i], ctypes.c_size_t(ntx),
+                   tx_blob, ctypes.c_size_t(ntx),
                    ctypes.c_int(mode), tx_out_ptr,
                    ctypes.byref(total), ctypes.byref(secs))
         assert rc == 0

@@ -1492,6 +1492,8 @@ static int64_t env_i64(const char* name, int64_t dflt) {
   return v && *v ? atoll(v) : dflt;
 }
 
+static int kb_trace() { static int t = env_i64("KB_TRACE", 0) ? 1 : 0; return t; }
+
 Slab* Slab::Create(int64_t max_rows, int64_t heap_cap, int device,
                    std::string* err) {
   int cnt = 0;
@@ -1678,6 +1680,8 @@ bool Slab::AppendRows(const uint8_t* keys, const uint64_t* meta,
   HIP_CHECK(hipMemcpyAsync(I->d_dvo, vo, m * 8, hipMemcpyHostToDevice, I->stream));
   HIP_CHECK(hipMemcpyAsync(I->d_dko, ko, m * 8, hipMemcpyHostToDevice, I->stream));
   int64_t new_dn = async ? known_new_dn : 0;  // async: mergeRuns reads it
+  if (kb_trace()) fprintf(stderr, "[trace] AppendRows n=%lld m=%lld async=%d\n",
+                          (long long)I->dn, (long long)m, (int)async);
   Run up{I->d_dkeys, I->d_dmeta, I->d_drev, I->d_dvo, I->d_dko};
   if (!I->mergeRuns(I->DA, I->dn, up, m, I->DB, &new_dn, err,
                     /*device_newn_ok=*/async))
@@ -1782,8 +1786,6 @@ bool Slab::RangeBatch(const std::vector<DevRangeQ>& qs, bool d2h,
                       const std::string& qtails) {
   return RangeBatchEx(qs, d2h, true, outs, err, qtails);
 }
-
-static int kb_trace() { static int t = env_i64("KB_TRACE", 0) ? 1 : 0; return t; }
 
 bool Slab::RangeBatchStart(const std::vector<DevRangeQ>& qs, std::string* err,
                            const std::string& qtails) {
