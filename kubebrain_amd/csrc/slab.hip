@@ -889,6 +889,58 @@ __global__ void k_merge_insert_inv(
   for (int t = 0; t < KEYW / 16; ++t) kd[t] = ks[t];
 }
 
+// multi-block inverted merge rank (any m into n>0 rows): each NEW row
+// binary-searches the big run once — m*log2(n) probes instead of the
+// forward path's n*log2(m) — and leaves the same difference array v as the
+// single-block variant. dup flags go to dup[] (scanned separately to give
+// each new row its drops-before count).
+__global__ void k_merge_rank_inv_big(Run a, int64_t n, Run dnew, int64_t m,
+                                     const uint8_t* __restrict__ spill,
+                                     uint64_t* __restrict__ v /*[n+2] zeroed*/,
+                                     uint64_t* __restrict__ drop /*[n] zeroed*/,
+                                     uint64_t* __restrict__ dup /*[m+1] zeroed*/,
+                                     uint64_t* __restrict__ lbs /*[m]*/) {
+  int64_t j = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  if (j >= m) return;
+  QKey k = row_qk(dnew, spill, j);
+  int64_t lb = d_lb_range(a, spill, 0, n, k, dnew.rev[j]);
+  bool isdup = lb < n && a.rev[lb] == dnew.rev[j] &&
+               rowcmp_q(a, spill, lb, k) == 0;
+  lbs[j] = (uint64_t)lb;
+  if (isdup) {
+    dup[j] = 1;
+    drop[lb] = 1;
+    atomicAdd(&v[lb + 1], (uint64_t)-1ll);
+  }
+  atomicAdd(&v[lb + (isdup ? 1 : 0)], 1ull);
+}
+
+// delta scatter for the big inverted path: slot = lb_j + j - dupx[j];
+// rewrites lbs[j] with the final slot for the same_next fixup
+__global__ void k_merge_scatter_delta2(
+    Run dnew, int64_t m, const uint64_t* __restrict__ dupx /*excl scan*/,
+    uint64_t* __restrict__ lbs, uint8_t* keysB, uint64_t* metaB,
+    uint64_t* revB, uint64_t* voB, uint64_t* koB) {
+  int64_t j = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  if (j >= m) return;
+  int64_t out = (int64_t)lbs[j] + j - (int64_t)dupx[j];
+  lbs[j] = (uint64_t)out;
+  revB[out] = dnew.rev[j];
+  metaB[out] = dnew.meta[j];
+  voB[out] = dnew.vo[j];
+  koB[out] = dnew.ko[j];
+  const uint4* ks = (const uint4*)(dnew.keys + j * KEYW);
+  uint4* kd = (uint4*)(keysB + out * KEYW);
+#pragma unroll
+  for (int t = 0; t < KEYW / 16; ++t) kd[t] = ks[t];
+}
+
+__global__ void k_set_newn(const uint64_t* __restrict__ dupx, int64_t n,
+                           int64_t m, int64_t* __restrict__ newn_out) {
+  if (threadIdx.x == 0 && blockIdx.x == 0)
+    *newn_out = n + m - (int64_t)dupx[m];
+}
+
 // base-row scatter for the inverted path: slot = i + excl_scan(v)[i+1]
 __global__ void k_merge_scatter_base2(
     Run a, const uint64_t* __restrict__ sv /*excl scan of v, [n+2]*/,
@@ -1185,6 +1237,21 @@ struct Slab::Impl {
 
   // scan scratch (u64, shared across ops)
   uint64_t* d_outpos = nullptr;  // [1024] inverted-insert dest slots
+  // big inverted merge scratch: dup flags / their scan / new-row lower
+  // bounds (then final slots), all m-sized, grown on demand
+  uint64_t *d_mdup = nullptr, *d_mdupx = nullptr, *d_mlbs = nullptr;
+  int64_t mergebuf_cap = 0;
+  bool ensure_mergebuf(int64_t m, std::string* err) {
+    if (m + 1 <= mergebuf_cap) return true;
+    int64_t cap = m + m / 2 + 64;
+    for (void* q : {(void*)d_mdup, (void*)d_mdupx, (void*)d_mlbs})
+      if (q) (void)hipFree(q);
+    HIP_CHECK(hipMalloc(&d_mdup, cap * 8));
+    HIP_CHECK(hipMalloc(&d_mdupx, cap * 8));
+    HIP_CHECK(hipMalloc(&d_mlbs, cap * 8));
+    mergebuf_cap = cap;
+    return true;
+  }
   uint64_t *s_a = nullptr, *s_b = nullptr, *s_c = nullptr, *s_d = nullptr,
            *s_e = nullptr;  // max_rows+2
   uint64_t *lv1 = nullptr, *lv1o = nullptr, *lv2 = nullptr, *lv2o = nullptr,
@@ -1319,7 +1386,8 @@ struct Slab::Impl {
                     (void*)d_gq, (void*)d_rows, (void*)d_offs,
                     (void*)d_found,  // base of the resmeta block
                     (void*)d_found32, (void*)d_orev, (void*)d_ometa,
-                    (void*)d_outpos, (void*)d_gbuf,
+                    (void*)d_outpos, (void*)d_mdup, (void*)d_mdupx,
+                    (void*)d_mlbs, (void*)d_gbuf,
                     (void*)d_pack, (void*)d_goffs, (void*)d_bounds,
                     (void*)d_bkeys, (void*)d_brevs, (void*)d_dkeys,
                     (void*)d_dmeta, (void*)d_drev, (void*)d_dvo, (void*)d_dko,
@@ -1440,6 +1508,42 @@ struct Slab::Impl {
       hipLaunchKernelGGL(k_same_next_fixup, dim3((uint32_t)ceil_div(2 * m, 256)),
                          dim3(256), 0, stream, dst.keys, dst.meta, dst.ko,
                          spillA, new_n, d_outpos, m);
+      *out_n = new_n;
+      return true;
+    }
+    if (n > 0 && m > 0) {
+      // big inverted path (folds, bulk appends): m*log2(n) probes + two
+      // prefix sums + streaming scatters. The forward k_merge_rank path
+      // (n*log2(m) random probes — ~400M for a 20.7M x 512k fold) is kept
+      // only for n==0 below.
+      if (!ensure_mergebuf(m, err)) return false;
+      HIP_CHECK(hipMemsetAsync(s_a, 0, (n + 2) * 8, stream));
+      HIP_CHECK(hipMemsetAsync(s_b, 0, n * 8, stream));
+      HIP_CHECK(hipMemsetAsync(d_mdup, 0, (m + 1) * 8, stream));
+      hipLaunchKernelGGL(k_merge_rank_inv_big, dim3((uint32_t)ceil_div(m, 256)),
+                         dim3(256), 0, stream, src.run(), n, dnew, m, spillA,
+                         s_a, s_b, d_mdup, d_mlbs);
+      if (!scan(d_mdup, d_mdupx, m + 1, nullptr, err)) return false;
+      if (!scan(s_a, s_c, n + 2, nullptr, err)) return false;
+      hipLaunchKernelGGL(k_merge_scatter_delta2, dim3((uint32_t)ceil_div(m, 256)),
+                         dim3(256), 0, stream, dnew, m, d_mdupx, d_mlbs,
+                         dst.keys, dst.meta, dst.rev, dst.vo, dst.ko);
+      hipLaunchKernelGGL(k_merge_scatter_base2, dim3((uint32_t)ceil_div(n, 256)),
+                         dim3(256), 0, stream, src.run(), s_c, s_b, dst.keys,
+                         dst.meta, dst.rev, dst.vo, dst.ko, n);
+      hipLaunchKernelGGL(k_set_newn, dim3(1), dim3(64), 0, stream, d_mdupx, n,
+                         m, (int64_t*)s_d);
+      int64_t new_n;
+      if (device_newn_ok) {
+        new_n = *out_n;
+      } else {
+        new_n = 0;
+        HIP_CHECK(hipMemcpyAsync(&new_n, s_d, 8, hipMemcpyDeviceToHost, stream));
+        HIP_CHECK(hipStreamSynchronize(stream));
+      }
+      hipLaunchKernelGGL(k_same_next_fixup, dim3((uint32_t)ceil_div(2 * m, 256)),
+                         dim3(256), 0, stream, dst.keys, dst.meta, dst.ko,
+                         spillA, new_n, d_mlbs, m);
       *out_n = new_n;
       return true;
     }
