@@ -1240,6 +1240,12 @@ struct Slab::Impl {
   // big inverted merge scratch: dup flags / their scan / new-row lower
   // bounds (then final slots), all m-sized, grown on demand
   uint64_t *d_mdup = nullptr, *d_mdupx = nullptr, *d_mlbs = nullptr;
+  uint8_t* h_uploads[2] = {nullptr, nullptr};  // pinned delta-upload mirrors
+  hipEvent_t ev_up[2] = {nullptr, nullptr};
+  int up_idx = 0;
+  DevRangeQ* h_qs[2] = {nullptr, nullptr};  // pinned query-upload mirrors
+  hipEvent_t ev_qs[2] = {nullptr, nullptr};
+  int qs_idx = 0;
   int64_t mergebuf_cap = 0;
   bool ensure_mergebuf(int64_t m, std::string* err) {
     if (m + 1 <= mergebuf_cap) return true;
@@ -1400,6 +1406,12 @@ struct Slab::Impl {
     }
     if (h_pack) (void)hipHostFree(h_pack);
     if (h_resmeta) (void)hipHostFree(h_resmeta);
+    for (int i = 0; i < 2; ++i) {
+      if (h_uploads[i]) (void)hipHostFree(h_uploads[i]);
+      if (ev_up[i]) (void)hipEventDestroy(ev_up[i]);
+      if (h_qs[i]) (void)hipHostFree(h_qs[i]);
+      if (ev_qs[i]) (void)hipEventDestroy(ev_qs[i]);
+    }
     if (h_gmeta) (void)hipHostFree(h_gmeta);
     if (h_gfound) (void)hipHostFree(h_gfound);
     if (ev_g0) (void)hipEventDestroy(ev_g0);
@@ -1454,14 +1466,21 @@ struct Slab::Impl {
   bool ensure_delta(int64_t m, std::string* err) {
     if (m <= upload_cap) return true;
     int64_t cap = m + m / 2 + 1024;
-    for (void* p : {(void*)d_dkeys, (void*)d_dmeta, (void*)d_drev,
-                    (void*)d_dvo, (void*)d_dko})
-      if (p) (void)hipFree(p);
-    HIP_CHECK(hipMalloc(&d_dkeys, cap * KEYW));
-    HIP_CHECK(hipMalloc(&d_dmeta, cap * 8));
-    HIP_CHECK(hipMalloc(&d_drev, cap * 8));
-    HIP_CHECK(hipMalloc(&d_dvo, cap * 8));
-    HIP_CHECK(hipMalloc(&d_dko, cap * 8));
+    // one device block + one pinned mirror: per-step uploads become a single
+    // host memcpy + ONE async H2D (five pageable copies each stalled the
+    // host on the staging pool)
+    if (d_dkeys) (void)hipFree(d_dkeys);
+    for (int i = 0; i < 2; ++i) {
+      if (h_uploads[i]) (void)hipHostFree(h_uploads[i]);
+      HIP_CHECK(hipHostMalloc(&h_uploads[i], cap * (KEYW + 32)));
+      if (!ev_up[i]) HIP_CHECK(hipEventCreate(&ev_up[i]));
+    }
+    int64_t bytes = cap * (KEYW + 32);
+    HIP_CHECK(hipMalloc(&d_dkeys, bytes));
+    d_dmeta = (uint64_t*)(d_dkeys + cap * KEYW);
+    d_drev = d_dmeta + cap;
+    d_dvo = d_drev + cap;
+    d_dko = d_dvo + cap;
     upload_cap = cap;
     return true;
   }
@@ -1778,11 +1797,26 @@ bool Slab::AppendRows(const uint8_t* keys, const uint64_t* meta,
   bool async = known_new_dn >= 0 && !validate;
   if (!async) HIP_CHECK(hipEventRecord(I->ev0, I->stream));
   if (!I->ensure_delta(m, err)) return false;
-  HIP_CHECK(hipMemcpyAsync(I->d_dkeys, keys, m * KEYW, hipMemcpyHostToDevice, I->stream));
-  HIP_CHECK(hipMemcpyAsync(I->d_dmeta, meta, m * 8, hipMemcpyHostToDevice, I->stream));
-  HIP_CHECK(hipMemcpyAsync(I->d_drev, rev, m * 8, hipMemcpyHostToDevice, I->stream));
-  HIP_CHECK(hipMemcpyAsync(I->d_dvo, vo, m * 8, hipMemcpyHostToDevice, I->stream));
-  HIP_CHECK(hipMemcpyAsync(I->d_dko, ko, m * 8, hipMemcpyHostToDevice, I->stream));
+  {
+    // double-buffered pinned mirror: wait only for THIS slot's previous
+    // upload (two steps back — long since done), never the whole stream
+    int slot = I->up_idx ^= 1;
+    HIP_CHECK(hipEventSynchronize(I->ev_up[slot]));
+    int64_t cap = I->upload_cap;
+    uint8_t* h = I->h_uploads[slot];
+    memcpy(h, keys, m * KEYW);
+    memcpy(h + cap * KEYW, meta, m * 8);
+    memcpy(h + cap * KEYW + cap * 8, rev, m * 8);
+    memcpy(h + cap * KEYW + cap * 16, vo, m * 8);
+    memcpy(h + cap * KEYW + cap * 24, ko, m * 8);
+    // pinned source => truly async enqueues (no staging-pool stall)
+    HIP_CHECK(hipMemcpyAsync(I->d_dkeys, h, m * KEYW, hipMemcpyHostToDevice, I->stream));
+    HIP_CHECK(hipMemcpyAsync(I->d_dmeta, h + cap * KEYW, m * 8, hipMemcpyHostToDevice, I->stream));
+    HIP_CHECK(hipMemcpyAsync(I->d_drev, h + cap * KEYW + cap * 8, m * 8, hipMemcpyHostToDevice, I->stream));
+    HIP_CHECK(hipMemcpyAsync(I->d_dvo, h + cap * KEYW + cap * 16, m * 8, hipMemcpyHostToDevice, I->stream));
+    HIP_CHECK(hipMemcpyAsync(I->d_dko, h + cap * KEYW + cap * 24, m * 8, hipMemcpyHostToDevice, I->stream));
+    HIP_CHECK(hipEventRecord(I->ev_up[slot], I->stream));
+  }
   int64_t new_dn = async ? known_new_dn : 0;  // async: mergeRuns reads it
   if (kb_trace()) fprintf(stderr, "[trace] AppendRows n=%lld m=%lld async=%d\n",
                           (long long)I->dn, (long long)m, (int)async);
@@ -1901,8 +1935,22 @@ bool Slab::RangeBatchStart(const std::vector<DevRangeQ>& qs, std::string* err,
   int64_t qcap = I->arena_bytes / nq;
   qcap &= ~15ll;
   if (!I->ensure_qtails(qtails, err)) return false;
-  HIP_CHECK(hipMemcpyAsync(I->d_qs, qs.data(), sizeof(DevRangeQ) * nq,
-                           hipMemcpyHostToDevice, I->stream));
+  {
+    // pinned double-buffered query upload (a 900-query batch is ~200 KB;
+    // pageable hipMemcpyAsync stalls the host on the staging pool)
+    if (!I->h_qs[0]) {
+      for (int i = 0; i < 2; ++i) {
+        HIP_CHECK(hipHostMalloc(&I->h_qs[i], sizeof(DevRangeQ) * I->max_q));
+        HIP_CHECK(hipEventCreate(&I->ev_qs[i]));
+      }
+    }
+    int slot = I->qs_idx ^= 1;
+    HIP_CHECK(hipEventSynchronize(I->ev_qs[slot]));
+    memcpy(I->h_qs[slot], qs.data(), sizeof(DevRangeQ) * nq);
+    HIP_CHECK(hipMemcpyAsync(I->d_qs, I->h_qs[slot], sizeof(DevRangeQ) * nq,
+                             hipMemcpyHostToDevice, I->stream));
+    HIP_CHECK(hipEventRecord(I->ev_qs[slot], I->stream));
+  }
   HIP_CHECK(hipMemsetAsync(I->d_scanned, 0, 8, I->stream));
   HIP_CHECK(hipMemsetAsync(I->d_bytes, 0, 8, I->stream));
   HIP_CHECK(hipEventRecord(I->ev0, I->stream));
