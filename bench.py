@@ -621,6 +621,26 @@ def main():
     p = perf(store)
 
     # one-off PCIe reference: pinned D2H bandwidth of this box (torch copy)
+    # box-health probe: host spin speed + GPU sclk, reported with every line
+    # (box-to-box variance on this pool reaches 3x; a slow `value` with a
+    # slow probe is the box, not the code)
+    box = {}
+    try:
+        t0 = time.time()
+        x = 0
+        for i in range(3_000_000):
+            x += i * i
+        box["host_spin_ms"] = round((time.time() - t0) * 1e3, 1)
+        import subprocess
+        out = subprocess.run(["rocm-smi", "--showgpuclocks", "--csv"],
+                             capture_output=True, text=True, timeout=10).stdout
+        for tok in out.replace(",", " ").split():
+            if tok.endswith("Mhz") or tok.endswith("MHz"):
+                box["sclk_mhz"] = int("".join(c for c in tok if c.isdigit()))
+                break
+    except Exception:
+        pass
+
     pcie_gbps = None
     if torch.cuda.is_available():
         try:
@@ -803,6 +823,7 @@ def main():
             "d2h_detail": d2h_detail,
             "d2h_keys_only_detail": d2h_ko_detail,
             "pcie_d2h_GBps": pcie_gbps,
+            "box": box,
             "step_split_ms": {"range": round(split_snapshot["range_s"] / args.steps * 1e3, 3),
                                "txn": round(split_snapshot["txn_s"] / args.steps * 1e3, 3)},
             "roofline": roofline,

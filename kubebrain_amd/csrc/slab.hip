@@ -1677,7 +1677,7 @@ Slab* Slab::Create(int64_t max_rows, int64_t heap_cap, int device,
   I->spill_cap = env_i64("KB_SPILL_BYTES", 256ll << 20);
   HIP_CHECK_NULL(hipMalloc(&I->spillA, I->spill_cap));
   HIP_CHECK_NULL(hipMalloc(&I->spillB, I->spill_cap));
-  I->delta_cap = env_i64("KB_DELTA_CAP", 1 << 19);
+  I->delta_cap = env_i64("KB_DELTA_CAP", 1 << 16);
   if (I->delta_cap > max_rows) I->delta_cap = max_rows;
   for (Impl::Col* c : {&I->DA, &I->DB}) {
     HIP_CHECK_NULL(hipMalloc(&c->keys, I->delta_cap * KEYW));
