@@ -1267,6 +1267,7 @@ struct Slab::Impl {
   int max_q = 1024;
   int scan_t = 1024;            // KB_SCAN_T: threads per scan block
   int gather_gwl = 4;           // KB_GATHER_GW: log2 lanes per record group
+  int gather_t = 512;           // KB_GATHER_T: threads per gather_copy block
   int supp_max = SUPP_MAX;      // KB_SUPP_MAX: LDS suppression stage cap
   int64_t max_cap = 4352;       // winners per query cap (>= limit+1 for etcd's 500)
   int64_t arena_bytes = 384ll << 20;
@@ -1636,6 +1637,10 @@ Slab* Slab::Create(int64_t max_rows, int64_t heap_cap, int device,
   I->scan_t = (int)env_i64("KB_SCAN_T", 1024);
   I->supp_max = (int)env_i64("KB_SUPP_MAX", SUPP_MAX);
   if (I->supp_max > SUPP_MAX) I->supp_max = SUPP_MAX;
+  I->gather_t = (int)env_i64("KB_GATHER_T", 512);
+  if (I->gather_t < 64) I->gather_t = 64;
+  if (I->gather_t > 1024) I->gather_t = 1024;
+  I->gather_t &= ~63;
   {
     int64_t gwv = env_i64("KB_GATHER_GW", 16);
     int l = 0;
@@ -1968,7 +1973,7 @@ bool Slab::RangeBatchStart(const std::vector<DevRangeQ>& qs, std::string* err,
                      I->DA.run(), I->heapA, I->d_rowsm, I->max_cap,
                      I->d_found, I->d_qs, nq, I->d_gbuf, qcap, I->d_offs,
                      I->d_gbytes, I->d_ovf, I->d_bytes);
-  hipLaunchKernelGGL(k_gather_copy, dim3(nq), dim3(512), 0, I->stream,
+  hipLaunchKernelGGL(k_gather_copy, dim3(nq), dim3((uint32_t)I->gather_t), 0, I->stream,
                      I->A.run(), I->DA.run(), I->spillA, I->heapA, I->d_rowsm,
                      I->max_cap, I->d_found, I->d_qs, nq, I->d_gbuf, qcap,
                      I->d_offs, I->d_ovf, I->gather_gwl);
