@@ -192,21 +192,27 @@ constexpr uint64_t ROW_MASK = ROW_TAG_DELTA - 1;
 // scan block width: runtime-tunable (KB_SCAN_T in {256,512,1024}); the
 // kernel reads blockDim.x, only the LDS wave-count array is sized for the max
 constexpr int SCAN_T_MAX = 1024;
-// max delta-suppression span staged in LDS per query block (~55 KB of LDS:
-// still 2 blocks/CU of 160 KB; the stage is a coalesced span-sized read that
-// replaces ~cap random binary searches over HBM)
-constexpr int SUPP_MAX = 512;
 
-// compare LDS-staged suppression row j against query key q (sign as
-// rowcmp_q); exact 96B-prefix ties where either side exceeds KEYW fall back
-// to the global-memory compare (spill tails are never staged)
-__device__ __forceinline__ int supp_cmp(const uint64_t* sl_keys,
-                                        const uint32_t* sl_klen, int j,
-                                        const Run& srun, const uint8_t* spill,
-                                        int64_t slo, const QKey& q) {
-  int c = keycmp96((const uint8_t*)(sl_keys + (int64_t)j * (KEYW / 8)), q.k96);
-  if (c || (sl_klen[j] <= (uint32_t)KEYW && q.len <= (uint32_t)KEYW)) return c;
-  return rowcmp_q(srun, spill, slo + j, q);
+// shadow-revision column (base run only): shadow[i] = smallest delta-run
+// revision (>=1) of base row i's key, UINT64_MAX = the delta holds no row of
+// that key. Maintained by k_shadow_mark at delta-insert time and reset to
+// MAX whenever the base is rebuilt (fold/compact empty the delta). Turns the
+// scan's winner-suppression probe — a per-winner binary search over the
+// delta — into one coalesced 8B stream read: base winner i is suppressed
+// iff shadow[i] != MAX && shadow[i] <= R (a strictly-newer delta row of the
+// key exists at or below the read revision).
+__global__ void k_shadow_mark(Run b, int64_t n, Run dnew, int64_t m,
+                              const uint8_t* __restrict__ spill,
+                              uint64_t* __restrict__ shadow) {
+  int64_t j = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  if (j >= m) return;
+  uint64_t rv = dnew.rev[j];
+  if (rv == 0) return;  // rev-row holes never suppress
+  QKey k = row_qk(dnew, spill, j);
+  for (int64_t i = d_lb_range(b, spill, 0, n, k, 1); i < n; ++i) {
+    if (rowcmp_q(b, spill, i, k) != 0) break;
+    atomicMin((unsigned long long*)&shadow[i], (unsigned long long)rv);
+  }
 }
 
 // one run's winner scan (ordered append); returns written, *total = seen.
@@ -219,12 +225,8 @@ __device__ __forceinline__ int supp_cmp(const uint64_t* sl_keys,
 __device__ int64_t scan_run_winners(
     const Run& run, const uint8_t* spill, int64_t lo, int64_t hi, uint64_t R,
     int64_t cap, uint64_t* out, int64_t out_cap, uint64_t tagbit,
-    const Run* srun, int64_t slo, int64_t shi,  // suppression run (null => none)
+    const uint64_t* shadow,  // shadow-revision column (null => no suppression)
     int64_t* total_out, int64_t* scanned_accum, int* wave_cnt,
-    // LDS-staged copy of the suppression span (sl_n = -1 => not staged):
-    // probes binary-search LDS instead of paying ~9 HBM round trips each
-    const uint64_t* sl_keys, const uint64_t* sl_rev, const uint32_t* sl_klen,
-    int sl_n,
     unsigned long long* dbg) {  // KB_SCAN_DBG phase cycles (null in prod)
   const uint64_t* __restrict__ rev = run.rev;
   const uint64_t* __restrict__ meta = run.meta;
@@ -254,6 +256,13 @@ __device__ int64_t scan_run_winners(
     // same_next set => row i+1 exists, shares the key, and (rows of one key
     // being contiguous and its key < qend) lies below hi, so the rev[i+1]
     // load is gated by i+1<hi alone, independent of the meta bits.
+    // pass 1: streaming winner predicate — rev/meta/shadow 8B streams only,
+    // no probes and no divergent chains, so the loads pipeline across
+    // rounds. same_next set => row i+1 exists, shares the key, and (rows of
+    // one key being contiguous and its key < qend) lies below hi, so the
+    // rev[i+1] load is gated by i+1<hi alone, independent of the meta bits.
+    // Suppression is the shadow-revision stream: a base winner loses iff a
+    // strictly-newer delta row of its key exists at or below R.
     uint64_t flags = 0;
 #pragma unroll 4
     for (int r = 0; r < rounds; ++r) {
@@ -262,50 +271,17 @@ __device__ int64_t scan_run_winners(
       if (i < hi) {
         uint64_t rv = rev[i], m = meta[i];
         uint64_t rvn = (i + 1 < hi) ? rev[i + 1] : 0;
-        if (rv > 0 && rv <= R && !(m & M_TOMB))
+        uint64_t sh = shadow ? shadow[i] : UINT64_MAX;
+        if (rv > 0 && rv <= R && !(m & M_TOMB) &&
+            !(sh != UINT64_MAX && sh <= R))
           win = !(m & M_SAME_NEXT) || rvn > R;
       }
       if (win) flags |= 1ull << r;
     }
     if (dbg0) { unsigned long long t1 = wall_clock64(); atomicAdd(&dbg[1], t1 - tprev); tprev = t1; }
-    // pass 1b: delta-suppression probes on flagged rows only (~cap of the
-    // 6*cap-row tile), so the random binary searches no longer stall the
-    // 16 B/row stream of pass 1a
     uint32_t wcnt = 0;
-    for (int r = 0; r < rounds; ++r) {
-      bool win = (flags >> r) & 1;
-      uint64_t b = __ballot(win);
-      if (srun && b) {
-        if (win) {
-          int64_t i = wbase + ((int64_t)r << 6) + lane;
-          QKey kk = row_qk(run, spill, i);
-          bool supp;
-          if (sl_n >= 0) {  // LDS-staged span: search without HBM latency
-            int lo2 = 0, hi2 = sl_n;
-            while (lo2 < hi2) {  // lower bound of (kk, rev>=1), as d_lb_range
-              int mid = (lo2 + hi2) >> 1;
-              int c = supp_cmp(sl_keys, sl_klen, mid, *srun, spill, slo, kk);
-              if (c < 0 || (c == 0 && sl_rev[mid] < 1))
-                lo2 = mid + 1;
-              else
-                hi2 = mid;
-            }
-            supp = lo2 < sl_n && sl_rev[lo2] <= R &&
-                   supp_cmp(sl_keys, sl_klen, lo2, *srun, spill, slo, kk) == 0;
-          } else {
-            int64_t lb = d_lb_range(*srun, spill, slo, shi, kk, 1);
-            supp = lb < shi && srun->rev[lb] <= R &&
-                   rowcmp_q(*srun, spill, lb, kk) == 0;
-          }
-          if (supp) {
-            win = false;  // a newer (delta) row of this key wins instead
-            flags &= ~(1ull << r);
-          }
-        }
-        b = __ballot(win);
-      }
-      wcnt += (uint32_t)__popcll(b);
-    }
+    for (int r = 0; r < rounds; ++r)
+      wcnt += (uint32_t)__popcll(__ballot((flags >> r) & 1));
     if (dbg0) { unsigned long long t2 = wall_clock64(); atomicAdd(&dbg[2], t2 - tprev); tprev = t2; }
     if (lane == 0) wave_cnt[w] = (int)wcnt;
     __syncthreads();
@@ -387,7 +363,7 @@ __global__ void k_range_scan2(
     uint64_t* __restrict__ rows_m, int64_t* __restrict__ found_out,
     int64_t* __restrict__ total_out, unsigned long long* __restrict__ scanned_out,
     const int64_t* __restrict__ bounds_g,  // from k_range_bounds
-    int supp_max,  // LDS suppression-stage span cap (KB_SUPP_MAX; 0=off)
+    const uint64_t* __restrict__ shadow,  // base shadow-revision column
     unsigned long long* dbg) {  // KB_SCAN_DBG phase cycles (null in prod)
   int q = blockIdx.x;
   if (q >= nq) return;
@@ -410,35 +386,11 @@ __global__ void k_range_scan2(
   uint64_t* outb = Q.count_only ? nullptr : rows_b + (int64_t)q * max_cap;
   int64_t nB = scan_run_winners(d, spill, dlo_s, dhi_s, Q.read_rev,
                                 cap, outd, max_cap, ROW_TAG_DELTA, nullptr,
-                                0, 0, &dtotal, &scanned, wave_cnt,
-                                nullptr, nullptr, nullptr, -1, dbg);
-  // stage the delta-suppression span in LDS when it is small (the common
-  // case: the delta run folds long before any namespace accrues >SUPP_MAX
-  // rows), so base-scan probes binary-search LDS instead of HBM
-  __shared__ uint64_t sl_keys[SUPP_MAX * (KEYW / 8)];
-  __shared__ uint64_t sl_rev[SUPP_MAX];
-  __shared__ uint32_t sl_klen[SUPP_MAX];
-  int sl_n = -1;
-  if (dn && supp_max > 0 && dhi_s - dlo_s <= (int64_t)supp_max) {
-    sl_n = (int)(dhi_s - dlo_s);
-    for (int64_t v = threadIdx.x; v < (int64_t)sl_n * (KEYW / 8);
-         v += blockDim.x)
-      sl_keys[v] =
-          ((const uint64_t*)(d.keys + dlo_s * KEYW))[v];
-    for (int64_t j = threadIdx.x; j < sl_n; j += blockDim.x) {
-      sl_rev[j] = d.rev[dlo_s + j];
-      sl_klen[j] = meta_klen(d.meta[dlo_s + j]);
-    }
-  }
-  if (dbg0) {
-    atomicAdd(&dbg[6], (unsigned long long)(sl_n >= 0 ? 1 : 0));
-    atomicAdd(&dbg[7], (unsigned long long)(dhi_s - dlo_s));
-  }
-  __syncthreads();  // wave_cnt handoff between the two runs + staged span
+                                &dtotal, &scanned, wave_cnt, dbg);
+  __syncthreads();  // wave_cnt handoff between the two runs
   int64_t nA = scan_run_winners(b, spill, lo_s, hi_s, Q.read_rev, cap,
-                                outb, max_cap, 0, dn ? &d : nullptr,
-                                dlo_s, dhi_s, &btotal, &scanned, wave_cnt,
-                                sl_keys, sl_rev, sl_klen, sl_n, dbg);
+                                outb, max_cap, 0, dn ? shadow : nullptr,
+                                &btotal, &scanned, wave_cnt, dbg);
   __syncthreads();  // winner lists complete before the merge reads them
   unsigned long long tkm = dbg0 ? wall_clock64() : 0;
   // merge by rank into rows_m (keys are disjoint across the two lists)
@@ -1236,6 +1188,7 @@ struct Slab::Impl {
   }
 
   // scan scratch (u64, shared across ops)
+  uint64_t* shadow = nullptr;  // [max_rows] base shadow-revision column
   uint64_t* d_outpos = nullptr;  // [1024] inverted-insert dest slots
   // big inverted merge scratch: dup flags / their scan / new-row lower
   // bounds (then final slots), all m-sized, grown on demand
@@ -1268,7 +1221,6 @@ struct Slab::Impl {
   int scan_t = 1024;            // KB_SCAN_T: threads per scan block
   int gather_gwl = 4;           // KB_GATHER_GW: log2 lanes per record group
   int gather_t = 512;           // KB_GATHER_T: threads per gather_copy block
-  int supp_max = SUPP_MAX;      // KB_SUPP_MAX: LDS suppression stage cap
   int64_t max_cap = 4352;       // winners per query cap (>= limit+1 for etcd's 500)
   int64_t arena_bytes = 384ll << 20;
   DevRangeQ* d_qs = nullptr;
@@ -1393,7 +1345,7 @@ struct Slab::Impl {
                     (void*)d_gq, (void*)d_rows, (void*)d_offs,
                     (void*)d_found,  // base of the resmeta block
                     (void*)d_found32, (void*)d_orev, (void*)d_ometa,
-                    (void*)d_outpos, (void*)d_mdup, (void*)d_mdupx,
+                    (void*)shadow, (void*)d_outpos, (void*)d_mdup, (void*)d_mdupx,
                     (void*)d_mlbs, (void*)d_gbuf,
                     (void*)d_pack, (void*)d_goffs, (void*)d_bounds,
                     (void*)d_bkeys, (void*)d_brevs, (void*)d_dkeys,
@@ -1635,8 +1587,6 @@ Slab* Slab::Create(int64_t max_rows, int64_t heap_cap, int device,
   I->heap_cap = heap_cap;
   I->max_q = (int)env_i64("KB_MAX_Q", 1024);
   I->scan_t = (int)env_i64("KB_SCAN_T", 1024);
-  I->supp_max = (int)env_i64("KB_SUPP_MAX", SUPP_MAX);
-  if (I->supp_max > SUPP_MAX) I->supp_max = SUPP_MAX;
   I->gather_t = (int)env_i64("KB_GATHER_T", 512);
   if (I->gather_t < 64) I->gather_t = 64;
   if (I->gather_t > 1024) I->gather_t = 1024;
@@ -1712,6 +1662,8 @@ Slab* Slab::Create(int64_t max_rows, int64_t heap_cap, int device,
   HIP_CHECK_NULL(hipMalloc(&I->d_rows2, (int64_t)I->max_q * I->max_cap * 8));
   HIP_CHECK_NULL(hipMalloc(&I->d_rowsm, (int64_t)I->max_q * I->max_cap * 8));
   HIP_CHECK_NULL(hipMalloc(&I->d_bounds4, (int64_t)I->max_q * 4 * 8));
+  HIP_CHECK_NULL(hipMalloc(&I->shadow, max_rows * 8));
+  HIP_CHECK_NULL(hipMemset(I->shadow, 0xFF, max_rows * 8));
   HIP_CHECK_NULL(hipMalloc(&I->d_offs, (int64_t)I->max_q * I->max_cap * 8));
   {
     // found/total/gbytes/ovf/scanned/bytes live in ONE allocation so
@@ -1826,6 +1778,13 @@ bool Slab::AppendRows(const uint8_t* keys, const uint64_t* meta,
   if (kb_trace()) fprintf(stderr, "[trace] AppendRows n=%lld m=%lld async=%d\n",
                           (long long)I->dn, (long long)m, (int)async);
   Run up{I->d_dkeys, I->d_dmeta, I->d_drev, I->d_dvo, I->d_dko};
+  // shadow-mark the base rows of every inserted key BEFORE the merge
+  // kernels queue (same stream => ordering is irrelevant, both read only
+  // the base run and the upload)
+  if (I->n > 0)
+    hipLaunchKernelGGL(k_shadow_mark, dim3((uint32_t)ceil_div(m, 256)),
+                       dim3(256), 0, I->stream, I->A.run(), I->n, up, m,
+                       I->spillA, I->shadow);
   if (!I->mergeRuns(I->DA, I->dn, up, m, I->DB, &new_dn, err,
                     /*device_newn_ok=*/async))
     return false;
@@ -1871,6 +1830,8 @@ bool Slab::Fold(std::string* err) {
   std::swap(I->A, I->B);
   I->n = new_n;
   I->dn = 0;
+  // base rebuilt + delta emptied => no key has a delta row
+  HIP_CHECK(hipMemsetAsync(I->shadow, 0xFF, I->n * 8, I->stream));
   HIP_CHECK(hipStreamSynchronize(I->stream));
   return true;
 }
@@ -1914,6 +1875,8 @@ bool Slab::Merge(const DeltaRows& d, std::string* err) {
       return false;
     std::swap(I->A, I->B);
     I->n = new_n;
+    // base rebuilt, delta empty (folded above) => clear shadows
+    HIP_CHECK(hipMemsetAsync(I->shadow, 0xFF, I->n * 8, I->stream));
   }
   HIP_CHECK(hipEventRecord(I->ev1, I->stream));
   HIP_CHECK(hipStreamSynchronize(I->stream));
@@ -1966,7 +1929,7 @@ bool Slab::RangeBatchStart(const std::vector<DevRangeQ>& qs, std::string* err,
                      I->A.run(), I->n, I->DA.run(), I->dn, I->spillA,
                      I->d_qtails, I->d_qs, nq,
                      I->max_cap, I->d_rows, I->d_rows2, I->d_rowsm, I->d_found,
-                     I->d_total, I->d_scanned, I->d_bounds4, I->supp_max,
+                     I->d_total, I->d_scanned, I->d_bounds4, I->shadow,
                      I->d_dbg);
   HIP_CHECK(hipEventRecord(I->ev1, I->stream));
   hipLaunchKernelGGL(k_gather, dim3(nq), dim3(256), 0, I->stream, I->A.run(),
@@ -2345,6 +2308,9 @@ bool Slab::Compact(const std::vector<std::pair<Bound, Bound>>& borders,
   std::swap(I->heapA, I->heapB);
   std::swap(I->spillA, I->spillB);
   I->n = (int64_t)kept;
+  // compaction folds the delta first and rewrites the base => clear shadows
+  if (I->n > 0)
+    HIP_CHECK(hipMemsetAsync(I->shadow, 0xFF, I->n * 8, I->stream));
   I->heap_used_ = (int64_t)new_heap;
   I->spill_used_ = (int64_t)new_spill;
   HIP_CHECK(hipStreamSynchronize(I->stream));
