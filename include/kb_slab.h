@@ -184,7 +184,13 @@ int kb_bench_txn(kb_store*, const uint8_t* tbuf, size_t n, uint64_t* out_revs);
 /* one bench step: Range batch launched async + txn batch overlapped on the
  * host while the kernels are in flight (results unchanged: kernels snapshot
  * device state at launch; writes stage host-side until the next sync).
- * mode bits as kb_bench_range. */
+ * mode bits as kb_bench_range, plus:
+ * 4 = pipelined (excludes bit 1) — this step's Range batch stays in flight;
+ *     `total` reports the PREVIOUS step's winners (first call reports 0) and
+ *     kb_sync collects the final batch. Exact under MVCC: every Range reads
+ *     at its fixed read_rev, so deferring collection by one step never
+ *     changes results; txn CAS results (out_revs) are still returned
+ *     in-step. */
 int kb_bench_step(kb_store*, const uint8_t* qbuf, size_t nq,
                   const uint8_t* tbuf, size_t ntx, int mode, uint64_t* out_revs,
                   unsigned long long* total, double* secs);
