@@ -264,8 +264,23 @@ __device__ int64_t scan_run_winners(
     // Suppression is the shadow-revision stream: a base winner loses iff a
     // strictly-newer delta row of its key exists at or below R.
     uint64_t flags = 0;
+    // full rounds carry no bounds check (every lane in range), so the
+    // rev/meta/shadow loads have no control dependence and pipeline across
+    // the unrolled iterations; only the ragged tail round is guarded
+    const int full = (int)min((int64_t)rounds, (hi - wbase) >> 6);
 #pragma unroll 4
-    for (int r = 0; r < rounds; ++r) {
+    for (int r = 0; r < full; ++r) {
+      int64_t i = wbase + ((int64_t)r << 6) + lane;
+      uint64_t rv = rev[i], m = meta[i];
+      uint64_t rvn = (i + 1 < hi) ? rev[i + 1] : 0;
+      uint64_t sh = shadow ? shadow[i] : UINT64_MAX;
+      bool win = false;
+      if (rv > 0 && rv <= R && !(m & M_TOMB) &&
+          !(sh != UINT64_MAX && sh <= R))
+        win = !(m & M_SAME_NEXT) || rvn > R;
+      if (win) flags |= 1ull << r;
+    }
+    for (int r = full; r < rounds; ++r) {
       int64_t i = wbase + ((int64_t)r << 6) + lane;
       bool win = false;
       if (i < hi) {
@@ -600,6 +615,58 @@ __global__ void k_gather_copy(Run rb, Run rd,
     if (gl == 0)
       for (uint32_t b = w16 * 16; b < vlen; ++b) vd[b] = vs[b];
   }
+}
+
+// wave-per-record gather (KB_GATHER_MODE=2): grid (nq, cap/4), 4 records per
+// 256-thread block, one 64-lane wave each — a 512B value is ONE uint4 round
+// over 32 lanes and the 96B key one u64 round, with no serial record chain
+// inside a lane group. Idle blocks (j >= found) retire immediately.
+__global__ void k_gather_copy2(Run rb, Run rd,
+                               const uint8_t* __restrict__ spill,
+                               const uint8_t* __restrict__ heap,
+                               const uint64_t* __restrict__ rows_out,
+                               int64_t max_cap,
+                               const int64_t* __restrict__ found_out,
+                               const DevRangeQ* __restrict__ qs, int nq,
+                               uint8_t* __restrict__ gbuf, int64_t qcap,
+                               const int64_t* __restrict__ offs,
+                               const int32_t* __restrict__ overflow) {
+  int q = blockIdx.x;
+  if (q >= nq || overflow[q]) return;
+  int64_t j = (int64_t)blockIdx.y * (blockDim.x >> 6) + (threadIdx.x >> 6);
+  if (j >= found_out[q]) return;
+  const bool konly = qs[q].keys_only != 0;
+  const int lane = threadIdx.x & 63;
+  uint64_t rt = rows_out[(int64_t)q * max_cap + j];
+  bool isd = (rt & ROW_TAG_DELTA) != 0;
+  int64_t row = (int64_t)(rt & ROW_MASK);
+  const Run& r = isd ? rd : rb;
+  uint64_t m = r.meta[row];
+  uint32_t klen = meta_klen(m), vlen = konly ? 0 : meta_vlen(m);
+  uint8_t* dst = gbuf + (int64_t)q * qcap + offs[(int64_t)q * max_cap + j];
+  if (lane == 0) {
+    *(uint64_t*)dst = r.rev[row];
+    ((uint32_t*)dst)[2] = klen;
+    ((uint32_t*)dst)[3] = vlen;
+  }
+  const uint8_t* ks = r.keys + row * KEYW;
+  uint8_t* kd = dst + 16;  // records are 16B-aligned
+  uint32_t kin = klen > (uint32_t)KEYW ? (uint32_t)KEYW : klen;
+  for (uint32_t b = lane; b < (kin + 7) / 8; b += 64)
+    ((uint64_t*)kd)[b] = ((const uint64_t*)ks)[b];
+  if (klen > (uint32_t)KEYW) {  // spill tail (keys > 96B)
+    const uint8_t* ts = spill + r.ko[row];
+    for (uint32_t t = lane; t < klen - (uint32_t)KEYW; t += 64)
+      kd[KEYW + t] = ts[t];
+  }
+  if (konly) return;
+  const uint8_t* vs = heap + r.vo[row];
+  uint8_t* vd = dst + 16 + ((klen + 15) & ~15u);
+  uint32_t w16 = vlen >> 4;
+  for (uint32_t b = lane; b < w16; b += 64)
+    ((uint4*)vd)[b] = ((const uint4*)vs)[b];
+  if (lane == 0)
+    for (uint32_t b = w16 * 16; b < vlen; ++b) vd[b] = vs[b];
 }
 
 // pack used per-query regions contiguously for one D2H
@@ -1221,6 +1288,7 @@ struct Slab::Impl {
   int scan_t = 1024;            // KB_SCAN_T: threads per scan block
   int gather_gwl = 4;           // KB_GATHER_GW: log2 lanes per record group
   int gather_t = 512;           // KB_GATHER_T: threads per gather_copy block
+  int gather_mode = 1;          // KB_GATHER_MODE: 1=record groups, 2=wave/record
   int64_t max_cap = 4352;       // winners per query cap (>= limit+1 for etcd's 500)
   int64_t arena_bytes = 384ll << 20;
   DevRangeQ* d_qs = nullptr;
@@ -1588,6 +1656,7 @@ Slab* Slab::Create(int64_t max_rows, int64_t heap_cap, int device,
   I->max_q = (int)env_i64("KB_MAX_Q", 1024);
   I->scan_t = (int)env_i64("KB_SCAN_T", 1024);
   I->gather_t = (int)env_i64("KB_GATHER_T", 512);
+  I->gather_mode = (int)env_i64("KB_GATHER_MODE", 1);
   if (I->gather_t < 64) I->gather_t = 64;
   if (I->gather_t > 1024) I->gather_t = 1024;
   I->gather_t &= ~63;
@@ -1936,10 +2005,23 @@ bool Slab::RangeBatchStart(const std::vector<DevRangeQ>& qs, std::string* err,
                      I->DA.run(), I->heapA, I->d_rowsm, I->max_cap,
                      I->d_found, I->d_qs, nq, I->d_gbuf, qcap, I->d_offs,
                      I->d_gbytes, I->d_ovf, I->d_bytes);
-  hipLaunchKernelGGL(k_gather_copy, dim3(nq), dim3((uint32_t)I->gather_t), 0, I->stream,
-                     I->A.run(), I->DA.run(), I->spillA, I->heapA, I->d_rowsm,
-                     I->max_cap, I->d_found, I->d_qs, nq, I->d_gbuf, qcap,
-                     I->d_offs, I->d_ovf, I->gather_gwl);
+  if (I->gather_mode == 2) {
+    int64_t maxw = 0;
+    for (const DevRangeQ& q : qs) {
+      int64_t c = q.cap > 0 && q.cap < I->max_cap ? q.cap : I->max_cap;
+      if (c > maxw) maxw = c;
+    }
+    hipLaunchKernelGGL(k_gather_copy2,
+                       dim3(nq, (uint32_t)ceil_div(maxw, 4)), dim3(256), 0,
+                       I->stream, I->A.run(), I->DA.run(), I->spillA, I->heapA,
+                       I->d_rowsm, I->max_cap, I->d_found, I->d_qs, nq,
+                       I->d_gbuf, qcap, I->d_offs, I->d_ovf);
+  } else {
+    hipLaunchKernelGGL(k_gather_copy, dim3(nq), dim3((uint32_t)I->gather_t), 0, I->stream,
+                       I->A.run(), I->DA.run(), I->spillA, I->heapA, I->d_rowsm,
+                       I->max_cap, I->d_found, I->d_qs, nq, I->d_gbuf, qcap,
+                       I->d_offs, I->d_ovf, I->gather_gwl);
+  }
   HIP_CHECK(hipEventRecord(I->ev2, I->stream));
   return true;
 }
