@@ -1208,11 +1208,24 @@ bool Store::BenchStep(const uint8_t* qbuf, size_t nq, const uint8_t* tbuf,
   const bool pipe = (mode & 4) != 0;
   const bool d2h = (mode & 1) != 0;
   if (pipe && d2h) { if (err) *err = "pipelined bench step excludes d2h"; return false; }
+  static const bool tr = getenv("KB_TRACE") && *getenv("KB_TRACE");
+  auto lt = std::chrono::steady_clock::now();
+  auto lap = [&](const char* what) {
+    if (!tr) return;
+    auto now = std::chrono::steady_clock::now();
+    static int n = 0;
+    if (++n % 397 < 8)
+      fprintf(stderr, "[step] %s %.0f us\n", what,
+              std::chrono::duration<double>(now - lt).count() * 1e6);
+    lt = now;
+  };
   if (!syncReads(err)) return false;
+  lap("syncReads");
   std::vector<DevRangeQ> qall;
   std::vector<int64_t> limits;
   std::string qtails;
   parseBenchQueries(qbuf, nq, committed_, mode, &qall, &limits, &qtails);
+  lap("parse");
   // pipelined (mode bit2): the previous step's range kernels ran while the
   // host applied its txns and parsed this step; collect them now, return
   // THEIR totals, and leave this step's batch in flight (range reads are
@@ -1220,6 +1233,7 @@ bool Store::BenchStep(const uint8_t* qbuf, size_t nq, const uint8_t* tbuf,
   // results)
   unsigned long long prev_tot = 0;
   if (pipe && !finishPendingBench(&prev_tot, err)) return false;
+  lap("finish_prev");
   // txn ops parsed up front so the batched CAS lookup (f1) can launch BEFORE
   // the range batch: stream order runs the small lookup first, the host
   // applies the conditional-update protocol against the device results while
@@ -1260,12 +1274,16 @@ bool Store::BenchStep(const uint8_t* qbuf, size_t nq, const uint8_t* tbuf,
     }
     if (!slab_->GetBatchStart(gq, err, gtails)) return false;
   }
+  lap("txn_parse_getstart");
   if (!slab_->RangeBatchStart(qall, err, qtails)) return false;
+  lap("range_start");
   if (ntx > 0) {
     if (tuniq) {
       std::vector<kbslab::GetResult> cur;
       if (!slab_->GetBatchFinish((int)ntx, &cur, err)) return false;
+      lap("get_finish");
       if (!applyTxnOps(tops.data(), ntx, cur, out_revs, err)) return false;
+      lap("apply_txn");
     } else {
       // duplicate keys: serial reference protocol (overlapped, host-only)
       for (size_t i = 0; i < ntx; ++i) {
