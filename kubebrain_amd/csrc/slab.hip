@@ -1266,6 +1266,30 @@ struct Slab::Impl {
   DevRangeQ* h_qs[2] = {nullptr, nullptr};  // pinned query-upload mirrors
   hipEvent_t ev_qs[2] = {nullptr, nullptr};
   int qs_idx = 0;
+  // pinned double-buffered staging for heap/spill appends: the old pageable
+  // copies needed a FULL stream sync per call (in the pipelined bench that
+  // waited out the previous step's scan+gather kernels)
+  uint8_t* h_blob[2] = {nullptr, nullptr};
+  int64_t h_blob_cap[2] = {0, 0};
+  hipEvent_t ev_blob[2] = {nullptr, nullptr};
+  int blob_idx = 0;
+  uint8_t* stage_blob(const void* src, int64_t len, std::string* err) {
+    int slot = blob_idx ^= 1;
+    if (!ev_blob[slot]) {
+      hipError_t _e = hipEventCreate(&ev_blob[slot]);
+      if (_e != hipSuccess) { if (err) *err = hipGetErrorString(_e); return nullptr; }
+    }
+    (void)hipEventSynchronize(ev_blob[slot]);
+    if (len > h_blob_cap[slot]) {
+      if (h_blob[slot]) (void)hipHostFree(h_blob[slot]);
+      int64_t cap = len + len / 2 + 4096;
+      hipError_t _e = hipHostMalloc(&h_blob[slot], cap);
+      if (_e != hipSuccess) { if (err) *err = hipGetErrorString(_e); h_blob[slot] = nullptr; h_blob_cap[slot] = 0; return nullptr; }
+      h_blob_cap[slot] = cap;
+    }
+    memcpy(h_blob[slot], src, len);
+    return h_blob[slot];
+  }
   int64_t mergebuf_cap = 0;
   bool ensure_mergebuf(int64_t m, std::string* err) {
     if (m + 1 <= mergebuf_cap) return true;
@@ -1432,6 +1456,8 @@ struct Slab::Impl {
       if (ev_up[i]) (void)hipEventDestroy(ev_up[i]);
       if (h_qs[i]) (void)hipHostFree(h_qs[i]);
       if (ev_qs[i]) (void)hipEventDestroy(ev_qs[i]);
+      if (h_blob[i]) (void)hipHostFree(h_blob[i]);
+      if (ev_blob[i]) (void)hipEventDestroy(ev_blob[i]);
     }
     if (h_gmeta) (void)hipHostFree(h_gmeta);
     if (h_gfound) (void)hipHostFree(h_gfound);
@@ -1779,11 +1805,23 @@ bool Slab::SpillAppend(const void* src, int64_t len, int64_t* off,
     if (err) *err = "key-spill heap full (KB_SPILL_BYTES)";
     return false;
   }
-  HIP_CHECK(hipMemcpyAsync(I->spillA + I->spill_used_, src, len,
+  if (len > (8 << 20)) {
+    HIP_CHECK(hipMemcpyAsync(I->spillA + I->spill_used_, src, len,
+                             hipMemcpyHostToDevice, I->stream));
+    HIP_CHECK(hipMemcpyAsync(I->spillB + I->spill_used_, src, len,
+                             hipMemcpyHostToDevice, I->stream));
+    HIP_CHECK(hipStreamSynchronize(I->stream));
+    *off = I->spill_used_;
+    I->spill_used_ += len;
+    return true;
+  }
+  uint8_t* h = I->stage_blob(src, len, err);
+  if (!h) return false;
+  HIP_CHECK(hipMemcpyAsync(I->spillA + I->spill_used_, h, len,
                            hipMemcpyHostToDevice, I->stream));
-  HIP_CHECK(hipMemcpyAsync(I->spillB + I->spill_used_, src, len,
+  HIP_CHECK(hipMemcpyAsync(I->spillB + I->spill_used_, h, len,
                            hipMemcpyHostToDevice, I->stream));
-  HIP_CHECK(hipStreamSynchronize(I->stream));
+  HIP_CHECK(hipEventRecord(I->ev_blob[I->blob_idx], I->stream));
   *off = I->spill_used_;
   I->spill_used_ += len;
   return true;
@@ -1796,11 +1834,23 @@ bool Slab::HeapAppend(const void* src, int64_t len, int64_t* off, std::string* e
     if (err) *err = "heap full (KB_HEAP_BYTES)";
     return false;
   }
-  HIP_CHECK(hipMemcpyAsync(I->heapA + I->heap_used_, src, len,
+  if (len > (8 << 20)) {  // bulk loads: pageable copy + sync beats GB-scale pinned staging
+    HIP_CHECK(hipMemcpyAsync(I->heapA + I->heap_used_, src, len,
+                             hipMemcpyHostToDevice, I->stream));
+    HIP_CHECK(hipMemcpyAsync(I->heapB + I->heap_used_, src, len,
+                             hipMemcpyHostToDevice, I->stream));
+    HIP_CHECK(hipStreamSynchronize(I->stream));
+    *off = I->heap_used_;
+    I->heap_used_ += len;
+    return true;
+  }
+  uint8_t* h = I->stage_blob(src, len, err);
+  if (!h) return false;
+  HIP_CHECK(hipMemcpyAsync(I->heapA + I->heap_used_, h, len,
                            hipMemcpyHostToDevice, I->stream));
-  HIP_CHECK(hipMemcpyAsync(I->heapB + I->heap_used_, src, len,
+  HIP_CHECK(hipMemcpyAsync(I->heapB + I->heap_used_, h, len,
                            hipMemcpyHostToDevice, I->stream));
-  HIP_CHECK(hipStreamSynchronize(I->stream));
+  HIP_CHECK(hipEventRecord(I->ev_blob[I->blob_idx], I->stream));
   *off = I->heap_used_;
   I->heap_used_ += len;
   return true;
