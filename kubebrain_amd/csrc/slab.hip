@@ -841,22 +841,6 @@ __global__ void k_heap_scatter(const uint64_t* __restrict__ keep,
 }
 
 // ---- merge (memtable flush): delta ranks + scatter ----------------------
-__global__ void k_merge_rank(Run a, int64_t n, Run dnew, int64_t m,
-                             const uint8_t* __restrict__ spill,
-                             uint64_t* __restrict__ rank,
-                             uint64_t* __restrict__ drop) {
-  int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
-  if (i >= n) return;
-  QKey k = row_qk(a, spill, i);
-  uint64_t r = a.rev[i];
-  int64_t lb = d_lb_range(dnew, spill, 0, m, k, r);
-  rank[i] = (uint64_t)lb;
-  // delta rev-rows REPLACE base rev-rows of the same key (memtable holds the
-  // updated revision-row value; equal internal key only happens at rev==0)
-  bool eq = lb < m && dnew.rev[lb] == r && rowcmp_q(dnew, spill, lb, k) == 0;
-  drop[i] = eq ? 1 : 0;
-}
-
 // inverted small-insert merge (m <= 1024 new rows into an n-row run): ONE
 // block searches each NEW row into the big run — m*log2(n) work instead of
 // k_merge_rank's n*log2(m) — writes the new rows at their final slots, and
@@ -996,24 +980,6 @@ __global__ void k_same_next_fixup(uint8_t* __restrict__ keys,
   uint64_t mm = meta[i];
   bool same = i + 1 < new_n && rows_same_key(keys, meta, ko, spill, i, i + 1);
   meta[i] = same ? (mm | M_SAME_NEXT) : (mm & ~M_SAME_NEXT);
-}
-
-__global__ void k_merge_scatter_base(
-    Run a, const uint64_t* __restrict__ rank,
-    const uint64_t* __restrict__ drop, const uint64_t* __restrict__ dropx,
-    uint8_t* keysB, uint64_t* metaB, uint64_t* revB, uint64_t* voB,
-    uint64_t* koB, int64_t n) {
-  int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
-  if (i >= n || drop[i]) return;
-  int64_t j = i - (int64_t)dropx[i] + (int64_t)rank[i];
-  revB[j] = a.rev[i];
-  metaB[j] = a.meta[i];
-  voB[j] = a.vo[i];
-  koB[j] = a.ko[i];
-  const uint64_t* ks = (const uint64_t*)(a.keys + i * KEYW);
-  uint64_t* kd = (uint64_t*)(keysB + j * KEYW);
-#pragma unroll
-  for (int k = 0; k < KEYW / 8; ++k) kd[k] = ks[k];
 }
 
 __global__ void k_merge_scatter_delta(
@@ -1579,9 +1545,8 @@ struct Slab::Impl {
     }
     if (n > 0 && m > 0) {
       // big inverted path (folds, bulk appends): m*log2(n) probes + two
-      // prefix sums + streaming scatters. The forward k_merge_rank path
-      // (n*log2(m) random probes — ~400M for a 20.7M x 512k fold) is kept
-      // only for n==0 below.
+      // prefix sums + streaming scatters (the forward n*log2(m) rank path
+      // paid ~400M random probes for a 20.7M x 512k fold and is gone).
       if (!ensure_mergebuf(m, err)) return false;
       HIP_CHECK(hipMemsetAsync(s_a, 0, (n + 2) * 8, stream));
       HIP_CHECK(hipMemsetAsync(s_b, 0, n * 8, stream));
@@ -1613,18 +1578,8 @@ struct Slab::Impl {
       *out_n = new_n;
       return true;
     }
-    if (n > 0) {
-      int64_t nb = ceil_div(n, 256);
-      hipLaunchKernelGGL(k_merge_rank, dim3((uint32_t)nb), dim3(256), 0, stream,
-                         src.run(), n, dnew, m, spillA, s_a, s_b);
-      HIP_CHECK(hipMemsetAsync(s_b + n, 0, 8, stream));
-      if (!scan(s_b, s_c, n + 1, nullptr, err)) return false;  // no host sync
-      hipLaunchKernelGGL(k_merge_scatter_base, dim3((uint32_t)nb), dim3(256), 0,
-                         stream, src.run(), s_a, s_b,
-                         s_c, dst.keys, dst.meta, dst.rev, dst.vo, dst.ko, n);
-    } else {
-      HIP_CHECK(hipMemsetAsync(s_c, 0, 8, stream));
-    }
+    // n == 0 here (every caller guards m > 0; n > 0 took the inverted path)
+    HIP_CHECK(hipMemsetAsync(s_c, 0, 8, stream));
     if (m > 0) {
       int64_t mb = ceil_div(m, 256);
       hipLaunchKernelGGL(k_merge_scatter_delta, dim3((uint32_t)mb), dim3(256),
