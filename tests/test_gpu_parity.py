@@ -782,3 +782,56 @@ def test_bench_step_pipelined_matches_sync():
     assert t_pipe[0] == 0
     assert t_pipe[1:] == t_sync[:-1]
     assert sum(t_sync) > 0
+
+
+@pytest.mark.parametrize("knobs", [
+    {"KB_SCAN_T": "256"},
+    {"KB_SCAN_T": "512"},
+    {"KB_GATHER_MODE": "2"},
+    {"KB_GATHER_GW": "8"},
+    {"KB_DELTA_CAP": "4096"},  # frequent folds exercise shadow resets
+])
+def test_knob_paths_keep_parity(knobs):
+    """The runtime-tunable kernel variants (scan block width, gather record
+    layout, delta fold cadence) must all produce byte-identical results —
+    the bench may run any of them. Env is read at store creation, so a fresh
+    Dual per knob-set suffices."""
+    old = {k: os.environ.get(k) for k in knobs}
+    os.environ.update(knobs)
+    try:
+        d = parity.Dual()
+        try:
+            rng = random.Random(hash(tuple(sorted(knobs))) & 0xFFFF)
+            ns = [b"/registry/pods/kn-%02d" % i for i in range(4)]
+            live = {}
+            for step in range(500):
+                op = rng.random()
+                k = ns[rng.randrange(4)] + b"/o-%04d" % rng.randrange(150)
+                if op < 0.4:
+                    r = d.create(k, b"v%d" % step)
+                    if r.succeeded:
+                        live[k] = r.header_revision
+                elif op < 0.6:
+                    r = d.update(k, b"u%d" % step, live.get(k, 0))
+                    if r.succeeded:
+                        live[k] = r.header_revision
+                elif op < 0.7:
+                    d.delete(k, 0)
+                    live.pop(k, None)
+                elif op < 0.95:
+                    lo = ns[rng.randrange(4)]
+                    d.list(lo + b"/", lo + b"0", 0,
+                           rng.choice([0, 5, 20, 200]))
+                else:
+                    d.get(k, 0)
+                if step == 300:
+                    d.compact(max(1, d.p.current_rev() - 50))
+            d.diff_dump()
+        finally:
+            d.close()
+    finally:
+        for k, v in old.items():
+            if v is None:
+                os.environ.pop(k, None)
+            else:
+                os.environ[k] = v
