@@ -197,14 +197,8 @@ int kb_watch_poll(kb_store* h, long long wid, uint8_t* out, size_t cap,
   // KB_ENOBUF can be retried with a larger buffer without losing events
   // (contiguous-revision delivery, backend.go:214-238). *out_len carries the
   // required size on KB_ENOBUF.
-  Status st;
-  size_t need = 0;
-  auto evs = ((Store*)h)->WatchPollLimited(wid, cap, &need, &st);
-  if (st == kbstore::NOBUF) { *out_len = need; return KB_ENOBUF; }
-  Writer w{out, cap};
-  writeEvents(w, evs);
-  *out_len = w.off;
-  if (w.overflow) return KB_ENOBUF;  // unreachable: size was pre-checked
+  Status st = ((Store*)h)->WatchPollWire(wid, out, cap, out_len);
+  if (st == kbstore::NOBUF) return KB_ENOBUF;
   return st;
 }
 

@@ -384,8 +384,6 @@ __global__ void k_range_scan2(
   if (q >= nq) return;
   __shared__ int wave_cnt[SCAN_T_MAX / 64];
   const DevRangeQ& Q = qs[q];
-  const QKey qstart{Q.start, Q.start_klen, Q.start_ko, qtails};
-  const QKey qend{Q.end, Q.end_klen, Q.end_ko, qtails};
   const bool dbg0 = dbg && threadIdx.x == 0;
   unsigned long long tk0 = dbg0 ? wall_clock64() : 0;
   // bounds resolved by the k_range_bounds pre-pass

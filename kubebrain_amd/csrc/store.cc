@@ -397,7 +397,6 @@ Status Store::StreamNext(int64_t sid, std::vector<KeyValue>* kvs) {
   // (key, rev=+inf) bound past the last chunk's final key (exact for
   // full-width keys; see List).
   const int64_t max_cap = slab_->max_winner_cap();
-  const bool tr = getenv("KB_TRACE") && *getenv("KB_TRACE");
   std::vector<kbslab::RangeResult> outs;
   while ((int64_t)kvs->size() < 300) {
     DevRangeQ q{};
@@ -604,6 +603,81 @@ std::vector<Event> Store::WatchPollLimited(int64_t wid, size_t max_bytes,
   w.pend_events = 0;
   *st = OK;
   return out;
+}
+
+// wire-direct poll: serializes delivered events straight from the ring into
+// the caller's buffer (the kb_watch_poll format: u32 count, then per event
+// i32 type | u64 rev | u64 kv_rev | str key | str value) — no intermediate
+// Event copies, and the NOBUF path sizes without copying. Same delivery
+// contract as WatchPollLimited: refs drain only when everything fits.
+Status Store::WatchPollWire(int64_t wid, uint8_t* out, size_t cap,
+                            size_t* out_len) {
+  std::lock_guard<std::recursive_mutex> lk(mu_);
+  pumpEvents();
+  *out_len = 0;
+  auto it = watchers_.find(wid);
+  if (it == watchers_.end()) return WATCH_DROPPED;
+  Watcher& w = it->second;
+  auto stale = [&]() {
+    for (const PendRef& pr : w.prefs)
+      if (pr.base < ring_.e - (int64_t)ring_.l) return true;
+    return false;
+  };
+  if (!w.dropped && stale()) {
+    w.dropped = true;
+    w.prefs.clear();
+    w.pend_events = 0;
+  }
+  if (w.dropped) {
+    releaseSlot(w);
+    watchers_.erase(it);
+    return WATCH_DROPPED;
+  }
+  const bool longpfx = w.prefix.size() > (size_t)KEYW;
+  // pass 1: exact count + wire size (no copies)
+  size_t need = 4;
+  uint32_t count = 0;
+  auto each = [&](auto&& fn) {
+    for (const PendRef& pr : w.prefs)
+      for (int c = 0; c * 64 < pr.count; ++c) {
+        uint64_t bits = pr.words[c];
+        while (bits) {
+          int j = __builtin_ctzll(bits);
+          bits &= bits - 1;
+          const Event& ev = ring_.arr[(pr.base + c * 64 + j) % ring_.l];
+          if (longpfx &&
+              ev.kv_key.compare(0, w.prefix.size(), w.prefix) != 0)
+            continue;
+          fn(ev);
+        }
+      }
+  };
+  each([&](const Event& ev) {
+    need += 28 + ev.kv_key.size() + ev.kv_value.size();
+    ++count;
+  });
+  *out_len = need;
+  if (need > cap) return NOBUF;  // refs intact: retry with a larger buffer
+  // pass 2: serialize straight from the ring
+  uint8_t* p = out;
+  auto put32 = [&](uint32_t v) { memcpy(p, &v, 4); p += 4; };
+  auto put64 = [&](uint64_t v) { memcpy(p, &v, 8); p += 8; };
+  put32(count);
+  each([&](const Event& ev) {
+    put32((uint32_t)ev.type);
+    put64(ev.revision);
+    put64(ev.kv_revision);
+    put32((uint32_t)ev.kv_key.size());
+    memcpy(p, ev.kv_key.data(), ev.kv_key.size());
+    p += ev.kv_key.size();
+    put32((uint32_t)ev.kv_value.size());
+    memcpy(p, ev.kv_value.data(), ev.kv_value.size());
+    p += ev.kv_value.size();
+  });
+  *out_len = (size_t)(p - out);
+  w.prefs.clear();
+  w.pend_events = 0;
+  return OK;
 }
 
 void Store::WatchCancel(int64_t wid) {

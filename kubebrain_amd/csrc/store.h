@@ -90,6 +90,8 @@ class Store {
 
   int64_t Watch(const Bytes& prefix, uint64_t revision, Status* st);
   std::vector<Event> WatchPoll(int64_t wid, Status* st);
+  // wire-direct poll (kb_watch_poll format), no intermediate Event copies
+  Status WatchPollWire(int64_t wid, uint8_t* out, size_t cap, size_t* out_len);
   // non-destructive on overflow: if the serialized size exceeds max_bytes,
   // returns NOBUF (need_bytes = required size) with the queue intact
   std::vector<Event> WatchPollLimited(int64_t wid, size_t max_bytes,
