@@ -197,6 +197,12 @@ void Store::putRevRow(const Bytes& key, uint64_t objrev, bool flag9) {
 
 bool Store::syncReads(std::string* err) {
   if (!fatal_.empty()) { if (err) *err = "store failed: " + fatal_; return false; }
+  // an uncollected pipelined bench batch shares the range result buffers
+  // with every other read path; drain it before any non-BenchStep work
+  // (BenchStep defers its own collection deliberately)
+  if (bench_pending_nq_ >= 0 && !in_bench_step_ &&
+      !finishPendingBench(nullptr, err))
+    return false;
   sync_n_++;
   auto t0 = std::chrono::steady_clock::now();
   if (!heap_pending_.empty()) {
@@ -1282,6 +1288,8 @@ bool Store::BenchStep(const uint8_t* qbuf, size_t nq, const uint8_t* tbuf,
   const bool pipe = (mode & 4) != 0;
   const bool d2h = (mode & 1) != 0;
   if (pipe && d2h) { if (err) *err = "pipelined bench step excludes d2h"; return false; }
+  in_bench_step_ = true;
+  struct Reset { bool* f; ~Reset() { *f = false; } } _rst{&in_bench_step_};
   static const bool tr = getenv("KB_TRACE") && *getenv("KB_TRACE");
   auto lt = std::chrono::steady_clock::now();
   auto lap = [&](const char* what) {

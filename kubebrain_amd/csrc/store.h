@@ -233,6 +233,7 @@ class Store {
   // overlap the host's txn apply + next-step prep. Valid only between
   // consecutive BenchStep calls; Sync() drains it.
   int bench_pending_nq_ = -1;
+  bool in_bench_step_ = false;
   std::vector<int64_t> bench_pending_limits_;
   bool finishPendingBench(unsigned long long* total, std::string* err);
   int64_t delivered_ = 0;  // watch events enqueued to watchers (fan-out)
