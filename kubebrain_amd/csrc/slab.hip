@@ -427,7 +427,13 @@ __global__ void k_range_scan2(
       if (pos < cap_m) outm[pos] = rt;
     }
   }
-  if (dbg0) { unsigned long long tke = wall_clock64(); atomicAdd(&dbg[4], tke - tkm); atomicAdd(&dbg[5], tke - tk0); }
+  if (dbg0) {
+    unsigned long long tke = wall_clock64();
+    atomicAdd(&dbg[4], tke - tkm);
+    atomicAdd(&dbg[5], tke - tk0);
+    atomicMax(&dbg[6], tke - tk0);  // slowest block (tail skew diagnostic)
+    atomicAdd(&dbg[7], 1ull);
+  }
   if (threadIdx.x == 0) {
     total_out[q] = btotal + dtotal;
     found_out[q] = Q.count_only ? 0 : cap_m;
@@ -2094,8 +2100,11 @@ bool Slab::RangeBatchFinish(int nq, bool d2h, bool parse,
     perf.dbg_c += (double)c[2] / I->wall_khz;
     perf.dbg_d += (double)c[3] / I->wall_khz;
     perf.dbg_e += (double)c[4] / I->wall_khz;
-    fprintf(stderr, "[scan_dbg] staged=%llu/%d span_sum=%llu\n", c[6], nq,
-            c[7]);
+    fprintf(stderr,
+            "[scan_dbg] blocks=%llu total_ms=%.2f max_block_us=%.1f avg_block_us=%.1f\n",
+            c[7], (double)c[5] / I->wall_khz,
+            c[7] ? (double)c[6] / I->wall_khz * 1e3 : 0.0,
+            c[7] ? (double)c[5] / c[7] / I->wall_khz * 1e3 : 0.0);
   }
   perf.scan_launches++;
   perf.gather_launches++;
