@@ -370,38 +370,6 @@ __global__ void k_range_bounds(Run b, int64_t n, Run d, int64_t dn,
   if (w < 4 && (threadIdx.x & 63) == 0) bounds_g[(int64_t)q * 4 + w] = r;
 }
 
-// longest-first block order for the scan: hot-namespace queries (large
-// spans, sparse winners) take 3-10x the average block time;
-// launching them first cuts the launch's tail. One block, bitonic sort of
-// up to 1024 (span, query) pairs in LDS — results are order-invariant
-// (every block still owns one whole query).
-__global__ void k_order_queries(const int64_t* __restrict__ bounds_g, int nq,
-                                int* __restrict__ order) {
-  __shared__ uint64_t kv[1024];  // span<<10 | q  (nq <= 1024 = max_q)
-  int t = threadIdx.x;
-  uint64_t v = 0;  // padding sorts last (descending)
-  if (t < nq) {
-    uint64_t span = (uint64_t)(bounds_g[t * 4 + 1] - bounds_g[t * 4 + 0]) +
-                    (uint64_t)(bounds_g[t * 4 + 3] - bounds_g[t * 4 + 2]);
-    if (span > (1ull << 53)) span = 1ull << 53;
-    v = (span << 10) | (uint64_t)t;
-  }
-  kv[t] = v;
-  __syncthreads();
-  for (int k = 2; k <= 1024; k <<= 1) {
-    for (int j = k >> 1; j > 0; j >>= 1) {
-      int ixj = t ^ j;
-      if (ixj > t) {
-        bool up = (t & k) == 0;  // descending overall
-        uint64_t a = kv[t], c = kv[ixj];
-        if (up ? (a < c) : (a > c)) { kv[t] = c; kv[ixj] = a; }
-      }
-      __syncthreads();
-    }
-  }
-  if (t < nq) order[t] = (int)(kv[t] & 1023);
-}
-
 __global__ void k_range_scan2(
     Run b, int64_t n, Run d, int64_t dn, const uint8_t* __restrict__ spill,
     const uint8_t* __restrict__ qtails, const DevRangeQ* __restrict__ qs,
@@ -411,10 +379,9 @@ __global__ void k_range_scan2(
     int64_t* __restrict__ total_out, unsigned long long* __restrict__ scanned_out,
     const int64_t* __restrict__ bounds_g,  // from k_range_bounds
     const uint64_t* __restrict__ shadow,  // base shadow-revision column
-    const int* __restrict__ qorder,  // longest-first block->query map
     unsigned long long* dbg) {  // KB_SCAN_DBG phase cycles (null in prod)
-  if (blockIdx.x >= (unsigned)nq) return;
-  int q = qorder ? qorder[blockIdx.x] : (int)blockIdx.x;
+  int q = blockIdx.x;
+  if (q >= nq) return;
   __shared__ int wave_cnt[SCAN_T_MAX / 64];
   const DevRangeQ& Q = qs[q];
   const bool dbg0 = dbg && threadIdx.x == 0;
@@ -1259,7 +1226,6 @@ struct Slab::Impl {
 
   // scan scratch (u64, shared across ops)
   uint64_t* shadow = nullptr;  // [max_rows] base shadow-revision column
-  int* d_qorder = nullptr;     // [1024] longest-first scan block order
   uint64_t* d_outpos = nullptr;  // [1024] inverted-insert dest slots
   // big inverted merge scratch: dup flags / their scan / new-row lower
   // bounds (then final slots), all m-sized, grown on demand
@@ -1441,8 +1407,7 @@ struct Slab::Impl {
                     (void*)d_gq, (void*)d_rows, (void*)d_offs,
                     (void*)d_found,  // base of the resmeta block
                     (void*)d_found32, (void*)d_orev, (void*)d_ometa,
-                    (void*)shadow, (void*)d_qorder, (void*)d_outpos,
-                    (void*)d_mdup, (void*)d_mdupx,
+                    (void*)shadow, (void*)d_outpos, (void*)d_mdup, (void*)d_mdupx,
                     (void*)d_mlbs, (void*)d_gbuf,
                     (void*)d_pack, (void*)d_goffs, (void*)d_bounds,
                     (void*)d_bkeys, (void*)d_brevs, (void*)d_dkeys,
@@ -2038,19 +2003,12 @@ bool Slab::RangeBatchStart(const std::vector<DevRangeQ>& qs, std::string* err,
   hipLaunchKernelGGL(k_range_bounds, dim3(nq), dim3(256), 0, I->stream,
                      I->A.run(), I->n, I->DA.run(), I->dn, I->spillA,
                      I->d_qtails, I->d_qs, nq, I->d_bounds4);
-  const int* qorder = nullptr;
-  if (nq > 64 && nq <= 1024) {  // longest-first pays once the grid is wide
-    if (!I->d_qorder) HIP_CHECK(hipMalloc(&I->d_qorder, 1024 * 4));
-    hipLaunchKernelGGL(k_order_queries, dim3(1), dim3(1024), 0, I->stream,
-                       I->d_bounds4, nq, I->d_qorder);
-    qorder = I->d_qorder;
-  }
   hipLaunchKernelGGL(k_range_scan2, dim3(nq), dim3((uint32_t)I->scan_t), 0, I->stream,
                      I->A.run(), I->n, I->DA.run(), I->dn, I->spillA,
                      I->d_qtails, I->d_qs, nq,
                      I->max_cap, I->d_rows, I->d_rows2, I->d_rowsm, I->d_found,
                      I->d_total, I->d_scanned, I->d_bounds4, I->shadow,
-                     qorder, I->d_dbg);
+                     I->d_dbg);
   HIP_CHECK(hipEventRecord(I->ev1, I->stream));
   hipLaunchKernelGGL(k_gather, dim3(nq), dim3(256), 0, I->stream, I->A.run(),
                      I->DA.run(), I->heapA, I->d_rowsm, I->max_cap,
