@@ -499,3 +499,41 @@ def test_list_enobuf_grow_and_retry():
             assert len(kvs) == 40
     finally:
         s.close()
+
+
+def test_stream_batch_enobuf_grow_and_retry():
+    """Same grow-and-retry contract on the streaming read path (the cgo
+    stream reader's KB_ENOBUF loop), via the oracle's one-shot batch API
+    (okb_stream_batch — the oracle mirrors ListByStream batch-wise; the
+    product's handle-based kb_stream_next is diffed against it on GPU)."""
+    import ctypes as C
+
+    from kbclient import ENOBUF, OK, open_oracle
+
+    s = open_oracle()
+    try:
+        s.set_current_rev(1000)
+        pfx = b"/registry/cfg/sb"
+        for i in range(20):
+            assert s.create(pfx + b"/k-%03d" % i, b"w" * 64).succeeded
+        f = s._f("stream_batch")
+        out_len = C.c_size_t()
+        rr = C.c_uint64()
+        tiny = C.create_string_buffer(8)
+        rc = f(C.c_void_p(s.h), pfx + b"/", C.c_size_t(len(pfx) + 1),
+               pfx + b"0", C.c_size_t(len(pfx) + 1), C.c_uint64(0),
+               C.c_uint64(0), tiny, C.c_size_t(8), C.byref(out_len),
+               C.byref(rr))
+        assert rc == ENOBUF and out_len.value > 8
+        need = out_len.value
+        exact = C.create_string_buffer(need)
+        rc = f(C.c_void_p(s.h), pfx + b"/", C.c_size_t(len(pfx) + 1),
+               pfx + b"0", C.c_size_t(len(pfx) + 1), C.c_uint64(0),
+               C.c_uint64(0), exact, C.c_size_t(need), C.byref(out_len),
+               C.byref(rr))
+        assert rc == OK and out_len.value == need
+        import kbclient
+        kvs = kbclient._parse_kvs(exact.raw[:out_len.value])
+        assert len(kvs) == 20
+    finally:
+        s.close()
